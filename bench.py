@@ -1,0 +1,193 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: NoisyNet 4-layer CIFAR-10 ConvNet, 4-bit quantized
+activations+weights, analog noise I_max=1nA, act_max=5, bf16, synthetic
+4-bit-CIFAR-shaped data (BASELINE.json metric: images/sec whole node).
+
+Single GPU:       python bench.py --steps 30 --warmup 10
+Multi-GPU (driver): python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Weak scaling: per-GPU batch is fixed (--batch), global batch = N * batch.
+Each timed step is a FULL training step: forward (fused quant+conv+sigma+
+noise kernels), cross-entropy, backward (dgrad/wgrad), bucketed RCCL
+all-reduce overlapped with backward, fused SGD update with weight clamp.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from noisynet_amd import distributed as dist_mod  # noqa: E402
+from noisynet_amd import optim as native_optim  # noqa: E402
+from noisynet_amd import utils  # noqa: E402
+from noisynet_amd.config import broadcast_per_layer, build_noisynet_parser  # noqa: E402
+from noisynet_amd.models.noisynet import Net  # noqa: E402
+from noisynet_amd.quant import finish_calibration, start_calibration  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--batch", type=int, default=2048,
+                   help="per-GPU batch size (weak scaling)")
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp32"])
+    p.add_argument("--no-noise", action="store_true",
+                   help="disable analog noise (noise-free baseline config)")
+    p.add_argument("--seed", type=int, default=42)
+    return p.parse_args()
+
+
+def flagship_args(bench):
+    """BASELINE.json config 2: NoisyNet CIFAR-10 4-bit, I_max=1nA, act_max=5."""
+    argv = ["--q_a", "4", "--act_max", "5", "--w_max1", "0.3",
+            "--LR", "0.005", "--L2_1", "0.0005", "--L2_2", "0.0002",
+            "--batch_size", str(bench.batch), "--optim", "SGD",
+            "--calculate_running", "--no-augment"]
+    if not bench.no_noise:
+        argv = ["--current", "1"] + argv
+    args = build_noisynet_parser().parse_args(argv)
+    broadcast_per_layer(args)
+    return args
+
+
+def main():
+    bench = parse_args()
+    torch.manual_seed(bench.seed)
+
+    world_size = dist_mod.env_world_size()
+    rank = dist_mod.env_rank()
+    distributed = dist_mod.init_distributed()
+    n_gpus = world_size if distributed else 1
+
+    if torch.cuda.is_available():
+        device = torch.device("cuda", dist_mod.env_local_rank())
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    dtype = torch.bfloat16 if bench.dtype == "bf16" else torch.float32
+    args = flagship_args(bench)
+
+    model = Net(args)
+    utils.init_model(model, args)
+    model = model.to(device)
+    if dtype is torch.bfloat16:
+        model = model.bfloat16()
+        # BN statistics stay fp32 for stability
+        for m in model.modules():
+            if isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+                m.float()
+    if device.type == "cuda":
+        model = model.to(memory_format=torch.channels_last)
+
+    dp = dist_mod.DataParallel(model) if distributed else None
+
+    # synthetic 4-bit CIFAR-shaped data, resident on device
+    g = torch.Generator().manual_seed(bench.seed + rank)
+    data = (torch.randint(0, 16, (bench.batch, 3, 32, 32), generator=g)
+            .to(device=device, dtype=dtype) / 15.0)
+    if device.type == "cuda":
+        data = data.contiguous(memory_format=torch.channels_last)
+    labels = torch.randint(0, 10, (bench.batch,), generator=g).to(device)
+
+    param_groups = [
+        {"params": model.conv1.parameters(), "weight_decay": args.L2_1,
+         "lr": args.LR, "clamp": (-args.w_max1, args.w_max1)},
+        {"params": model.conv2.parameters(), "weight_decay": args.L2_2,
+         "lr": args.LR},
+        {"params": model.linear1.parameters(), "weight_decay": args.L2_3,
+         "lr": args.LR},
+        {"params": model.linear2.parameters(), "weight_decay": args.L2_4,
+         "lr": args.LR},
+        {"params": [p for m in (model.bn1, model.bn2, model.bn3, model.bn4)
+                    for p in m.parameters()], "weight_decay": 0.0,
+         "lr": args.LR},
+    ]
+    optimizer = native_optim.SGD(param_groups, lr=args.LR,
+                                 momentum=args.momentum, nesterov=True)
+    criterion = torch.nn.CrossEntropyLoss()
+
+    def step(i):
+        model.train()
+        out = model(data, 0, i)
+        loss = criterion(out.float(), labels)
+        optimizer.zero_grad(set_to_none=False)
+        loss.backward()
+        if dp is not None:
+            dp.finish()
+        optimizer.step()
+        return loss
+
+    # calibration (5 batches) then freeze ranges, as the reference does
+    start_calibration(model)
+    with torch.no_grad():
+        for i in range(5):
+            model(data, 0, i)
+    finish_calibration(model, device)
+
+    for i in range(bench.warmup):
+        step(i + 100)  # i>=20: telemetry off in steady state
+
+    if distributed:
+        torch.distributed.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(bench.steps):
+        step(i + 1000)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    if distributed:
+        torch.distributed.barrier()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if distributed:
+        elapsed = elapsed.to(device if device.type == "cuda" else "cpu")
+        torch.distributed.all_reduce(elapsed, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(elapsed.item())
+
+    ms_per_step = elapsed / bench.steps * 1000.0
+    images_per_sec = n_gpus * bench.batch * bench.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node) + top-1, NoisyNet 4-bit "
+                      "CIFAR-10 I_max=1nA, 1/2/4/8 GPU",
+            "value": round(images_per_sec, 2),
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": bench.steps,
+            "warmup": bench.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": bench.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "noisynet-4layer-cifar",
+                "global_batch": n_gpus * bench.batch,
+                "seq_len": None,
+                "image_size": 32,
+                "q_a": 4, "q_w": 0, "current_nA": 0 if bench.no_noise else 1,
+                "act_max": 5,
+                "parallelism": "dp%d" % n_gpus,
+            },
+        }))
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

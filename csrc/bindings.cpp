@@ -1,0 +1,93 @@
+// Python bindings for the NoisyNet-MI355X HIP extension.
+
+#include <torch/extension.h>
+
+// elementwise.hip
+torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value,
+                             double max_value, double stochastic, int64_t seed);
+torch::Tensor ste_mask(torch::Tensor grad, torch::Tensor x, double min_value,
+                       double max_value);
+torch::Tensor mult_uniform_noise(torch::Tensor x, double a, int64_t seed);
+torch::Tensor relu_clip_fwd(torch::Tensor x, bool relu, double act_max);
+std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed);
+
+// optimizer.hip
+void sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
+              double momentum, double wd, bool nesterov, double cmin,
+              double cmax);
+void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, int64_t step, double lr, double beta1,
+                double beta2, double eps, double wd, double cmin, double cmax);
+
+// softmax_xent.hip
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor target);
+torch::Tensor softmax_xent_bwd(torch::Tensor softmax, torch::Tensor target,
+                               double gscale);
+
+// bn_act.hip
+std::vector<torch::Tensor> bn_stats(torch::Tensor x);
+torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
+                         torch::Tensor invstd, torch::Tensor gamma,
+                         torch::Tensor beta, bool relu, double act_max);
+
+// pool.hip
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
+torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
+                             int64_t W);
+
+// conv_mfma.hip
+torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                       int64_t pad);
+std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
+                                          torch::Tensor wraw, torch::Tensor bias,
+                                          int64_t stride, int64_t pad,
+                                          int64_t sigma_mode, double factor,
+                                          int64_t seed, bool telem);
+torch::Tensor conv_dgrad(torch::Tensor gy, torch::Tensor w, int64_t stride,
+                         int64_t pad, int64_t H, int64_t W);
+torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
+                         int64_t pad, int64_t R, int64_t S);
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w);
+torch::Tensor linear_dgrad(torch::Tensor gy, torch::Tensor w);
+torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x);
+std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
+                                            torch::Tensor wraw,
+                                            torch::Tensor bias,
+                                            int64_t sigma_mode, double factor,
+                                            int64_t seed, bool telem);
+std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
+                                            int64_t stride, int64_t pad,
+                                            int64_t sigma_mode, double factor,
+                                            int64_t seed, bool telem);
+std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
+                                                   torch::Tensor wraw,
+                                                   int64_t sigma_mode,
+                                                   double factor, int64_t seed,
+                                                   bool telem);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fake_quant_fwd", &fake_quant_fwd);
+  m.def("ste_mask", &ste_mask);
+  m.def("mult_uniform_noise", &mult_uniform_noise);
+  m.def("relu_clip_fwd", &relu_clip_fwd);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("sgd_step", &sgd_step);
+  m.def("adamw_step", &adamw_step);
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("bn_stats", &bn_stats);
+  m.def("bn_act_fwd", &bn_act_fwd);
+  m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
+  m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
+  m.def("conv_fwd", &conv_fwd);
+  m.def("conv_fwd_fused", &conv_fwd_fused);
+  m.def("conv_dgrad", &conv_dgrad);
+  m.def("conv_wgrad", &conv_wgrad);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("linear_dgrad", &linear_dgrad);
+  m.def("linear_wgrad", &linear_wgrad);
+  m.def("linear_fwd_fused", &linear_fwd_fused);
+  m.def("sigma_noise_conv", &sigma_noise_conv);
+  m.def("sigma_noise_linear", &sigma_noise_linear_impl);
+}

@@ -1,0 +1,120 @@
+// Fused BatchNorm + ReLU + clip kernels (NHWC layout on GPU).
+//
+// bn_stats: per-channel biased mean/var via a single pass over the NHWC
+// tensor -- each block owns a channel-chunk and reduces sum / sumsq with
+// f32 accumulation (channels are the fastest-varying dim in NHWC, so lanes
+// read consecutive channels: fully coalesced).
+// bn_act_fwd: one elementwise pass applying (x-mean)*invstd*gamma+beta,
+// ReLU and the act_max clip.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+// Grid: (C + 63)/64 blocks in x, rows-chunks in y. Each block handles 64
+// channels x kBlock/64-row slab, accumulating into global atomics (f32).
+template <typename T>
+__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
+                                float* __restrict__ sumsq, int64_t rows, int C) {
+  int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
+  if (c >= C) return;
+  float s = 0.0f, sq = 0.0f;
+  for (int64_t r = rstart; r < rows; r += gridDim.y * (kBlock / 64)) {
+    float v = to_f32(x[r * C + c]);
+    s += v;
+    sq += v * v;
+  }
+  atomicAdd(&sum[c], s);
+  atomicAdd(&sumsq[c], sq);
+}
+
+template <typename T>
+__global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ invstd,
+                                  const float* __restrict__ gamma,
+                                  const float* __restrict__ beta, int64_t n,
+                                  int C, int do_relu, float act_max) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float v = (to_f32(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (do_relu) v = fmaxf(v, 0.0f);
+    if (act_max > 0.0f) v = fminf(v, act_max);
+    y[i] = from_f32<T>(v);
+  }
+}
+
+template <typename scalar_t> struct DevT { using type = scalar_t; };
+template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
+template <> struct DevT<at::Half> { using type = _Float16; };
+
+}  // namespace
+
+// x: NHWC-contiguous 4-D (memory_format=channels_last, passed as NCHW logical)
+// or 2-D [N, C] row-major. Returns (mean, biased var) as f32.
+std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
+  int C;
+  int64_t rows;
+  const void* ptr = x.data_ptr();
+  if (x.dim() == 4) {
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "bn_stats: expected channels_last");
+    C = (int)x.size(1);
+    rows = x.size(0) * x.size(2) * x.size(3);
+  } else {
+    TORCH_CHECK(x.is_contiguous());
+    C = (int)x.size(1);
+    rows = x.size(0);
+  }
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sum = torch::zeros({C}, opts);
+  auto sumsq = torch::zeros({C}, opts);
+  int gx = (C + 63) / 64;
+  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64), 1024);
+  NN_DISPATCH(x.scalar_type(),
+                                  "bn_stats", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)ptr,
+                       sum.data_ptr<float>(), sumsq.data_ptr<float>(), rows, C);
+  });
+  HIP_CHECK_LAST();
+  auto mean = sum / (double)rows;
+  auto var = sumsq / (double)rows - mean * mean;
+  return {mean, var.clamp_min(0)};
+}
+
+torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
+                         torch::Tensor invstd, torch::Tensor gamma,
+                         torch::Tensor beta, bool relu, double act_max) {
+  int C;
+  if (x.dim() == 4) {
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    C = (int)x.size(1);
+  } else {
+    TORCH_CHECK(x.is_contiguous());
+    C = (int)x.size(1);
+  }
+  auto y = torch::empty_like(x);
+  int64_t n = x.numel();
+  int blocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
+  NN_DISPATCH(x.scalar_type(),
+                                  "bn_act_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((bn_act_fwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)x.data_ptr(),
+                       (T*)y.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), n, C, relu ? 1 : 0,
+                       (float)act_max);
+  });
+  HIP_CHECK_LAST();
+  return y;
+}

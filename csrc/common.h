@@ -1,0 +1,141 @@
+// Common device utilities for the NoisyNet-MI355X kernels (gfx950 / CDNA4).
+//
+// Wave width on CDNA4 is 64; all kernels use 256-thread blocks (4 waves).
+// RNG is counter-based Philox4x32-10 keyed by (seed, linear element index),
+// so results are deterministic per seed and independent of launch geometry.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <cmath>
+#include <cstdint>
+
+#define WAVE 64
+#define DEV_INLINE __device__ __forceinline__
+
+// ---------------------------------------------------------------------------
+// dtype conversion helpers
+// ---------------------------------------------------------------------------
+template <typename T> DEV_INLINE float to_f32(T v);
+template <> DEV_INLINE float to_f32<float>(float v) { return v; }
+template <> DEV_INLINE float to_f32<__hip_bfloat16>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+template <> DEV_INLINE float to_f32<_Float16>(_Float16 v) { return (float)v; }
+
+template <typename T> DEV_INLINE T from_f32(float v);
+template <> DEV_INLINE float from_f32<float>(float v) { return v; }
+template <> DEV_INLINE __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+template <> DEV_INLINE _Float16 from_f32<_Float16>(float v) {
+  return (_Float16)v;
+}
+
+// ---------------------------------------------------------------------------
+// Philox4x32-10 counter-based RNG
+// ---------------------------------------------------------------------------
+struct Philox4 {
+  uint32_t x, y, z, w;
+};
+
+DEV_INLINE uint32_t mulhilo(uint32_t a, uint32_t b, uint32_t* hip) {
+  uint64_t p = (uint64_t)a * (uint64_t)b;
+  *hip = (uint32_t)(p >> 32);
+  return (uint32_t)p;
+}
+
+DEV_INLINE Philox4 philox4x32(uint64_t seed, uint64_t counter) {
+  uint32_t c0 = (uint32_t)counter, c1 = (uint32_t)(counter >> 32);
+  uint32_t c2 = 0u, c3 = 0u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int i = 0; i < 10; ++i) {
+    uint32_t hi0, hi1;
+    uint32_t lo0 = mulhilo(0xD2511F53u, c0, &hi0);
+    uint32_t lo1 = mulhilo(0xCD9E8D57u, c2, &hi1);
+    uint32_t n0 = hi1 ^ c1 ^ k0;
+    uint32_t n1 = lo1;
+    uint32_t n2 = hi0 ^ c3 ^ k1;
+    uint32_t n3 = lo0;
+    c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  return {c0, c1, c2, c3};
+}
+
+// uniform in [0, 1)
+DEV_INLINE float u01(uint32_t v) { return (float)v * (1.0f / 4294967296.0f); }
+// uniform in (0, 1]  (for log() in Box-Muller)
+DEV_INLINE float u01_open(uint32_t v) {
+  return ((float)v + 1.0f) * (1.0f / 4294967296.0f);
+}
+
+// two N(0,1) samples from one Philox draw (Box-Muller)
+DEV_INLINE void gauss2(uint64_t seed, uint64_t ctr, float* g0, float* g1) {
+  Philox4 p = philox4x32(seed, ctr);
+  float r = sqrtf(-2.0f * logf(u01_open(p.x)));
+  float theta = 6.2831853071795864f * u01(p.y);
+  float s, c;
+  __sincosf(theta, &s, &c);
+  *g0 = r * c;
+  *g1 = r * s;
+}
+
+DEV_INLINE float gauss1(uint64_t seed, uint64_t ctr) {
+  Philox4 p = philox4x32(seed, ctr);
+  float r = sqrtf(-2.0f * logf(u01_open(p.x)));
+  float theta = 6.2831853071795864f * u01(p.y);
+  return r * __cosf(theta);
+}
+
+// uniform in [-a, a]
+DEV_INLINE float uniform_pm(uint64_t seed, uint64_t ctr, float a) {
+  Philox4 p = philox4x32(seed, ctr);
+  return (2.0f * u01(p.x) - 1.0f) * a;
+}
+
+// ---------------------------------------------------------------------------
+// wave/block reductions
+// ---------------------------------------------------------------------------
+DEV_INLINE float wave_sum(float v) {
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
+  return v;
+}
+
+DEV_INLINE float wave_max(float v) {
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_down(v, off, WAVE));
+  return v;
+}
+
+// block reduce using LDS; blockDim.x <= 1024, result valid on thread 0
+DEV_INLINE float block_sum(float v, float* lds /* >= blockDim/WAVE floats */) {
+  int lane = threadIdx.x & (WAVE - 1);
+  int wid = threadIdx.x / WAVE;
+  v = wave_sum(v);
+  if (lane == 0) lds[wid] = v;
+  __syncthreads();
+  int nw = blockDim.x / WAVE;
+  v = (threadIdx.x < nw) ? lds[threadIdx.x] : 0.0f;
+  if (wid == 0) v = wave_sum(v);
+  return v;
+}
+
+#define HIP_CHECK_LAST()                                                   \
+  do {                                                                     \
+    hipError_t e = hipGetLastError();                                      \
+    TORCH_CHECK(e == hipSuccess, "HIP kernel launch failed: ",             \
+                hipGetErrorString(e));                                     \
+  } while (0)
+
+// dispatch over {float, bf16, fp16} only (no double on the GPU path)
+#define NN_DISPATCH(TYPE, NAME, ...)                       \
+  AT_DISPATCH_SWITCH(TYPE, NAME,                           \
+    AT_DISPATCH_CASE(at::kFloat, __VA_ARGS__)              \
+    AT_DISPATCH_CASE(at::kBFloat16, __VA_ARGS__)           \
+    AT_DISPATCH_CASE(at::kHalf, __VA_ARGS__))

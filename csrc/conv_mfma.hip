@@ -1,0 +1,832 @@
+// MFMA implicit-GEMM convolution kernels for gfx950 (CDNA4), NHWC layout.
+//
+// One kernel family covers conv2d forward (optionally FUSED with the analog
+// noise model: a second "sigma" accumulator fed by |W| or |W|^2+|W| tiles
+// derived IN-REGISTER from the raw-weight tile, plus in-kernel Philox
+// Gaussian sampling -- the reference pays a whole second cuDNN conv +
+// curand pass for this, hardware_model.py:49-83), dgrad and wgrad.
+// Linear layers are the R=S=1, H=W=1 case (host wrappers reshape).
+//
+// Structure (v1, correctness-first; tuned against rocprof after):
+//   * block = 256 threads = 4 waves in a 2x2 wave grid
+//   * tile = 64(M) x 64(N), K-step 32, v_mfma_f32_16x16x32_bf16
+//   * K-loop runs over filter taps (r,s) outer, input-channel slices inner,
+//     so NHWC staging loads are channel-contiguous (vectorizable) and there
+//     is no im2col materialization
+//   * LDS tiles use an 80-byte row stride (64 B data + 16 B skew) so
+//     ds_read_b128 fragment reads avoid the row-power-of-2 bank pattern
+//   * fp32 path uses v_mfma_f32_16x16x4_f32 (exact f32, for unit tests)
+//
+// A/B/C fragment maps for v_mfma_f32_16x16x32_bf16 (cdna_hip_programming.md
+// §3): A[i][k]: i = lane&15, k = 8*(lane>>4)+j (j=0..7, 8 bf16 = 4 VGPRs);
+// B[k][j]: j = lane&15, same k split; C/D: col = lane&15,
+// row = 4*(lane>>4) + reg.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int BM = 64;
+constexpr int BN = 64;
+constexpr int BK = 32;
+constexpr int LDS_STRIDE = 80;  // bytes per tile row: 64B data + 16B skew
+
+using bf16 = __hip_bfloat16;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+DEV_INLINE float atomic_max_f32(float* addr, float val) {
+  // monotone int mapping for IEEE floats
+  int* ia = (int*)addr;
+  int old = __float_as_int(val);
+  if (val >= 0.0f) {
+    return __int_as_float(atomicMax(ia, old));
+  }
+  return __uint_as_float(atomicMin((unsigned int*)ia, (unsigned int)old));
+}
+
+struct ConvGeom {
+  int N, H, W, C;      // input
+  int K, R, S;         // filter
+  int OH, OW;          // output
+  int stride, pad;
+  int64_t M;           // = N*OH*OW rows of the implicit GEMM
+};
+
+// --------------------------------------------------------------------------
+// Staging helpers: global (NHWC) -> LDS tile [rows][BK], bf16, skewed rows.
+// Each of the 256 threads owns 8 consecutive k-elements of one row:
+//   row = tid >> 2, seg = tid & 3  (4 segs * 8 el * 2B = 64B per row)
+// --------------------------------------------------------------------------
+
+DEV_INLINE void* lds_at(char* base, int row, int byte_in_row) {
+  return base + row * LDS_STRIDE + byte_in_row;
+}
+
+// stage activation tile for tap (r,s): rows are output pixels m0+row,
+// cols are input channels ck+seg*8 .. +8
+template <typename T>
+DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
+                            const ConvGeom g, int64_t m0, int ck, int r, int s) {
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int64_t m = m0 + row;
+  bf16 vals[8];
+  if (m < g.M) {
+    int64_t t = m;
+    int ow = (int)(t % g.OW); t /= g.OW;
+    int oh = (int)(t % g.OH); t /= g.OH;
+    int n = (int)t;
+    int ih = oh * g.stride - g.pad + r;
+    int iw = ow * g.stride - g.pad + s;
+    if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
+      const T* px = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
+      int c0 = ck + seg * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int c = c0 + j;
+        vals[j] = (c < g.C) ? __float2bfloat16(to_f32(px[c]))
+                            : __float2bfloat16(0.0f);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+  }
+  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+}
+
+// stage weight tile for tap (r,s): rows are output channels n0+row,
+// cols are input channels. w layout: [K, R, S, C] (NHWC filter).
+template <typename T, bool ABS_TRANSFORM, int SIGMA_MODE>
+DEV_INLINE void stage_w_tap(char* lds, const T* __restrict__ w,
+                            const ConvGeom g, int n0, int ck, int r, int s) {
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int k = n0 + row;
+  bf16 vals[8];
+  if (k < g.K) {
+    const T* pw = w + (((int64_t)k * g.R + r) * g.S + s) * g.C;
+    int c0 = ck + seg * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float v = (c < g.C) ? to_f32(pw[c]) : 0.0f;
+      if (ABS_TRANSFORM) {
+        v = fabsf(v);
+        if (SIGMA_MODE == 2) v = v * v + v;
+      }
+      vals[j] = __float2bfloat16(v);
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+  }
+  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+}
+
+// read one 16x16x32 fragment (8 bf16) for this lane from an LDS tile
+DEV_INLINE bf16x8 frag_from_lds(char* lds, int row_base) {
+  int lane = threadIdx.x & (WAVE - 1);
+  int row = row_base + (lane & 15);
+  int byte = (lane >> 4) * 16;
+  return *(bf16x8*)lds_at(lds, row, byte);
+}
+
+// --------------------------------------------------------------------------
+// Forward kernel.
+// WANT_Y:     accumulate the main output (conv with wq)
+// SIGMA_MODE: 0 none, 1 abs(|w|), 2 abs2(|w|^2+|w|)  (uses w_raw tiles)
+// TELEM:      also accumulate sum(sigma_abs), sum|noise|, max(y_clean)
+//             (when SIGMA_MODE==2 an extra |w| accumulator is carried)
+// BIAS:       add bias[n]
+// --------------------------------------------------------------------------
+
+template <typename T, bool WANT_Y, int SIGMA_MODE, bool TELEM, bool BIAS>
+__global__ __launch_bounds__(kBlock)
+void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
+                     const T* __restrict__ wraw, const float* __restrict__ bias,
+                     T* __restrict__ out, ConvGeom g, float factor,
+                     uint64_t seed, float* __restrict__ telem /* [3] */) {
+  // grid: x = n-tiles, y = m-tiles
+  int n0 = blockIdx.x * BN;
+  int64_t m0 = (int64_t)blockIdx.y * BM;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* a_lds = smem;                                   // BM rows
+  char* b_lds = smem + BM * LDS_STRIDE;                 // BN rows (wq or wraw)
+  char* c_lds = smem + (BM + BN) * LDS_STRIDE;          // BN rows (wraw sigma)
+  char* d_lds = smem + (BM + 2 * BN) * LDS_STRIDE;      // BN rows (|w| telem)
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+
+  f32x4 acc[2][2] = {};     // y
+  f32x4 sacc[2][2] = {};    // sigma (mode 1 or 2)
+  f32x4 tacc[2][2] = {};    // sigma_abs for telemetry when mode==2
+
+  for (int r = 0; r < g.R; ++r) {
+    for (int s = 0; s < g.S; ++s) {
+      for (int ck = 0; ck < g.C; ck += BK) {
+        stage_x_tap(a_lds, x, g, m0, ck, r, s);
+        if (WANT_Y) stage_w_tap<T, false, 0>(b_lds, wq, g, n0, ck, r, s);
+        if (SIGMA_MODE > 0)
+          stage_w_tap<T, true, SIGMA_MODE>(c_lds, wraw, g, n0, ck, r, s);
+        if (TELEM && SIGMA_MODE == 2)
+          stage_w_tap<T, true, 1>(d_lds, wraw, g, n0, ck, r, s);
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm) {
+          bf16x8 a = frag_from_lds(a_lds, wm * 32 + fm * 16);
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn) {
+            if (WANT_Y) {
+              bf16x8 b = frag_from_lds(b_lds, wn * 32 + fn * 16);
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, b, acc[fm][fn], 0, 0, 0);
+            }
+            if (SIGMA_MODE > 0) {
+              bf16x8 bs = frag_from_lds(c_lds, wn * 32 + fn * 16);
+              sacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, bs, sacc[fm][fn], 0, 0, 0);
+            }
+            if (TELEM && SIGMA_MODE == 2) {
+              bf16x8 bt = frag_from_lds(d_lds, wn * 32 + fn * 16);
+              tacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, bt, tacc[fm][fn], 0, 0, 0);
+            }
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+
+  // epilogue: bias, noise, stores (+ telemetry reductions)
+  float t_sum_sigma = 0.0f, t_sum_noise = 0.0f, t_max_y = -INFINITY;
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int64_t m = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
+        int n = n0 + wn * 32 + fn * 16 + (lane & 15);
+        if (m < g.M && n < g.K) {
+          float y = WANT_Y ? acc[fm][fn][reg] : 0.0f;
+          if (BIAS) y += bias[n];
+          float v = y;
+          if (SIGMA_MODE > 0) {
+            float sig = fmaxf(sacc[fm][fn][reg], 0.0f);
+            float noise = gauss1(seed, (uint64_t)(m * g.K + n))
+                          * sqrtf(factor * sig);
+            v = y + noise;
+            if (TELEM) {
+              t_sum_noise += fabsf(noise);
+              t_max_y = fmaxf(t_max_y, y);
+              t_sum_sigma += (SIGMA_MODE == 2) ? tacc[fm][fn][reg]
+                                               : sacc[fm][fn][reg];
+            }
+          }
+          out[m * g.K + n] = from_f32<T>(v);
+        }
+      }
+    }
+  }
+  if (TELEM && SIGMA_MODE > 0) {
+    __shared__ float red[4];
+    float s0 = block_sum(t_sum_sigma, red);
+    __syncthreads();
+    float s1 = block_sum(t_sum_noise, red);
+    __syncthreads();
+    // block max via wave max + lds
+    float m0v = wave_max(t_max_y);
+    __shared__ float redm[4];
+    if ((threadIdx.x & (WAVE - 1)) == 0) redm[threadIdx.x / WAVE] = m0v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      atomicAdd(&telem[0], s0);
+      atomicAdd(&telem[1], s1);
+      float mm = fmaxf(fmaxf(redm[0], redm[1]), fmaxf(redm[2], redm[3]));
+      atomic_max_f32(&telem[2], mm);
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// Dgrad: dx[n,ih,iw,c] = sum_{r,s,k} g[n,oh,ow,k] * w[k,r,s,c]
+// with oh = (ih + pad - r)/stride when divisible. Implicit GEMM over taps:
+// A rows = input pixels, contraction = output channels K, B = wt[r,s,c,k]
+// ([R*S*C, K] row-major: contraction contiguous).
+// --------------------------------------------------------------------------
+
+template <typename T>
+DEV_INLINE void stage_g_tap_dgrad(char* lds, const T* __restrict__ gy,
+                                  const ConvGeom g, int64_t m0, int kk, int r,
+                                  int s) {
+  // rows = input pixels (n, ih, iw); cols = output channels kk+seg*8..
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int64_t m = m0 + row;
+  int64_t MI = (int64_t)g.N * g.H * g.W;
+  bf16 vals[8];
+  bool ok = false;
+  const T* pg = nullptr;
+  if (m < MI) {
+    int64_t t = m;
+    int iw = (int)(t % g.W); t /= g.W;
+    int ih = (int)(t % g.H); t /= g.H;
+    int n = (int)t;
+    int ohs = ih + g.pad - r;
+    int ows = iw + g.pad - s;
+    if (ohs >= 0 && ows >= 0 && ohs % g.stride == 0 && ows % g.stride == 0) {
+      int oh = ohs / g.stride, ow = ows / g.stride;
+      if (oh < g.OH && ow < g.OW) {
+        pg = gy + (((int64_t)n * g.OH + oh) * g.OW + ow) * g.K;
+        ok = true;
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int k = kk + seg * 8 + j;
+    vals[j] = (ok && k < g.K) ? __float2bfloat16(to_f32(pg[k]))
+                              : __float2bfloat16(0.0f);
+  }
+  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+}
+
+template <typename T>
+DEV_INLINE void stage_wt_tap(char* lds, const T* __restrict__ wt,
+                             const ConvGeom g, int n0, int kk, int r, int s) {
+  // wt layout [R, S, C, K]; rows = input channels n0+row, cols = K
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int c = n0 + row;
+  bf16 vals[8];
+  if (c < g.C) {
+    const T* pw = wt + (((int64_t)r * g.S + s) * g.C + c) * g.K;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = kk + seg * 8 + j;
+      vals[j] = (k < g.K) ? __float2bfloat16(to_f32(pw[k]))
+                          : __float2bfloat16(0.0f);
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+  }
+  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock)
+void conv_dgrad_kernel(const T* __restrict__ gy, const T* __restrict__ wt,
+                       T* __restrict__ dx, ConvGeom g) {
+  int n0 = blockIdx.x * BN;  // over input channels C
+  int64_t m0 = (int64_t)blockIdx.y * BM;  // over input pixels
+  int64_t MI = (int64_t)g.N * g.H * g.W;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* a_lds = smem;
+  char* b_lds = smem + BM * LDS_STRIDE;
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+  f32x4 acc[2][2] = {};
+
+  for (int r = 0; r < g.R; ++r) {
+    for (int s = 0; s < g.S; ++s) {
+      for (int kk = 0; kk < g.K; kk += BK) {
+        stage_g_tap_dgrad(a_lds, gy, g, m0, kk, r, s);
+        stage_wt_tap(b_lds, wt, g, n0, kk, r, s);
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm) {
+          bf16x8 a = frag_from_lds(a_lds, wm * 32 + fm * 16);
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn) {
+            bf16x8 b = frag_from_lds(b_lds, wn * 32 + fn * 16);
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[fm][fn], 0, 0, 0);
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int64_t m = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
+        int c = n0 + wn * 32 + fn * 16 + (lane & 15);
+        if (m < MI && c < g.C) dx[m * g.C + c] = from_f32<T>(acc[fm][fn][reg]);
+      }
+}
+
+// --------------------------------------------------------------------------
+// Wgrad: dw[k,r,s,c] = sum_{n,oh,ow} g[n,oh,ow,k] * x[n,ih,iw,c].
+// Contraction over output pixels (huge): each block owns (tap, k-tile,
+// c-tile) and a grid-z slice of the M range; partials atomicAdd into an
+// f32 buffer. Staging transposes G and X into LDS so fragments read
+// contraction-contiguous.
+// --------------------------------------------------------------------------
+
+template <typename T>
+DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
+                                    const T* __restrict__ gy,
+                                    const T* __restrict__ x, const ConvGeom g,
+                                    int64_t m0, int k0, int c0, int r, int s) {
+  // 256 threads load a [BK=32 m] x [64 col] slab of G and X each, writing
+  // transposed into LDS rows [col][m]. Thread: mi = tid&31, colseg = tid>>5
+  // (8 segs of 8 cols).
+  int mi = threadIdx.x & 31;
+  int colseg = threadIdx.x >> 5;
+  int64_t m = m0 + mi;
+  bool mok = m < g.M;
+  int n = 0, ih = 0, iw = 0;
+  const T* px = nullptr;
+  const T* pg = nullptr;
+  if (mok) {
+    int64_t t = m;
+    int ow = (int)(t % g.OW); t /= g.OW;
+    int oh = (int)(t % g.OH); t /= g.OH;
+    n = (int)t;
+    ih = oh * g.stride - g.pad + r;
+    iw = ow * g.stride - g.pad + s;
+    pg = gy + (((int64_t)n * g.OH + oh) * g.OW + ow) * g.K;
+    if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
+      px = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int kcol = k0 + colseg * 8 + j;
+    float gv = (mok && kcol < g.K) ? to_f32(pg[kcol]) : 0.0f;
+    *(bf16*)lds_at(g_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(gv);
+    int ccol = c0 + colseg * 8 + j;
+    float xv = (px != nullptr && ccol < g.C) ? to_f32(px[ccol]) : 0.0f;
+    *(bf16*)lds_at(x_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(xv);
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock)
+void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
+                       float* __restrict__ dw /* [K,R,S,C] f32 */, ConvGeom g,
+                       int mchunks_per_block) {
+  // grid: x = c-tiles, y = k-tiles, z = taps * m-slices
+  int tap = blockIdx.z % (g.R * g.S);
+  int mslice = blockIdx.z / (g.R * g.S);
+  int r = tap / g.S, s = tap % g.S;
+  int c0 = blockIdx.x * BN;
+  int k0 = blockIdx.y * BM;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* g_lds = smem;                        // [64 k rows][32 m]
+  char* x_lds = smem + BM * LDS_STRIDE;      // [64 c rows][32 m]
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+  f32x4 acc[2][2] = {};
+
+  int64_t m_start = (int64_t)mslice * mchunks_per_block * BK;
+  int64_t m_end = m_start + (int64_t)mchunks_per_block * BK;
+  if (m_end > g.M) m_end = g.M;
+
+  for (int64_t m0 = m_start; m0 < m_end; m0 += BK) {
+    stage_gx_transposed(g_lds, x_lds, gy, x, g, m0, k0, c0, r, s);
+    __syncthreads();
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+      bf16x8 a = frag_from_lds(g_lds, wm * 32 + fm * 16);  // A: [k][m]
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+        bf16x8 b = frag_from_lds(x_lds, wn * 32 + fn * 16);  // B: [c][m]
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, acc[fm][fn], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int k = k0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
+        int c = c0 + wn * 32 + fn * 16 + (lane & 15);
+        if (k < g.K && c < g.C) {
+          float v = acc[fm][fn][reg];
+          if (v != 0.0f || gridDim.z > (unsigned)(g.R * g.S))
+            atomicAdd(&dw[(((int64_t)k * g.R + r) * g.S + s) * g.C + c], v);
+        }
+      }
+}
+
+template <typename scalar_t> struct DevT { using type = scalar_t; };
+template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
+template <> struct DevT<at::Half> { using type = _Float16; };
+
+ConvGeom make_geom(int N, int H, int W, int C, int K, int R, int S, int stride,
+                   int pad) {
+  ConvGeom g;
+  g.N = N; g.H = H; g.W = W; g.C = C; g.K = K; g.R = R; g.S = S;
+  g.stride = stride; g.pad = pad;
+  g.OH = (H + 2 * pad - R) / stride + 1;
+  g.OW = (W + 2 * pad - S) / stride + 1;
+  g.M = (int64_t)N * g.OH * g.OW;
+  return g;
+}
+
+// NHWC raw pointer views of channels_last tensors
+inline void check_cl(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.dim() == 4 && t.is_contiguous(at::MemoryFormat::ChannelsLast),
+              name, ": expected 4-D channels_last tensor");
+}
+
+}  // namespace
+
+// ===========================================================================
+// host wrappers
+// ===========================================================================
+
+torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
+                       int64_t pad) {
+  check_cl(x, "conv_fwd x");
+  check_cl(w, "conv_fwd w");
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)w.size(0), (int)w.size(2),
+                     (int)w.size(3), (int)stride, (int)pad);
+  auto out = torch::empty({g.N, g.K, g.OH, g.OW},
+                          x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
+  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  NN_DISPATCH(x.scalar_type(),
+                                  "conv_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((conv_fwd_kernel<T, true, 0, false, false>), grid,
+                       dim3(kBlock), lds, c10::hip::getCurrentHIPStream(),
+                       (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+                       (const T*)w.data_ptr(), nullptr, (T*)out.data_ptr(), g,
+                       0.0f, 0, nullptr);
+  });
+  HIP_CHECK_LAST();
+  return out;
+}
+
+std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
+                                               torch::Tensor wq,
+                                               torch::Tensor wraw,
+                                               torch::Tensor bias,
+                                               int64_t stride, int64_t pad,
+                                               int64_t sigma_mode,
+                                               double factor, int64_t seed,
+                                               bool telem, bool want_y) {
+  check_cl(x, "conv_fwd_fused x");
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
+                     (int)wraw.size(3), (int)stride, (int)pad);
+  auto out = torch::empty({g.N, g.K, g.OH, g.OW},
+                          x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  bool has_bias = bias.numel() > 0;
+  torch::Tensor bias_f;
+  if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
+  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+  tele[2] = -std::numeric_limits<float>::infinity();
+  dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
+  size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
+  NN_DISPATCH(x.scalar_type(),
+                                  "conv_fwd_fused", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto stream = c10::hip::getCurrentHIPStream();
+    const T* xp = (const T*)x.data_ptr();
+    const T* wqp = (const T*)wq.data_ptr();
+    const T* wrp = (const T*)wraw.data_ptr();
+    const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
+    T* op = (T*)out.data_ptr();
+    float* tp = tele.data_ptr<float>();
+    float f = (float)factor;
+    uint64_t sd = (uint64_t)seed;
+    auto launch = [&](auto wy, auto sm, auto tl, auto bi) {
+      hipLaunchKernelGGL((conv_fwd_kernel<T, decltype(wy)::value,
+                          decltype(sm)::value, decltype(tl)::value,
+                          decltype(bi)::value>), grid, dim3(kBlock), lds,
+                         stream, xp, wqp, wrp, bp, op, g, f, sd, tp);
+    };
+    using TT = std::true_type; using FF = std::false_type;
+    using S1 = std::integral_constant<int, 1>;
+    using S2 = std::integral_constant<int, 2>;
+    if (want_y) {
+      if (sigma_mode == 1) {
+        if (telem) { if (has_bias) launch(TT{}, S1{}, TT{}, TT{}); else launch(TT{}, S1{}, TT{}, FF{}); }
+        else { if (has_bias) launch(TT{}, S1{}, FF{}, TT{}); else launch(TT{}, S1{}, FF{}, FF{}); }
+      } else {
+        if (telem) { if (has_bias) launch(TT{}, S2{}, TT{}, TT{}); else launch(TT{}, S2{}, TT{}, FF{}); }
+        else { if (has_bias) launch(TT{}, S2{}, FF{}, TT{}); else launch(TT{}, S2{}, FF{}, FF{}); }
+      }
+    } else {
+      if (sigma_mode == 1) {
+        if (telem) launch(FF{}, S1{}, TT{}, FF{}); else launch(FF{}, S1{}, FF{}, FF{});
+      } else {
+        if (telem) launch(FF{}, S2{}, TT{}, FF{}); else launch(FF{}, S2{}, FF{}, FF{});
+      }
+    }
+  });
+  HIP_CHECK_LAST();
+  return {out, tele};
+}
+
+torch::Tensor conv_dgrad(torch::Tensor gy, torch::Tensor w, int64_t stride,
+                         int64_t pad, int64_t H, int64_t W) {
+  check_cl(gy, "conv_dgrad gy");
+  check_cl(w, "conv_dgrad w");
+  auto g = make_geom((int)gy.size(0), (int)H, (int)W, (int)w.size(1),
+                     (int)w.size(0), (int)w.size(2), (int)w.size(3),
+                     (int)stride, (int)pad);
+  TORCH_CHECK(g.OH == (int)gy.size(2) && g.OW == (int)gy.size(3),
+              "conv_dgrad: grad shape mismatch");
+  // wt[r,s,c,k]: contraction-contiguous weight view (tiny tensor)
+  auto wt = w.permute({2, 3, 1, 0}).contiguous();
+  auto dx = torch::empty({g.N, g.C, g.H, g.W},
+                         gy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t MI = (int64_t)g.N * g.H * g.W;
+  dim3 grid((g.C + BN - 1) / BN, (int)((MI + BM - 1) / BM));
+  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  NN_DISPATCH(gy.scalar_type(),
+                                  "conv_dgrad", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((conv_dgrad_kernel<T>), grid, dim3(kBlock), lds,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)gy.data_ptr(), (const T*)wt.data_ptr(),
+                       (T*)dx.data_ptr(), g);
+  });
+  HIP_CHECK_LAST();
+  return dx;
+}
+
+torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
+                         int64_t pad, int64_t R, int64_t S) {
+  check_cl(gy, "conv_wgrad gy");
+  check_cl(x, "conv_wgrad x");
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)gy.size(1), (int)R, (int)S,
+                     (int)stride, (int)pad);
+  auto dw_f = torch::zeros({g.K, g.R, g.S, g.C},
+                           x.options().dtype(torch::kFloat32));
+  // slice the contraction so ~1024+ blocks are in flight
+  int taps = g.R * g.S;
+  int64_t mtotal = (g.M + BK - 1) / BK;  // number of BK chunks
+  int ctiles = (g.C + BN - 1) / BN;
+  int ktiles = (g.K + BM - 1) / BM;
+  int64_t target_z = std::max<int64_t>(1, 2048 / std::max(1, ctiles * ktiles));
+  int mslices = (int)std::min<int64_t>(mtotal, std::max<int64_t>(1, target_z / taps ? target_z / taps : 1));
+  if (mslices < 1) mslices = 1;
+  int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
+  dim3 grid(ctiles, ktiles, taps * mslices);
+  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  NN_DISPATCH(gy.scalar_type(),
+                                  "conv_wgrad", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((conv_wgrad_kernel<T>), grid, dim3(kBlock), lds,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                       dw_f.data_ptr<float>(), g, chunks_per_block);
+  });
+  HIP_CHECK_LAST();
+  // dw as NCHW-logical [K, C, R, S] channels_last == raw [K,R,S,C]
+  auto dw = dw_f.to(x.scalar_type());
+  return dw.permute({0, 3, 1, 2}).contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+// ===========================================================================
+// Linear layers = the R=S=1, H=W=1 case: a 2-D row-major [B, F] tensor has
+// exactly the raw layout of an NHWC (B, F, 1, 1) channels_last tensor, so
+// the conv kernels run unchanged on raw pointers.
+// ===========================================================================
+
+namespace {
+ConvGeom linear_geom(const torch::Tensor& x, int64_t out_features) {
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous(), "linear: 2-D contiguous");
+  return make_geom((int)x.size(0), 1, 1, (int)x.size(1), (int)out_features,
+                   1, 1, 1, 0);
+}
+}  // namespace
+
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w) {
+  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
+  auto g = linear_geom(x, w.size(0));
+  auto out = torch::empty({x.size(0), w.size(0)}, x.options());
+  dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
+  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  NN_DISPATCH(x.scalar_type(),
+                                  "linear_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((conv_fwd_kernel<T, true, 0, false, false>), grid,
+                       dim3(kBlock), lds, c10::hip::getCurrentHIPStream(),
+                       (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+                       (const T*)w.data_ptr(), nullptr, (T*)out.data_ptr(), g,
+                       0.0f, 0, nullptr);
+  });
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// dx = g @ W: contraction over out_features -> conv_fwd with A=g [B,O] and
+// "weights" = W^T [I, O] (pre-transposed; W is small).
+torch::Tensor linear_dgrad(torch::Tensor gy, torch::Tensor w) {
+  auto wt = w.t().contiguous();  // [I, O]
+  return linear_fwd(gy, wt);
+}
+
+// dW = g^T @ x: the wgrad kernel with the 1x1-tap geometry.
+torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
+  TORCH_CHECK(gy.dim() == 2 && gy.is_contiguous());
+  TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
+  auto g = linear_geom(x, gy.size(1));
+  auto dw_f = torch::zeros({g.K, g.C}, x.options().dtype(torch::kFloat32));
+  int64_t mtotal = (g.M + BK - 1) / BK;
+  int ctiles = (g.C + BN - 1) / BN;
+  int ktiles = (g.K + BM - 1) / BM;
+  int64_t target_z = std::max<int64_t>(1, 1024 / std::max(1, ctiles * ktiles));
+  int mslices = (int)std::min<int64_t>(mtotal, target_z);
+  if (mslices < 1) mslices = 1;
+  int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
+  dim3 grid(ctiles, ktiles, mslices);
+  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  NN_DISPATCH(gy.scalar_type(),
+                                  "linear_wgrad", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((conv_wgrad_kernel<T>), grid, dim3(kBlock), lds,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                       dw_f.data_ptr<float>(), g, chunks_per_block);
+  });
+  HIP_CHECK_LAST();
+  return dw_f.to(x.scalar_type());
+}
+
+std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
+                                            torch::Tensor wraw,
+                                            torch::Tensor bias,
+                                            int64_t sigma_mode, double factor,
+                                            int64_t seed, bool telem) {
+  auto g = linear_geom(x, wraw.size(0));
+  auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
+  bool has_bias = bias.numel() > 0;
+  torch::Tensor bias_f;
+  if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
+  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+  tele[2] = -std::numeric_limits<float>::infinity();
+  dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
+  size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
+  NN_DISPATCH(x.scalar_type(),
+                                  "linear_fwd_fused", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto stream = c10::hip::getCurrentHIPStream();
+    const T* xp = (const T*)x.data_ptr();
+    const T* wqp = (const T*)wq.data_ptr();
+    const T* wrp = (const T*)wraw.data_ptr();
+    const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
+    T* op = (T*)out.data_ptr();
+    float* tp = tele.data_ptr<float>();
+    float f = (float)factor;
+    uint64_t sd = (uint64_t)seed;
+    auto launch = [&](auto sm, auto tl, auto bi) {
+      hipLaunchKernelGGL((conv_fwd_kernel<T, true, decltype(sm)::value,
+                          decltype(tl)::value, decltype(bi)::value>), grid,
+                         dim3(kBlock), lds, stream, xp, wqp, wrp, bp, op, g,
+                         f, sd, tp);
+    };
+    using TT = std::true_type; using FF = std::false_type;
+    using S1 = std::integral_constant<int, 1>;
+    using S2 = std::integral_constant<int, 2>;
+    if (sigma_mode == 1) {
+      if (telem) { if (has_bias) launch(S1{}, TT{}, TT{}); else launch(S1{}, TT{}, FF{}); }
+      else { if (has_bias) launch(S1{}, FF{}, TT{}); else launch(S1{}, FF{}, FF{}); }
+    } else {
+      if (telem) { if (has_bias) launch(S2{}, TT{}, TT{}); else launch(S2{}, TT{}, FF{}); }
+      else { if (has_bias) launch(S2{}, FF{}, TT{}); else launch(S2{}, FF{}, FF{}); }
+    }
+  });
+  HIP_CHECK_LAST();
+  return {out, tele};
+}
+
+std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
+                                          torch::Tensor wraw, torch::Tensor bias,
+                                          int64_t stride, int64_t pad,
+                                          int64_t sigma_mode, double factor,
+                                          int64_t seed, bool telem) {
+  return conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
+                             factor, seed, telem, /*want_y=*/true);
+}
+
+std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
+                                            int64_t stride, int64_t pad,
+                                            int64_t sigma_mode, double factor,
+                                            int64_t seed, bool telem) {
+  auto empty_bias = torch::empty({0}, x.options());
+  return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,
+                             sigma_mode, factor, seed, telem, /*want_y=*/false);
+}
+
+std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
+                                                   torch::Tensor wraw,
+                                                   int64_t sigma_mode,
+                                                   double factor, int64_t seed,
+                                                   bool telem) {
+  auto g = linear_geom(x, wraw.size(0));
+  auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
+  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+  tele[2] = -std::numeric_limits<float>::infinity();
+  dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
+  size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
+  NN_DISPATCH(x.scalar_type(),
+                                  "sigma_noise_linear", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto stream = c10::hip::getCurrentHIPStream();
+    const T* xp = (const T*)x.data_ptr();
+    const T* wrp = (const T*)wraw.data_ptr();
+    T* op = (T*)out.data_ptr();
+    float* tp = tele.data_ptr<float>();
+    float f = (float)factor;
+    uint64_t sd = (uint64_t)seed;
+    if (sigma_mode == 1) {
+      if (telem)
+        hipLaunchKernelGGL((conv_fwd_kernel<T, false, 1, true, false>), grid,
+                           dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
+                           op, g, f, sd, tp);
+      else
+        hipLaunchKernelGGL((conv_fwd_kernel<T, false, 1, false, false>), grid,
+                           dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
+                           op, g, f, sd, tp);
+    } else {
+      if (telem)
+        hipLaunchKernelGGL((conv_fwd_kernel<T, false, 2, true, false>), grid,
+                           dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
+                           op, g, f, sd, tp);
+      else
+        hipLaunchKernelGGL((conv_fwd_kernel<T, false, 2, false, false>), grid,
+                           dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
+                           op, g, f, sd, tp);
+    }
+  });
+  HIP_CHECK_LAST();
+  return {out, tele};
+}

@@ -1,0 +1,104 @@
+// MaxPool 2x2 stride 2, NHWC. Forward stores a 2-bit argmax code per output
+// (packed in uint8) so backward is a gather-free scatter.
+
+#include <torch/extension.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+template <typename T>
+__global__ void maxpool2x2_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                      uint8_t* __restrict__ code, int64_t n_out,
+                                      int C, int H, int W, int OH, int OW) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t t = i / C;
+    int ow = (int)(t % OW);
+    t /= OW;
+    int oh = (int)(t % OH);
+    int64_t nb = t / OH;
+    const T* base = x + ((nb * H + 2 * oh) * W + 2 * ow) * C + c;
+    float v00 = to_f32(base[0]);
+    float v01 = to_f32(base[C]);
+    float v10 = to_f32(base[(int64_t)W * C]);
+    float v11 = to_f32(base[(int64_t)W * C + C]);
+    float m = v00;
+    int k = 0;
+    if (v01 > m) { m = v01; k = 1; }
+    if (v10 > m) { m = v10; k = 2; }
+    if (v11 > m) { m = v11; k = 3; }
+    y[i] = from_f32<T>(m);
+    code[i] = (uint8_t)k;
+  }
+}
+
+template <typename T>
+__global__ void maxpool2x2_bwd_kernel(const T* __restrict__ g,
+                                      const uint8_t* __restrict__ code,
+                                      T* __restrict__ gx, int64_t n_out, int C,
+                                      int H, int W, int OH, int OW) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t t = i / C;
+    int ow = (int)(t % OW);
+    t /= OW;
+    int oh = (int)(t % OH);
+    int64_t nb = t / OH;
+    int k = code[i];
+    int dh = k >> 1, dw = k & 1;
+    gx[((nb * H + 2 * oh + dh) * W + 2 * ow + dw) * C + c] = g[i];
+  }
+}
+
+template <typename scalar_t> struct DevT { using type = scalar_t; };
+template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
+template <> struct DevT<at::Half> { using type = _Float16; };
+
+}  // namespace
+
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.dim() == 4 && x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2), W = (int)x.size(3);
+  int OH = H / 2, OW = W / 2;
+  auto y = torch::empty({N, C, OH, OW},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto code = torch::empty({N, OH, OW, C}, x.options().dtype(torch::kUInt8));
+  int64_t n_out = (int64_t)N * C * OH * OW;
+  int blocks = (int)std::min<int64_t>((n_out + kBlock - 1) / kBlock, 8192);
+  NN_DISPATCH(x.scalar_type(),
+                                  "maxpool2x2_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((maxpool2x2_fwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)x.data_ptr(),
+                       (T*)y.data_ptr(), code.data_ptr<uint8_t>(), n_out, C, H,
+                       W, OH, OW);
+  });
+  HIP_CHECK_LAST();
+  return {y, code};
+}
+
+torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
+                             int64_t W) {
+  TORCH_CHECK(g.dim() == 4 && g.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = (int)g.size(0), C = (int)g.size(1), OH = (int)g.size(2), OW = (int)g.size(3);
+  auto gx = torch::zeros({N, C, (int)H, (int)W},
+                         g.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t n_out = (int64_t)N * C * OH * OW;
+  int blocks = (int)std::min<int64_t>((n_out + kBlock - 1) / kBlock, 8192);
+  NN_DISPATCH(g.scalar_type(),
+                                  "maxpool2x2_bwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((maxpool2x2_bwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)g.data_ptr(),
+                       code.data_ptr<uint8_t>(), (T*)gx.data_ptr(), n_out, C,
+                       (int)H, (int)W, OH, OW);
+  });
+  HIP_CHECK_LAST();
+  return gx;
+}

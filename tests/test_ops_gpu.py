@@ -1,0 +1,402 @@
+"""GPU numerics tests: every HIP kernel vs the plain-PyTorch fp32 reference.
+
+All marked @pytest.mark.gpu (run on MI355X via gpurun / at round end).
+bf16 kernels compare against fp32 references with bf16-appropriate
+tolerances (inputs are drawn bf16-representable where exactness matters).
+"""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+from noisynet_amd import ops  # noqa: E402
+from noisynet_amd.ops import reference as ref  # noqa: E402
+
+
+def dev():
+    return torch.device("cuda")
+
+
+def cl(t):
+    return t.contiguous(memory_format=torch.channels_last)
+
+
+def to_bf16_representable(x):
+    return x.bfloat16().float()
+
+
+# ---------------------------------------------------------------------------
+# elementwise
+# ---------------------------------------------------------------------------
+
+
+def test_ext_loaded_native_path_required():
+    assert ops.has_ext(), "HIP extension must be loadable on the GPU box"
+    x = torch.randn(16, device=dev())
+    # native op must run (raises if ext missing)
+    ops.fake_quant(x, 4, 0.0, 1.0, 0.0)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fake_quant_gpu_matches_reference(dtype):
+    torch.manual_seed(0)
+    x = torch.randn(4096, device=dev(), dtype=dtype) * 2
+    out = ops.fake_quant(x, 4, 0.0, 1.0, 0.0)
+    expect = ref.fake_quant_forward(x.float().cpu(), 4, 0.0, 1.0, 0.0)
+    got = out.float().cpu()
+    # rounding boundaries can differ by one level in bf16; compare levels
+    atol = 1e-6 if dtype == torch.float32 else 1.5 / 15
+    assert torch.allclose(got, expect, atol=atol)
+    if dtype == torch.float32:
+        assert torch.equal(got, expect)
+
+
+def test_fake_quant_gpu_stochastic_unbiased():
+    x = torch.full((1 << 20,), 0.3, device=dev())
+    out = ops.fake_quant(x, 4, 0.0, 1.0, 0.5)
+    assert abs(out.mean().item() - 0.3) < 2e-3
+
+
+def test_ste_mask_gpu():
+    x = torch.tensor([-2.0, 0.5, 1.5], device=dev())
+    g = torch.ones_like(x)
+    out = ops.ext().ste_mask(g, x, 0.0, 1.0)
+    assert out.cpu().tolist() == [0.0, 1.0, 0.0]
+
+
+def test_mult_uniform_noise_gpu_bounds():
+    x = torch.randn(100000, device=dev())
+    out = ops.ext().mult_uniform_noise(x, 0.1, 1234)
+    d = (out - x).abs()
+    assert (d <= 0.1 * x.abs() + 1e-5).all()
+    # roughly uniform: mean |d| ~ 0.05 * |x|
+    ratio = (d / x.abs().clamp_min(1e-8)).mean().item()
+    assert 0.045 < ratio < 0.055
+
+
+def test_dropout_gpu_stats():
+    x = torch.ones(1 << 20, device=dev())
+    y, mask = ops.ext().dropout_fwd(x, 0.25, 999)
+    keep = (mask > 0).float().mean().item()
+    assert abs(keep - 0.75) < 0.01
+    assert abs(y.mean().item() - 1.0) < 0.01  # inverted dropout preserves mean
+
+
+# ---------------------------------------------------------------------------
+# conv / linear MFMA kernels
+# ---------------------------------------------------------------------------
+
+
+CONV_SHAPES = [
+    # N, C, H, W, K, R, stride, pad      -- NoisyNet + ResNet shape classes
+    (4, 3, 32, 32, 65, 5, 1, 0),
+    (4, 65, 14, 14, 120, 5, 1, 0),
+    (2, 16, 16, 16, 32, 3, 1, 1),
+    (2, 8, 16, 16, 16, 3, 2, 1),
+    (2, 64, 7, 7, 128, 1, 1, 0),
+    (3, 33, 9, 9, 17, 3, 1, 1),
+]
+
+
+@pytest.mark.parametrize("shape", CONV_SHAPES)
+def test_conv_fwd_matches_torch(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(0)
+    x = to_bf16_representable(torch.randn(N, C, H, W)).to(dev()).bfloat16()
+    w = to_bf16_representable(torch.randn(K, C, R, R) * 0.2).to(dev()).bfloat16()
+    out = ops.ext().conv_fwd(cl(x), cl(w), stride, pad)
+    expect = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad)
+    got = out.float().cpu()
+    err = (got - expect).abs().max().item()
+    scale = expect.abs().max().item() + 1e-6
+    assert err / scale < 0.02, f"rel err {err/scale}"
+
+
+@pytest.mark.parametrize("shape", CONV_SHAPES)
+def test_conv_dgrad_matches_torch(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(1)
+    OH = (H + 2 * pad - R) // stride + 1
+    g = to_bf16_representable(torch.randn(N, K, OH, OH)).to(dev()).bfloat16()
+    w = to_bf16_representable(torch.randn(K, C, R, R) * 0.2).to(dev()).bfloat16()
+    dx = ops.ext().conv_dgrad(cl(g), cl(w), stride, pad, H, W)
+    expect = torch.nn.grad.conv2d_input((N, C, H, W), w.float().cpu(),
+                                        g.float().cpu(), stride, pad)
+    got = dx.float().cpu()
+    err = (got - expect).abs().max().item()
+    scale = expect.abs().max().item() + 1e-6
+    assert err / scale < 0.02, f"rel err {err/scale}"
+
+
+@pytest.mark.parametrize("shape", CONV_SHAPES)
+def test_conv_wgrad_matches_torch(shape):
+    N, C, H, W, K, R, stride, pad = shape
+    torch.manual_seed(2)
+    OH = (H + 2 * pad - R) // stride + 1
+    g = to_bf16_representable(torch.randn(N, K, OH, OH) * 0.1).to(dev()).bfloat16()
+    x = to_bf16_representable(torch.randn(N, C, H, W)).to(dev()).bfloat16()
+    dw = ops.ext().conv_wgrad(cl(g), cl(x), stride, pad, R, R)
+    expect = torch.nn.grad.conv2d_weight(x.float().cpu(), (K, C, R, R),
+                                         g.float().cpu(), stride, pad)
+    got = dw.float().cpu()
+    err = (got - expect).abs().max().item()
+    scale = expect.abs().max().item() + 1e-6
+    assert err / scale < 0.03, f"rel err {err/scale}"
+
+
+def test_linear_fwd_dgrad_wgrad():
+    torch.manual_seed(3)
+    B, I, O = 128, 3000, 390
+    x = to_bf16_representable(torch.randn(B, I) * 0.5).to(dev()).bfloat16()
+    w = to_bf16_representable(torch.randn(O, I) * 0.05).to(dev()).bfloat16()
+    g = to_bf16_representable(torch.randn(B, O) * 0.1).to(dev()).bfloat16()
+
+    y = ops.ext().linear_fwd(x, w)
+    y_ref = x.float().cpu() @ w.float().cpu().t()
+    rel = (y.float().cpu() - y_ref).abs().max() / (y_ref.abs().max() + 1e-6)
+    assert rel < 0.02, rel
+
+    dx = ops.ext().linear_dgrad(g, w)
+    dx_ref = g.float().cpu() @ w.float().cpu()
+    rel = (dx.float().cpu() - dx_ref).abs().max() / (dx_ref.abs().max() + 1e-6)
+    assert rel < 0.02, rel
+
+    dw = ops.ext().linear_wgrad(g, x)
+    dw_ref = g.float().cpu().t() @ x.float().cpu()
+    rel = (dw.float().cpu() - dw_ref).abs().max() / (dw_ref.abs().max() + 1e-6)
+    assert rel < 0.03, rel
+
+
+def test_conv_asymmetric_transpose_detection():
+    """A=I with asymmetric W catches transposed output layouts (§5.4 r16)."""
+    C = 16
+    x = torch.zeros(1, C, 4, 4, device=dev()).bfloat16()
+    # delta input at one pixel
+    x[0, :, 1, 2] = torch.arange(C, device=dev()).bfloat16() / C
+    w = torch.zeros(8, C, 1, 1, device=dev()).bfloat16()
+    for k in range(8):
+        w[k, (3 * k + 1) % C, 0, 0] = 1.0 + 0.125 * k
+    out = ops.ext().conv_fwd(cl(x), cl(w), 1, 0)
+    expect = F.conv2d(x.float().cpu(), w.float().cpu())
+    assert torch.allclose(out.float().cpu(), expect, atol=1e-2)
+
+
+def test_fused_conv_noise_statistics():
+    """The fused kernel's noise must be N(0, sqrt(factor * conv(x, |w|)))."""
+    torch.manual_seed(4)
+    N, C, H, W, K, R = 64, 16, 12, 12, 32, 3
+    x = torch.rand(N, C, H, W).to(dev()).bfloat16()
+    w = (torch.randn(K, C, R, R) * 0.2).to(dev()).bfloat16()
+    wq = w.clone()
+    factor = 0.05
+    out, tele = ops.ext().conv_fwd_fused(
+        cl(x), cl(wq), cl(w), torch.empty(0, device=dev(), dtype=torch.bfloat16),
+        1, 0, 1, factor, 42, True)
+    clean = F.conv2d(x.float().cpu(), w.float().cpu())
+    noise = out.float().cpu() - clean
+    sig = F.conv2d(x.float().cpu(), w.float().cpu().abs())
+    # standardize: noise / sqrt(factor*sig) ~ N(0,1)
+    z = noise / (factor * sig).clamp_min(1e-9).sqrt()
+    assert abs(z.mean().item()) < 0.02
+    assert abs(z.std().item() - 1.0) < 0.05
+    # telemetry: sum sigma_abs
+    assert abs(tele[0].item() - sig.sum().item()) / sig.sum().item() < 0.02
+    # max of clean y
+    assert abs(tele[2].item() - clean.max().item()) < 0.1
+
+
+def test_fused_conv_abs2_mode():
+    torch.manual_seed(5)
+    N, C, H, W, K, R = 32, 8, 10, 10, 16, 3
+    x = torch.rand(N, C, H, W).to(dev()).bfloat16()
+    w = (torch.randn(K, C, R, R) * 0.3).to(dev()).bfloat16()
+    factor = 0.02
+    out, tele = ops.ext().conv_fwd_fused(
+        cl(x), cl(w), cl(w), torch.empty(0, device=dev(), dtype=torch.bfloat16),
+        1, 0, 2, factor, 7, True)
+    clean = F.conv2d(x.float().cpu(), w.float().cpu())
+    noise = out.float().cpu() - clean
+    aw = w.float().cpu().abs()
+    sig2 = F.conv2d(x.float().cpu(), aw * aw + aw)
+    z = noise / (factor * sig2).clamp_min(1e-9).sqrt()
+    assert abs(z.std().item() - 1.0) < 0.05
+    # telemetry sigma_abs uses |w| even in abs2 mode
+    sig1 = F.conv2d(x.float().cpu(), aw)
+    assert abs(tele[0].item() - sig1.sum().item()) / sig1.sum().item() < 0.02
+
+
+def test_sigma_noise_only():
+    torch.manual_seed(6)
+    x = torch.rand(32, 8, 10, 10).to(dev()).bfloat16()
+    w = (torch.randn(16, 8, 3, 3) * 0.3).to(dev()).bfloat16()
+    noise, tele = ops.ext().sigma_noise_conv(cl(x), cl(w), 1, 0, 1, 0.05, 9, False)
+    aw = w.float().cpu().abs()
+    sig = F.conv2d(x.float().cpu(), aw)
+    z = noise.float().cpu() / (0.05 * sig).clamp_min(1e-9).sqrt()
+    assert abs(z.std().item() - 1.0) < 0.05
+    assert abs(z.mean().item()) < 0.02
+
+
+# ---------------------------------------------------------------------------
+# bn / pool / loss / optim
+# ---------------------------------------------------------------------------
+
+
+def test_bn_stats_and_act():
+    torch.manual_seed(7)
+    x = torch.randn(32, 65, 14, 14, device=dev()).bfloat16()
+    mean, var = ops.ext().bn_stats(cl(x))
+    xe = x.float()
+    em = xe.mean(dim=(0, 2, 3))
+    ev = xe.var(dim=(0, 2, 3), unbiased=False)
+    assert torch.allclose(mean, em, atol=2e-3, rtol=1e-2)
+    assert torch.allclose(var, ev, atol=2e-3, rtol=2e-2)
+
+    gamma = torch.randn(65, device=dev())
+    beta = torch.randn(65, device=dev())
+    invstd = (var + 1e-5).rsqrt()
+    y = ops.ext().bn_act_fwd(cl(x), mean, invstd, gamma, beta, True, 2.0)
+    ye = ref.bn_act_forward(xe, gamma, beta, mean, invstd, 2.0, True)
+    assert (y.float() - ye).abs().max().item() < 0.05
+
+
+def test_maxpool_gpu():
+    torch.manual_seed(8)
+    x = torch.randn(8, 65, 28, 28, device=dev()).bfloat16()
+    y, code = ops.ext().maxpool2x2_fwd(cl(x))
+    ye = F.max_pool2d(x.float(), 2, 2)
+    assert torch.equal(y.float().cpu(), ye.cpu())
+    g = torch.randn_like(y)
+    gx = ops.ext().maxpool2x2_bwd(cl(g), code, 28, 28)
+    # scatter sums equal
+    assert torch.allclose(gx.sum().float().cpu(), g.sum().float().cpu(), rtol=1e-2)
+
+
+def test_softmax_xent_gpu():
+    torch.manual_seed(9)
+    logits = torch.randn(512, 10, device=dev())
+    target = torch.randint(0, 10, (512,), device=dev())
+    loss, sm = ops.ext().softmax_xent_fwd(logits, target)
+    le = F.cross_entropy(logits, target)
+    assert abs(loss.item() - le.item()) < 1e-4
+    grad = ops.ext().softmax_xent_bwd(sm, target, 1.0)
+    logits2 = logits.clone().requires_grad_(True)
+    F.cross_entropy(logits2, target).backward()
+    assert torch.allclose(grad, logits2.grad, atol=1e-5)
+
+
+def test_sgd_step_gpu_matches_torch():
+    torch.manual_seed(10)
+    p0 = torch.randn(1000, device=dev())
+    g = torch.randn(1000, device=dev())
+    # native
+    p1 = p0.clone()
+    buf1 = torch.zeros_like(p1)
+    for _ in range(3):
+        ops.ext().sgd_step(p1, g, buf1, 0.1, 0.9, 1e-4, True, -0.3, 0.3)
+    # torch reference
+    p2 = p0.clone().requires_grad_(True)
+    opt = torch.optim.SGD([p2], lr=0.1, momentum=0.9, weight_decay=1e-4,
+                          nesterov=True)
+    for _ in range(3):
+        p2.grad = g.clone()
+        opt.step()
+        with torch.no_grad():
+            p2.clamp_(-0.3, 0.3)
+    assert torch.allclose(p1, p2.detach(), atol=1e-5)
+
+
+def test_adamw_step_gpu_matches_torch():
+    torch.manual_seed(11)
+    p0 = torch.randn(1000, device=dev())
+    g = torch.randn(1000, device=dev())
+    p1 = p0.clone()
+    m = torch.zeros_like(p1)
+    v = torch.zeros_like(p1)
+    for step in range(1, 4):
+        ops.ext().adamw_step(p1, g, m, v, step, 1e-3, 0.9, 0.999, 1e-8, 0.01,
+                             0.0, 0.0)
+    p2 = p0.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([p2], lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                            weight_decay=0.01)
+    for _ in range(3):
+        p2.grad = g.clone()
+        opt.step()
+    assert torch.allclose(p1, p2.detach(), atol=1e-6)
+
+
+# ---------------------------------------------------------------------------
+# end-to-end on GPU
+# ---------------------------------------------------------------------------
+
+
+def test_model_step_gpu_bf16():
+    from noisynet_amd import utils
+    from noisynet_amd.config import broadcast_per_layer, build_noisynet_parser
+    from noisynet_amd.models.noisynet import Net
+    from noisynet_amd import optim as native_optim
+    from noisynet_amd.quant import finish_calibration, start_calibration
+
+    args = build_noisynet_parser().parse_args(
+        ['--current', '1', '--q_a', '4', '--act_max', '5', '--w_max1', '0.3',
+         '--calculate_running'])
+    broadcast_per_layer(args)
+    torch.manual_seed(0)
+    model = Net(args)
+    utils.init_model(model, args)
+    model = model.cuda().bfloat16()
+    for m in model.modules():
+        if isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+            m.float()
+    model = model.to(memory_format=torch.channels_last)
+
+    x = (torch.randint(0, 16, (64, 3, 32, 32)).cuda().bfloat16() / 15.0)
+    x = cl(x)
+    y = torch.randint(0, 10, (64,)).cuda()
+
+    start_calibration(model)
+    with torch.no_grad():
+        model(x, 0, 0)
+    finish_calibration(model, 'cuda')
+
+    opt = native_optim.SGD(model.parameters(), lr=0.01, momentum=0.9,
+                           nesterov=True)
+    model.train()
+    losses = []
+    for i in range(8):
+        out = model(x, 0, 100 + i)
+        loss = F.cross_entropy(out.float(), y)
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(np.isfinite(losses))
+    # training on a fixed batch must reduce the loss
+    assert losses[-1] < losses[0], losses
+
+
+def test_model_gpu_matches_cpu_noisefree():
+    """Noise-free fp32 model: GPU forward (HIP kernels) vs CPU forward."""
+    from noisynet_amd import utils
+    from noisynet_amd.config import broadcast_per_layer, build_noisynet_parser
+    from noisynet_amd.models.noisynet import Net
+
+    args = build_noisynet_parser().parse_args(['--batch_size', '16'])
+    broadcast_per_layer(args)
+    torch.manual_seed(0)
+    model = Net(args)
+    utils.init_model(model, args)
+    model.eval()
+    x = torch.rand(16, 3, 32, 32)
+    with torch.no_grad():
+        out_cpu = model(x)
+        gm = model.cuda().to(memory_format=torch.channels_last)
+        out_gpu = gm(cl(x.cuda()))
+    err = (out_gpu.cpu() - out_cpu).abs().max().item()
+    scale = out_cpu.abs().max().item()
+    assert err / scale < 0.05, err / scale
