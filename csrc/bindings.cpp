@@ -36,6 +36,9 @@ std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
                              int64_t W);
 
+// percentile.hip
+torch::Tensor kth_percentile(torch::Tensor x, double pctl);
+
 // conv_mfma.hip
 torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                        int64_t pad);
@@ -80,6 +83,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
+  m.def("kth_percentile", &kth_percentile);
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_fwd_fused", &conv_fwd_fused);
   m.def("conv_dgrad", &conv_dgrad);

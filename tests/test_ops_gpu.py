@@ -400,3 +400,14 @@ def test_model_gpu_matches_cpu_noisefree():
     err = (out_gpu.cpu() - out_cpu).abs().max().item()
     scale = out_cpu.abs().max().item()
     assert err / scale < 0.05, err / scale
+
+
+def test_kth_percentile_gpu_matches_kthvalue():
+    torch.manual_seed(12)
+    for dtype in (torch.float32, torch.bfloat16):
+        x = (torch.randn(100000) * 3).to(dev(), dtype)
+        for pctl in (50.0, 99.0, 99.98):
+            got = ops.ext().kth_percentile(x.view(-1), pctl).float().item()
+            k = max(1, int(x.numel() * pctl / 100.0))
+            expect, _ = torch.kthvalue(x.float().cpu().flatten(), k)
+            assert got == pytest.approx(expect.item(), abs=1e-6), (dtype, pctl)
