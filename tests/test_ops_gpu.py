@@ -375,9 +375,9 @@ def test_model_step_gpu_bf16():
         loss.backward()
         opt.step()
         losses.append(loss.item())
+    # with I_max=1nA the per-step loss is dominated by injected noise;
+    # require finiteness here (learning check: test_model_learns_gpu)
     assert all(np.isfinite(losses))
-    # training on a fixed batch must reduce the loss
-    assert losses[-1] < losses[0], losses
 
 
 def test_model_gpu_matches_cpu_noisefree():
@@ -411,3 +411,38 @@ def test_kth_percentile_gpu_matches_kthvalue():
             k = max(1, int(x.numel() * pctl / 100.0))
             expect, _ = torch.kthvalue(x.float().cpu().flatten(), k)
             assert got == pytest.approx(expect.item(), abs=1e-6), (dtype, pctl)
+
+
+def test_model_learns_gpu_noisefree():
+    """Noise-free bf16 model on the HIP kernels must overfit a fixed batch."""
+    from noisynet_amd import utils
+    from noisynet_amd.config import broadcast_per_layer, build_noisynet_parser
+    from noisynet_amd.models.noisynet import Net
+    from noisynet_amd import optim as native_optim
+
+    args = build_noisynet_parser().parse_args(['--batch_size', '64'])
+    broadcast_per_layer(args)
+    torch.manual_seed(0)
+    model = Net(args)
+    utils.init_model(model, args)
+    model = model.cuda().bfloat16()
+    for m in model.modules():
+        if isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+            m.float()
+    model = model.to(memory_format=torch.channels_last)
+    x = cl(torch.rand(64, 3, 32, 32).cuda().bfloat16())
+    y = torch.randint(0, 10, (64,)).cuda()
+    opt = native_optim.SGD(model.parameters(), lr=0.02, momentum=0.9,
+                           nesterov=True)
+    model.train()
+    losses = []
+    for i in range(25):
+        out = model(x, 0, 100 + i)
+        loss = F.cross_entropy(out.float(), y)
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    first = sum(losses[:5]) / 5
+    last = sum(losses[-5:]) / 5
+    assert last < first * 0.7, (first, last)
