@@ -55,6 +55,9 @@ struct ConvGeom {
   int OH, OW;          // output
   int stride, pad;
   int64_t M;           // = N*OH*OW rows of the implicit GEMM
+  int flat;            // 1: contraction runs over flattened (r,s,c) in one
+                       // K-loop (small-C layers: conv1's C=3 would otherwise
+                       // waste 90% of each 32-deep MFMA K-step per tap)
 };
 
 // --------------------------------------------------------------------------
@@ -67,8 +70,10 @@ DEV_INLINE void* lds_at(char* base, int row, int byte_in_row) {
   return base + row * LDS_STRIDE + byte_in_row;
 }
 
-// stage activation tile for tap (r,s): rows are output pixels m0+row,
-// cols are input channels ck+seg*8 .. +8
+// stage activation tile: rows are output pixels m0+row, cols are the
+// contraction slice ck+seg*8 .. +8 (input channels of tap (r,s), or the
+// flattened (r,s,c) index when g.flat). bf16 inputs with C%8==0 take one
+// 16-byte vector load (NHWC rows are then 16B-aligned).
 template <typename T>
 DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
                             const ConvGeom g, int64_t m0, int ck, int r, int s) {
@@ -76,25 +81,51 @@ DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
   int seg = threadIdx.x & 3;
   int64_t m = m0 + row;
   bf16 vals[8];
-  if (m < g.M) {
+  bool zero = m >= g.M;
+  int n = 0, oh = 0, ow = 0;
+  if (!zero) {
     int64_t t = m;
-    int ow = (int)(t % g.OW); t /= g.OW;
-    int oh = (int)(t % g.OH); t /= g.OH;
-    int n = (int)t;
-    int ih = oh * g.stride - g.pad + r;
-    int iw = ow * g.stride - g.pad + s;
-    if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
-      const T* px = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
-      int c0 = ck + seg * 8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int c = c0 + j;
-        vals[j] = (c < g.C) ? __float2bfloat16(to_f32(px[c]))
-                            : __float2bfloat16(0.0f);
-      }
-    } else {
+    ow = (int)(t % g.OW); t /= g.OW;
+    oh = (int)(t % g.OH); t /= g.OH;
+    n = (int)t;
+  }
+  int c0 = ck + seg * 8;
+  if (g.flat) {
+    if (zero) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = c0 + j;
+        int c = k % g.C;
+        int rs = k / g.C;
+        int ss = rs % g.S;
+        int rr = rs / g.S;
+        int ih = oh * g.stride - g.pad + rr;
+        int iw = ow * g.stride - g.pad + ss;
+        float v = 0.0f;
+        if (rr < g.R && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
+          v = to_f32(x[(((int64_t)n * g.H + ih) * g.W + iw) * g.C + c]);
+        vals[j] = __float2bfloat16(v);
+      }
+    }
+    *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+    return;
+  }
+  int ih = oh * g.stride - g.pad + r;
+  int iw = ow * g.stride - g.pad + s;
+  if (!zero && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
+    const T* px = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
+    if (sizeof(T) == 2 && (g.C & 7) == 0 && c0 + 8 <= g.C) {
+      *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(px + c0);
+      return;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      vals[j] = (c < g.C) ? __float2bfloat16(to_f32(px[c]))
+                          : __float2bfloat16(0.0f);
     }
   } else {
 #pragma unroll
@@ -104,7 +135,8 @@ DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
 }
 
 // stage weight tile for tap (r,s): rows are output channels n0+row,
-// cols are input channels. w layout: [K, R, S, C] (NHWC filter).
+// cols are input channels (or the flattened (r,s,c) slice when g.flat --
+// the [K,R,S,C] filter is contiguous in exactly that order).
 template <typename T, bool ABS_TRANSFORM, int SIGMA_MODE>
 DEV_INLINE void stage_w_tap(char* lds, const T* __restrict__ w,
                             const ConvGeom g, int n0, int ck, int r, int s) {
@@ -112,13 +144,15 @@ DEV_INLINE void stage_w_tap(char* lds, const T* __restrict__ w,
   int seg = threadIdx.x & 3;
   int k = n0 + row;
   bf16 vals[8];
+  int span = g.flat ? g.R * g.S * g.C : g.C;
   if (k < g.K) {
-    const T* pw = w + (((int64_t)k * g.R + r) * g.S + s) * g.C;
+    const T* pw = g.flat ? (w + (int64_t)k * span)
+                         : (w + (((int64_t)k * g.R + r) * g.S + s) * g.C);
     int c0 = ck + seg * 8;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = c0 + j;
-      float v = (c < g.C) ? to_f32(pw[c]) : 0.0f;
+      float v = (c < span) ? to_f32(pw[c]) : 0.0f;
       if (ABS_TRANSFORM) {
         v = fabsf(v);
         if (SIGMA_MODE == 2) v = v * v + v;
@@ -173,9 +207,12 @@ void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
   f32x4 sacc[2][2] = {};    // sigma (mode 1 or 2)
   f32x4 tacc[2][2] = {};    // sigma_abs for telemetry when mode==2
 
-  for (int r = 0; r < g.R; ++r) {
-    for (int s = 0; s < g.S; ++s) {
-      for (int ck = 0; ck < g.C; ck += BK) {
+  const int Rl = g.flat ? 1 : g.R;
+  const int Sl = g.flat ? 1 : g.S;
+  const int Cl = g.flat ? g.R * g.S * g.C : g.C;
+  for (int r = 0; r < Rl; ++r) {
+    for (int s = 0; s < Sl; ++s) {
+      for (int ck = 0; ck < Cl; ck += BK) {
         stage_x_tap(a_lds, x, g, m0, ck, r, s);
         if (WANT_Y) stage_w_tap<T, false, 0>(b_lds, wq, g, n0, ck, r, s);
         if (SIGMA_MODE > 0)
@@ -295,9 +332,14 @@ DEV_INLINE void stage_g_tap_dgrad(char* lds, const T* __restrict__ gy,
       }
     }
   }
+  int k0 = kk + seg * 8;
+  if (ok && sizeof(T) == 2 && (g.K & 7) == 0 && k0 + 8 <= g.K) {
+    *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(pg + k0);
+    return;
+  }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    int k = kk + seg * 8 + j;
+    int k = k0 + j;
     vals[j] = (ok && k < g.K) ? __float2bfloat16(to_f32(pg[k]))
                               : __float2bfloat16(0.0f);
   }
@@ -314,9 +356,14 @@ DEV_INLINE void stage_wt_tap(char* lds, const T* __restrict__ wt,
   bf16 vals[8];
   if (c < g.C) {
     const T* pw = wt + (((int64_t)r * g.S + s) * g.C + c) * g.K;
+    int k0 = kk + seg * 8;
+    if (sizeof(T) == 2 && (g.K & 7) == 0 && k0 + 8 <= g.K) {
+      *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(pw + k0);
+      return;
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int k = kk + seg * 8 + j;
+      int k = k0 + j;
       vals[j] = (k < g.K) ? __float2bfloat16(to_f32(pw[k]))
                           : __float2bfloat16(0.0f);
     }
@@ -411,14 +458,60 @@ DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
     if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
       px = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
   }
+  if (g.flat) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int kcol = k0 + colseg * 8 + j;
+      float gv = (mok && kcol < g.K) ? to_f32(pg[kcol]) : 0.0f;
+      *(bf16*)lds_at(g_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(gv);
+      // flat column -> (r', s', c) decoded per element
+      int fcol = c0 + colseg * 8 + j;
+      float xv = 0.0f;
+      if (mok && fcol < g.R * g.S * g.C) {
+        int c = fcol % g.C;
+        int rs = fcol / g.C;
+        int ss = rs % g.S;
+        int rr = rs / g.S;
+        int t2 = m % ((int64_t)g.OH * g.OW);
+        int ow2 = (int)(t2 % g.OW);
+        int oh2 = (int)(t2 / g.OW);
+        int ih2 = oh2 * g.stride - g.pad + rr;
+        int iw2 = ow2 * g.stride - g.pad + ss;
+        if (ih2 >= 0 && ih2 < g.H && iw2 >= 0 && iw2 < g.W)
+          xv = to_f32(x[(((int64_t)n * g.H + ih2) * g.W + iw2) * g.C + c]);
+      }
+      *(bf16*)lds_at(x_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(xv);
+    }
+    return;
+  }
+  bf16 gvals[8], xvals[8];
+  int kc0 = k0 + colseg * 8;
+  if (mok && sizeof(T) == 2 && (g.K & 7) == 0 && kc0 + 8 <= g.K) {
+    *(bf16x8*)gvals = *(const bf16x8*)(pg + kc0);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int kcol = kc0 + j;
+      gvals[j] = (mok && kcol < g.K) ? __float2bfloat16(to_f32(pg[kcol]))
+                                     : __float2bfloat16(0.0f);
+    }
+  }
+  int cc0 = c0 + colseg * 8;
+  if (px != nullptr && sizeof(T) == 2 && (g.C & 7) == 0 && cc0 + 8 <= g.C) {
+    *(bf16x8*)xvals = *(const bf16x8*)(px + cc0);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int ccol = cc0 + j;
+      xvals[j] = (px != nullptr && ccol < g.C)
+                     ? __float2bfloat16(to_f32(px[ccol]))
+                     : __float2bfloat16(0.0f);
+    }
+  }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    int kcol = k0 + colseg * 8 + j;
-    float gv = (mok && kcol < g.K) ? to_f32(pg[kcol]) : 0.0f;
-    *(bf16*)lds_at(g_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(gv);
-    int ccol = c0 + colseg * 8 + j;
-    float xv = (px != nullptr && ccol < g.C) ? to_f32(px[ccol]) : 0.0f;
-    *(bf16*)lds_at(x_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(xv);
+    *(bf16*)lds_at(g_lds, colseg * 8 + j, mi * 2) = gvals[j];
+    *(bf16*)lds_at(x_lds, colseg * 8 + j, mi * 2) = xvals[j];
   }
 }
 
@@ -427,9 +520,10 @@ __global__ __launch_bounds__(kBlock)
 void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
                        float* __restrict__ dw /* [K,R,S,C] f32 */, ConvGeom g,
                        int mchunks_per_block) {
-  // grid: x = c-tiles, y = k-tiles, z = taps * m-slices
-  int tap = blockIdx.z % (g.R * g.S);
-  int mslice = blockIdx.z / (g.R * g.S);
+  // grid: x = c-tiles (or flat rsc-tiles), y = k-tiles, z = taps * m-slices
+  int taps = g.flat ? 1 : g.R * g.S;
+  int tap = blockIdx.z % taps;
+  int mslice = blockIdx.z / taps;
   int r = tap / g.S, s = tap % g.S;
   int c0 = blockIdx.x * BN;
   int k0 = blockIdx.y * BM;
@@ -471,10 +565,12 @@ void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
       for (int reg = 0; reg < 4; ++reg) {
         int k = k0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
         int c = c0 + wn * 32 + fn * 16 + (lane & 15);
-        if (k < g.K && c < g.C) {
-          float v = acc[fm][fn][reg];
-          if (v != 0.0f || gridDim.z > (unsigned)(g.R * g.S))
-            atomicAdd(&dw[(((int64_t)k * g.R + r) * g.S + s) * g.C + c], v);
+        int span = g.flat ? g.R * g.S * g.C : g.C;
+        if (k < g.K && c < span) {
+          int64_t off = g.flat
+              ? (int64_t)k * span + c
+              : (((int64_t)k * g.R + r) * g.S + s) * g.C + c;
+          atomicAdd(&dw[off], acc[fm][fn][reg]);
         }
       }
 }
@@ -491,6 +587,8 @@ ConvGeom make_geom(int N, int H, int W, int C, int K, int R, int S, int stride,
   g.OH = (H + 2 * pad - R) / stride + 1;
   g.OW = (W + 2 * pad - S) / stride + 1;
   g.M = (int64_t)N * g.OH * g.OW;
+  // flatten the contraction when per-tap C wastes most of a 32-deep K-step
+  g.flat = (C < 16 && R * S > 1) ? 1 : 0;
   return g;
 }
 
@@ -629,14 +727,15 @@ torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
                      (int)stride, (int)pad);
   auto dw_f = torch::zeros({g.K, g.R, g.S, g.C},
                            x.options().dtype(torch::kFloat32));
-  // slice the contraction so ~1024+ blocks are in flight
-  int taps = g.R * g.S;
+  // slice the contraction so ~2048 blocks are in flight
+  int taps = g.flat ? 1 : g.R * g.S;
+  int span = g.flat ? g.R * g.S * g.C : g.C;
   int64_t mtotal = (g.M + BK - 1) / BK;  // number of BK chunks
-  int ctiles = (g.C + BN - 1) / BN;
+  int ctiles = (span + BN - 1) / BN;
   int ktiles = (g.K + BM - 1) / BM;
   int64_t target_z = std::max<int64_t>(1, 2048 / std::max(1, ctiles * ktiles));
-  int mslices = (int)std::min<int64_t>(mtotal, std::max<int64_t>(1, target_z / taps ? target_z / taps : 1));
-  if (mslices < 1) mslices = 1;
+  int mslices = (int)std::min<int64_t>(
+      mtotal, std::max<int64_t>(1, target_z / taps));
   int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
   dim3 grid(ctiles, ktiles, taps * mslices);
   size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
