@@ -35,6 +35,14 @@ torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
 torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
                              int64_t W);
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k,
+                                       int64_t stride, int64_t pad);
+torch::Tensor maxpool_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
+                          int64_t W, int64_t k, int64_t stride, int64_t pad);
+torch::Tensor avgpool_fwd(torch::Tensor x, int64_t k, int64_t stride,
+                          int64_t pad);
+torch::Tensor avgpool_bwd(torch::Tensor g, int64_t H, int64_t W, int64_t k,
+                          int64_t stride, int64_t pad);
 
 // percentile.hip
 torch::Tensor kth_percentile(torch::Tensor x, double pctl);
@@ -83,6 +91,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("avgpool_fwd", &avgpool_fwd);
+  m.def("avgpool_bwd", &avgpool_bwd);
   m.def("kth_percentile", &kth_percentile);
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_fwd_fused", &conv_fwd_fused);

@@ -102,3 +102,206 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
   HIP_CHECK_LAST();
   return gx;
 }
+
+// ---------------------------------------------------------------------------
+// Generic pooling (kernel k x k, stride s, padding p), NHWC.
+// Max pool stores a k*k argmax code (uint8, k <= 15) for the backward
+// scatter; avg pool backward distributes g/(k*k) (count_include_pad=True,
+// matching nn.AvgPool2d defaults used by the reference ResNet).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <typename T>
+__global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   uint8_t* __restrict__ code, int64_t n_out,
+                                   int C, int H, int W, int OH, int OW, int k,
+                                   int stride, int pad) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t t = i / C;
+    int ow = (int)(t % OW);
+    t /= OW;
+    int oh = (int)(t % OH);
+    int64_t nb = t / OH;
+    float best = -INFINITY;
+    int bestk = 0;
+    for (int r = 0; r < k; ++r) {
+      int ih = oh * stride - pad + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < k; ++s) {
+        int iw = ow * stride - pad + s;
+        if (iw < 0 || iw >= W) continue;
+        float v = to_f32(x[((nb * H + ih) * W + iw) * C + c]);
+        if (v > best) { best = v; bestk = r * k + s; }
+      }
+    }
+    y[i] = from_f32<T>(best);
+    code[i] = (uint8_t)bestk;
+  }
+}
+
+template <typename T>
+__global__ void maxpool_bwd_kernel(const T* __restrict__ g,
+                                   const uint8_t* __restrict__ code,
+                                   float* __restrict__ gx, int64_t n_out, int C,
+                                   int H, int W, int OH, int OW, int k,
+                                   int stride, int pad) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t t = i / C;
+    int ow = (int)(t % OW);
+    t /= OW;
+    int oh = (int)(t % OH);
+    int64_t nb = t / OH;
+    int kk = code[i];
+    int r = kk / k, s = kk % k;
+    int ih = oh * stride - pad + r;
+    int iw = ow * stride - pad + s;
+    // overlapping windows (stride < k) need atomic accumulation
+    atomicAdd(&gx[((nb * H + ih) * W + iw) * C + c], to_f32(g[i]));
+  }
+}
+
+template <typename T>
+__global__ void avgpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int64_t n_out, int C, int H, int W, int OH,
+                                   int OW, int k, int stride, int pad) {
+  float inv = 1.0f / (k * k);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t t = i / C;
+    int ow = (int)(t % OW);
+    t /= OW;
+    int oh = (int)(t % OH);
+    int64_t nb = t / OH;
+    float acc = 0.0f;
+    for (int r = 0; r < k; ++r) {
+      int ih = oh * stride - pad + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < k; ++s) {
+        int iw = ow * stride - pad + s;
+        if (iw < 0 || iw >= W) continue;
+        acc += to_f32(x[((nb * H + ih) * W + iw) * C + c]);
+      }
+    }
+    y[i] = from_f32<T>(acc * inv);
+  }
+}
+
+template <typename T>
+__global__ void avgpool_bwd_kernel(const T* __restrict__ g,
+                                   float* __restrict__ gx, int64_t n_out, int C,
+                                   int H, int W, int OH, int OW, int k,
+                                   int stride, int pad) {
+  float inv = 1.0f / (k * k);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t t = i / C;
+    int ow = (int)(t % OW);
+    t /= OW;
+    int oh = (int)(t % OH);
+    int64_t nb = t / OH;
+    float gv = to_f32(g[i]) * inv;
+    for (int r = 0; r < k; ++r) {
+      int ih = oh * stride - pad + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < k; ++s) {
+        int iw = ow * stride - pad + s;
+        if (iw < 0 || iw >= W) continue;
+        atomicAdd(&gx[((nb * H + ih) * W + iw) * C + c], gv);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k,
+                                       int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.dim() == 4 && x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(k <= 15);
+  int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2), W = (int)x.size(3);
+  int OH = (H + 2 * (int)pad - (int)k) / (int)stride + 1;
+  int OW = (W + 2 * (int)pad - (int)k) / (int)stride + 1;
+  auto y = torch::empty({N, C, OH, OW},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto code = torch::empty({N, OH, OW, C}, x.options().dtype(torch::kUInt8));
+  int64_t n_out = (int64_t)N * C * OH * OW;
+  int blocks = (int)std::min<int64_t>((n_out + 255) / 256, 8192);
+  NN_DISPATCH(x.scalar_type(), "maxpool_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((maxpool_fwd_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)x.data_ptr(),
+                       (T*)y.data_ptr(), code.data_ptr<uint8_t>(), n_out, C, H,
+                       W, OH, OW, (int)k, (int)stride, (int)pad);
+  });
+  HIP_CHECK_LAST();
+  return {y, code};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
+                          int64_t W, int64_t k, int64_t stride, int64_t pad) {
+  TORCH_CHECK(g.dim() == 4 && g.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = (int)g.size(0), C = (int)g.size(1), OH = (int)g.size(2), OW = (int)g.size(3);
+  auto gx_f = torch::zeros({N, (int)H, (int)W, C},
+                           g.options().dtype(torch::kFloat32));
+  int64_t n_out = (int64_t)N * C * OH * OW;
+  int blocks = (int)std::min<int64_t>((n_out + 255) / 256, 8192);
+  NN_DISPATCH(g.scalar_type(), "maxpool_bwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((maxpool_bwd_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)g.data_ptr(),
+                       code.data_ptr<uint8_t>(), gx_f.data_ptr<float>(), n_out,
+                       C, (int)H, (int)W, OH, OW, (int)k, (int)stride, (int)pad);
+  });
+  HIP_CHECK_LAST();
+  // raw NHWC f32 -> logical NCHW channels_last in g's dtype
+  return gx_f.permute({0, 3, 1, 2}).to(g.scalar_type())
+      .contiguous(at::MemoryFormat::ChannelsLast);
+}
+
+torch::Tensor avgpool_fwd(torch::Tensor x, int64_t k, int64_t stride,
+                          int64_t pad) {
+  TORCH_CHECK(x.dim() == 4 && x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2), W = (int)x.size(3);
+  int OH = (H + 2 * (int)pad - (int)k) / (int)stride + 1;
+  int OW = (W + 2 * (int)pad - (int)k) / (int)stride + 1;
+  auto y = torch::empty({N, C, OH, OW},
+                        x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t n_out = (int64_t)N * C * OH * OW;
+  int blocks = (int)std::min<int64_t>((n_out + 255) / 256, 8192);
+  NN_DISPATCH(x.scalar_type(), "avgpool_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((avgpool_fwd_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)x.data_ptr(),
+                       (T*)y.data_ptr(), n_out, C, H, W, OH, OW, (int)k,
+                       (int)stride, (int)pad);
+  });
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor avgpool_bwd(torch::Tensor g, int64_t H, int64_t W, int64_t k,
+                          int64_t stride, int64_t pad) {
+  TORCH_CHECK(g.dim() == 4 && g.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = (int)g.size(0), C = (int)g.size(1), OH = (int)g.size(2), OW = (int)g.size(3);
+  auto gx_f = torch::zeros({N, (int)H, (int)W, C},
+                           g.options().dtype(torch::kFloat32));
+  int64_t n_out = (int64_t)N * C * OH * OW;
+  int blocks = (int)std::min<int64_t>((n_out + 255) / 256, 8192);
+  NN_DISPATCH(g.scalar_type(), "avgpool_bwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((avgpool_bwd_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)g.data_ptr(),
+                       gx_f.data_ptr<float>(), n_out, C, (int)H, (int)W, OH,
+                       OW, (int)k, (int)stride, (int)pad);
+  });
+  HIP_CHECK_LAST();
+  return gx_f.permute({0, 3, 1, 2}).to(g.scalar_type())
+      .contiguous(at::MemoryFormat::ChannelsLast);
+}

@@ -176,3 +176,97 @@ def var_list_for(var_name, current=0.0):
         'L2_w_max': [0.1],
     }
     return grids.get(var_name, [' '])
+
+
+# ---------------------------------------------------------------------------
+# main.py (ImageNet) parser -- reference main.py:40-192
+# ---------------------------------------------------------------------------
+
+_MAIN_BOOL_PAIRS = [
+    ('pretrained', False), ('debug_quant', False), ('normalize', False),
+    ('dali', True), ('amp', False), ('dali_cpu', True), ('merge_bn', False),
+    ('bn_out', False), ('fp16', False), ('track_running_stats', True),
+    ('plot', False), ('print_shapes', False), ('plot_basic', False),
+    ('calculate_running', False), ('q_inplace', False),
+    ('ignore_best_acc', True), ('reset_start_epoch', False),
+]
+
+
+def build_main_parser():
+    parser = argparse.ArgumentParser(
+        description='NoisyNet-MI355X ImageNet Training',
+        formatter_class=argparse.ArgumentDefaultsHelpFormatter)
+    parser.add_argument('--data', default='/data/imagenet/', metavar='DIR')
+    parser.add_argument('-a', '--arch', metavar='ARCH', default='resnet18')
+    parser.add_argument('-j', '--workers', default=10, type=int, metavar='N')
+    parser.add_argument('--epochs', default=150, type=int, metavar='N')
+    parser.add_argument('--start-epoch', default=0, type=int, metavar='N',
+                        dest='start_epoch')
+    parser.add_argument('-b', '--batch_size', '--batchsize', '--batch-size',
+                        '--bs', default=256, type=int, metavar='N')
+    parser.add_argument('--lr', '--LR', '--learning-rate', default=0.1,
+                        type=float, metavar='LR', dest='lr')
+    parser.add_argument('--gamma', type=float, default=0.1)
+    parser.add_argument('--momentum', default=0.9, type=float, metavar='M')
+    parser.add_argument('--L1', type=float, default=0.000, metavar='')
+    parser.add_argument('--wd', '--L2', '--weight-decay', default=1e-4,
+                        type=float, metavar='W', dest='weight_decay')
+    parser.add_argument('--L3', type=float, default=0.000, metavar='')
+    parser.add_argument('-p', '--print-freq', default=1000, type=int,
+                        metavar='N', dest='print_freq')
+    parser.add_argument('--resume', default='', type=str, metavar='PATH')
+    parser.add_argument('--tag', default='', type=str, metavar='PATH')
+    parser.add_argument('-e', '--evaluate', dest='evaluate', action='store_true')
+    parser.add_argument('--debug', dest='debug', action='store_true')
+    parser.add_argument('--distort_w_test', dest='distort_w_test', action='store_true')
+    parser.add_argument('--distort_act', dest='distort_act', action='store_true')
+    parser.add_argument('--distort_pre_act', dest='distort_pre_act', action='store_true')
+    parser.add_argument('--distort_act_test', dest='distort_act_test', action='store_true')
+    parser.add_argument('--noise', default=0, type=float)
+    parser.add_argument('--stochastic', default=0.5, type=float)
+    parser.add_argument('--step-after', default=30, type=int, dest='step_after')
+    parser.add_argument('--seed', default=None, type=int)
+    parser.add_argument('--num_sims', default=1, type=int)
+    parser.add_argument('--var_name', default=None, type=str)
+    parser.add_argument('--q_a', default=4, type=int)
+    parser.add_argument('--q_a_first', default=0, type=int)
+    parser.add_argument('--q_w', default=0, type=int)
+    parser.add_argument('--n_w', type=float, default=0, metavar='')
+    parser.add_argument('--n_w_test', type=float, default=0, metavar='')
+    parser.add_argument('--local_rank', '--local-rank', default=0, type=int)
+    parser.add_argument('--world_size', default=1, type=int)
+    parser.add_argument('--block_size', type=int, default=None, metavar='')
+    parser.add_argument('--act_max', default=0, type=float)
+    parser.add_argument('--w_max', default=0, type=float)
+    parser.add_argument('--eps', default=1e-7, type=float)
+    parser.add_argument('--grad_clip', default=0, type=float)
+    parser.add_argument('--q_scale', default=1, type=float)
+    parser.add_argument('--scale_bias', default=0, type=float)
+    parser.add_argument('--pctl', default=99.98, type=float)
+    parser.add_argument('--w_pctl', default=0, type=float)
+    parser.add_argument('--offset', default=0, type=float)
+    parser.add_argument('--offset_input', default=0, type=float)
+    parser.add_argument('--gpu', default=None, type=str)
+    parser.add_argument('--amp_level', default='O1', type=str)
+    parser.add_argument('--loss_scale', default=128.0, type=float)
+    parser.add_argument('--keep-batchnorm-fp32', type=str, default=None,
+                        dest='keep_batchnorm_fp32')
+    parser.add_argument('--selected_weights', type=float, default=0, metavar='')
+    parser.add_argument('--selection_criteria', type=str, default=None, metavar='')
+    parser.add_argument('--selected_weights_noise_scale', type=float, default=0, metavar='')
+    parser.add_argument('--scale_weights', type=float, default=0, metavar='')
+    parser.add_argument('--test_temp', type=float, default=0, metavar='')
+    parser.add_argument('--temperature', type=float, default=0, metavar='')
+    parser.add_argument('--debug_noise', dest='debug_noise', action='store_true')
+    parser.add_argument('--old_checkpoint', dest='old_checkpoint', action='store_true')
+    parser.add_argument('--warmup', action='store_true')
+    parser.add_argument('--lr-decay', type=str, default='step', dest='lr_decay')
+    parser.add_argument('--stuck_at_weights', type=str, default=None, metavar='')
+    parser.add_argument('--sync-bn', action='store_true', dest='sync_bn')
+    for name, default in _MAIN_BOOL_PAIRS:
+        _bool_pair(parser, name, default)
+    # MI355X additions
+    parser.add_argument('--bf16', dest='bf16', action='store_true')
+    parser.add_argument('--synthetic_batches', type=int, default=50,
+                        help='batches per epoch for the synthetic loader')
+    return parser

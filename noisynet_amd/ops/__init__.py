@@ -10,9 +10,11 @@ from .functional import (  # noqa: F401
     fake_quant,
     fused_noisy_conv2d,
     fused_noisy_linear,
+    avgpool_nhwc,
     kth_percentile,
     linear,
     maxpool2x2,
+    maxpool_nhwc,
     NoiseTelemetry,
     relu_clip,
     sgd_step,

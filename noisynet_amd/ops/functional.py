@@ -589,6 +589,68 @@ def maxpool2x2(x):
     return MaxPool2x2.apply(x)
 
 
+class MaxPoolNHWC(torch.autograd.Function):
+    """Generic k x k / stride / pad max pool (ResNet 3x3 s2 stem etc.)."""
+
+    @staticmethod
+    def forward(ctx, x, k, stride, pad):
+        if use_native(x):
+            y, code = ext().maxpool_fwd(_nhwc(x), k, stride, pad)
+        else:
+            y, code = F.max_pool2d(x, k, stride, pad, return_indices=True)
+        ctx.save_for_backward(code)
+        ctx.meta = (x.shape, k, stride, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        (code,) = ctx.saved_tensors
+        x_shape, k, stride, pad = ctx.meta
+        if use_native(g):
+            gx = ext().maxpool_bwd(_nhwc(g), code, x_shape[2], x_shape[3],
+                                   k, stride, pad)
+        else:
+            gx = F.max_unpool2d(g, code, k, stride, pad,
+                                output_size=x_shape[2:])
+        return gx, None, None, None
+
+
+def maxpool_nhwc(x, k, stride, pad=0):
+    return MaxPoolNHWC.apply(x, k, stride, pad)
+
+
+class AvgPoolNHWC(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k, stride, pad):
+        ctx.meta = (x.shape, k, stride, pad)
+        if use_native(x):
+            return ext().avgpool_fwd(_nhwc(x), k, stride, pad)
+        return F.avg_pool2d(x, k, stride, pad)
+
+    @staticmethod
+    def backward(ctx, g):
+        x_shape, k, stride, pad = ctx.meta
+        if use_native(g):
+            gx = ext().avgpool_bwd(_nhwc(g), x_shape[2], x_shape[3], k,
+                                   stride, pad)
+        else:
+            # distribute g/(k*k) over each window via conv_transpose of ones
+            w = torch.ones(x_shape[1], 1, k, k, device=g.device,
+                           dtype=g.dtype) / (k * k)
+            gx = F.conv_transpose2d(g, w, None, stride, pad,
+                                    groups=x_shape[1],
+                                    output_padding=0)
+            # pad to original size if needed
+            if gx.shape[2] != x_shape[2] or gx.shape[3] != x_shape[3]:
+                gx = F.pad(gx, (0, x_shape[3] - gx.shape[3],
+                                0, x_shape[2] - gx.shape[2]))
+        return gx, None, None, None
+
+
+def avgpool_nhwc(x, k, stride, pad=0):
+    return AvgPoolNHWC.apply(x, k, stride, pad)
+
+
 # ---------------------------------------------------------------------------
 # ReLU + clip (standalone, used where BN is off / merged)
 # ---------------------------------------------------------------------------
