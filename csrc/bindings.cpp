@@ -44,6 +44,18 @@ torch::Tensor avgpool_fwd(torch::Tensor x, int64_t k, int64_t stride,
 torch::Tensor avgpool_bwd(torch::Tensor g, int64_t H, int64_t W, int64_t k,
                           int64_t stride, int64_t pad);
 
+// depthwise.hip
+torch::Tensor dwconv_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                         int64_t stride, int64_t pad);
+torch::Tensor dwconv_dgrad(torch::Tensor gy, torch::Tensor w, int64_t stride,
+                           int64_t pad, int64_t H, int64_t W);
+torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
+                           int64_t pad, int64_t R, int64_t S);
+
+// activations.hip
+torch::Tensor act_fwd(torch::Tensor x, int64_t act);
+torch::Tensor act_bwd(torch::Tensor g, torch::Tensor x, int64_t act);
+
 // percentile.hip
 torch::Tensor kth_percentile(torch::Tensor x, double pctl);
 
@@ -91,6 +103,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
+  m.def("dwconv_fwd", &dwconv_fwd);
+  m.def("dwconv_dgrad", &dwconv_dgrad);
+  m.def("dwconv_wgrad", &dwconv_wgrad);
+  m.def("act_fwd", &act_fwd);
+  m.def("act_bwd", &act_bwd);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_fwd", &avgpool_fwd);

@@ -162,6 +162,12 @@ class NoisyConv2d(nn.Conv2d):
         weight, bias = self.effective_weight()
         if self.groups == 1 and self.dilation == (1, 1):
             return ops.conv2d(qinput, weight, bias, self.stride, self.padding)
+        if (self.groups == self.in_channels == self.out_channels
+                and self.dilation == (1, 1)):
+            return ops.depthwise_conv2d(qinput, weight, bias, self.stride,
+                                        self.padding)
+        # exotic group counts (e.g. CondConv's per-sample grouped conv) go
+        # through the library path
         return nn.functional.conv2d(qinput, weight, bias, self.stride,
                                     self.padding, self.dilation, self.groups)
 

@@ -652,6 +652,134 @@ def avgpool_nhwc(x, k, stride, pad=0):
 
 
 # ---------------------------------------------------------------------------
+# Depthwise conv (groups == channels): MobileNetV2 / EfficientNet dw layers
+# ---------------------------------------------------------------------------
+
+
+class DwConv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, stride, pad):
+        ctx.stride, ctx.pad = stride, pad
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        if use_native(x, w):
+            return ext().dwconv_fwd(
+                _nhwc(x), w.contiguous(),
+                bias if bias is not None else torch.empty(0, device=x.device,
+                                                          dtype=x.dtype),
+                stride, pad)
+        return F.conv2d(x, w, bias, stride, pad, 1, x.shape[1])
+
+    @staticmethod
+    def backward(ctx, g):
+        x, w = ctx.saved_tensors
+        gx = gw = gb = None
+        if use_native(g):
+            if ctx.needs_input_grad[0]:
+                gx = ext().dwconv_dgrad(_nhwc(g), w.contiguous(), ctx.stride,
+                                        ctx.pad, x.shape[2], x.shape[3])
+            if ctx.needs_input_grad[1]:
+                gw = ext().dwconv_wgrad(_nhwc(g), _nhwc(x), ctx.stride,
+                                        ctx.pad, w.shape[2], w.shape[3])
+        else:
+            if ctx.needs_input_grad[0]:
+                gx = torch.nn.grad.conv2d_input(x.shape, w, g, ctx.stride,
+                                                ctx.pad, 1, x.shape[1])
+            if ctx.needs_input_grad[1]:
+                gw = torch.nn.grad.conv2d_weight(x, w.shape, g, ctx.stride,
+                                                 ctx.pad, 1, x.shape[1])
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            gb = g.sum(dim=(0, 2, 3))
+        return gx, gw, gb, None, None
+
+
+def depthwise_conv2d(x, w, bias=None, stride=1, padding=0):
+    if isinstance(stride, (tuple, list)):
+        stride = stride[0]
+    if isinstance(padding, (tuple, list)):
+        padding = padding[0]
+    return DwConv2d.apply(x, w, bias, stride, padding)
+
+
+# ---------------------------------------------------------------------------
+# EfficientNet-family activations (memory-efficient analytic backward)
+# ---------------------------------------------------------------------------
+
+_ACT_IDS = {"swish": 0, "mish": 1, "hardswish": 2, "hardsigmoid": 3,
+            "sigmoid": 4}
+
+
+def _act_ref_fwd(x, act):
+    if act == "swish":
+        return x * torch.sigmoid(x)
+    if act == "mish":
+        return x * F.softplus(x).tanh()
+    if act == "hardswish":
+        return x * F.relu6(x + 3.0) / 6.0
+    if act == "hardsigmoid":
+        return F.relu6(x + 3.0) / 6.0
+    return torch.sigmoid(x)
+
+
+def _act_ref_bwd(g, x, act):
+    if act == "swish":
+        s = torch.sigmoid(x)
+        return g * (s * (1 + x * (1 - s)))
+    if act == "mish":
+        sp = F.softplus(x)
+        tsp = sp.tanh()
+        s = torch.sigmoid(x)
+        return g * (tsp + x * s * (1 - tsp * tsp))
+    if act == "hardswish":
+        d = torch.where(x <= -3.0, torch.zeros_like(x),
+                        torch.where(x >= 3.0, torch.ones_like(x),
+                                    (2 * x + 3.0) / 6.0))
+        return g * d
+    if act == "hardsigmoid":
+        d = ((x > -3.0) & (x < 3.0)).to(g.dtype) / 6.0
+        return g * d
+    s = torch.sigmoid(x)
+    return g * s * (1 - s)
+
+
+class ActFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, act):
+        ctx.act = act
+        ctx.save_for_backward(x)
+        if use_native(x):
+            return ext().act_fwd(x, _ACT_IDS[act])
+        return _act_ref_fwd(x, act)
+
+    @staticmethod
+    def backward(ctx, g):
+        (x,) = ctx.saved_tensors
+        if use_native(x):
+            return ext().act_bwd(g.contiguous(), x, _ACT_IDS[ctx.act]), None
+        return _act_ref_bwd(g, x, ctx.act), None
+
+
+def swish(x):
+    return ActFn.apply(x, "swish")
+
+
+def mish(x):
+    return ActFn.apply(x, "mish")
+
+
+def hard_swish(x):
+    return ActFn.apply(x, "hardswish")
+
+
+def hard_sigmoid(x):
+    return ActFn.apply(x, "hardsigmoid")
+
+
+def sigmoid(x):
+    return ActFn.apply(x, "sigmoid")
+
+
+# ---------------------------------------------------------------------------
 # ReLU + clip (standalone, used where BN is off / merged)
 # ---------------------------------------------------------------------------
 

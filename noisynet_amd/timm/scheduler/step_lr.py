@@ -1,0 +1,36 @@
+"""Step decay (reference timm/scheduler/step_lr.py)."""
+
+from .scheduler import Scheduler
+
+
+class StepLRScheduler(Scheduler):
+    def __init__(self, optimizer, decay_t, decay_rate=1., warmup_t=0,
+                 warmup_lr_init=0, t_in_epochs=True, noise_range_t=None,
+                 noise_pct=0.67, noise_std=1.0, noise_seed=42,
+                 initialize=True):
+        super().__init__(optimizer, 'lr', noise_range_t=noise_range_t,
+                         noise_pct=noise_pct, noise_std=noise_std,
+                         noise_seed=noise_seed, initialize=initialize)
+        self.decay_t = decay_t
+        self.decay_rate = decay_rate
+        self.warmup_t = warmup_t
+        self.warmup_lr_init = warmup_lr_init
+        self.t_in_epochs = t_in_epochs
+        if self.warmup_t:
+            self.warmup_steps = [(v - warmup_lr_init) / self.warmup_t
+                                 for v in self.base_values]
+            super().update_groups(self.warmup_lr_init)
+        else:
+            self.warmup_steps = [1 for _ in self.base_values]
+
+    def _get_lr(self, t):
+        if t < self.warmup_t:
+            return [self.warmup_lr_init + t * s for s in self.warmup_steps]
+        return [v * (self.decay_rate ** (t // self.decay_t))
+                for v in self.base_values]
+
+    def get_epoch_values(self, epoch):
+        return self._get_lr(epoch) if self.t_in_epochs else None
+
+    def get_update_values(self, num_updates):
+        return self._get_lr(num_updates) if not self.t_in_epochs else None
