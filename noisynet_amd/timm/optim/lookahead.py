@@ -3,10 +3,13 @@
 from collections import defaultdict
 
 import torch
-from torch.optim.optimizer import Optimizer
 
 
-class Lookahead(Optimizer):
+class Lookahead:
+    """Wraps a base optimizer; not an Optimizer subclass (modern torch
+    Optimizer internals require going through Optimizer.__init__, which a
+    wrapper sharing the base's param_groups cannot do)."""
+
     def __init__(self, base_optimizer, alpha=0.5, k=6):
         if not 0.0 <= alpha <= 1.0:
             raise ValueError('Invalid slow update rate: %f' % alpha)
@@ -60,3 +63,9 @@ class Lookahead(Optimizer):
             {'state': state_dict['state'],
              'param_groups': state_dict['param_groups']})
         self.param_groups = self.base_optimizer.param_groups
+
+    def zero_grad(self, set_to_none=True):
+        self.base_optimizer.zero_grad(set_to_none=set_to_none)
+
+    def add_param_group(self, param_group):
+        self.base_optimizer.add_param_group(param_group)

@@ -1,0 +1,277 @@
+"""CPU tests for the model zoo: ResNet-18, MobileNetV2, EfficientNet family,
+conv variant layers, activations, timm subsystems."""
+
+import numpy as np
+import pytest
+import torch
+
+from noisynet_amd.config import build_main_parser
+
+
+def margs(extra=None):
+    argv = ['-a', 'resnet18', '--q_a', '0']
+    if extra:
+        argv += extra
+    return build_main_parser().parse_args(argv)
+
+
+def test_resnet18_forward_backward():
+    from noisynet_amd.models.resnet import ResNet18
+    args = margs(['--q_a', '4', '--calculate_running'])
+    m = ResNet18(args)
+    m.train()
+    x = torch.rand(2, 3, 224, 224)
+    out = m(x)
+    assert out.shape == (2, 1000)
+    out.sum().backward()
+    assert m.conv1.weight.grad is not None
+    assert m.layer4[1].conv2.weight.grad is not None
+
+
+def test_resnet18_act_max_hardtanh():
+    from noisynet_amd.models.resnet import ResNet18
+    args = margs(['--act_max', '4'])
+    m = ResNet18(args)
+    m.eval()
+    x = torch.rand(2, 3, 224, 224)
+    with torch.no_grad():
+        m(x)  # runs the clip path
+
+
+def test_mobilenet_v2_forward_backward():
+    from noisynet_amd.models.mobilenet import mobilenet_v2
+    args = margs(['-a', 'mobilenet_v2', '--q_a', '4', '--calculate_running'])
+    m = mobilenet_v2(args)
+    m.train()
+    x = torch.rand(2, 3, 224, 224)
+    out = m(x)
+    assert out.shape == (2, 1000)
+    out.sum().backward()
+    assert m.features[0].conv.weight.grad is not None
+
+
+def test_efficientnet_mini_local():
+    from noisynet_amd.models.efficientnet import efficientnet_b0
+    args = margs(['-a', 'efficientnet_b0'])
+    m = efficientnet_b0(args)
+    m.train()
+    x = torch.rand(2, 3, 64, 64)
+    out = m(x)
+    assert out.shape == (2, 1000)
+    out.sum().backward()
+
+
+def test_timm_registry_and_family():
+    from noisynet_amd.timm.models import create_model, is_model, list_models
+    assert is_model('efficientnet_b0')
+    models = list_models('efficientnet*')
+    assert 'efficientnet_b0' in models and 'efficientnet_b7' in models
+    m0 = create_model('efficientnet_b0', num_classes=10)
+    m1 = create_model('efficientnet_b1', num_classes=10)
+    # b1 is deeper than b0
+    n0 = sum(1 for _ in m0.blocks.modules())
+    n1 = sum(1 for _ in m1.blocks.modules())
+    assert n1 > n0
+    out = m0(torch.rand(2, 3, 96, 96))
+    assert out.shape == (2, 10)
+
+
+def test_efficientnet_gradients_healthy_at_init():
+    """Regression: dw-conv goog init (fan_out//groups); grads bounded."""
+    from noisynet_amd.timm.models import create_model
+    torch.manual_seed(0)
+    m = create_model('efficientnet_b0', num_classes=1000)
+    m.train()
+    out = m(torch.rand(2, 3, 224, 224))
+    loss = torch.nn.functional.cross_entropy(out, torch.randint(0, 1000, (2,)))
+    loss.backward()
+    worst = max(p.grad.abs().max().item() for p in m.parameters()
+                if p.grad is not None)
+    assert worst < 1e6, worst  # was ~1e14 before the fix
+
+
+def test_efficientnet_features_backbone():
+    from noisynet_amd.timm.models.efficientnet import EfficientNetFeatures
+    m = EfficientNetFeatures(out_indices=(0, 2, 4))
+    m.eval()
+    feats = m(torch.rand(1, 3, 64, 64))
+    assert len(feats) == 3
+    assert feats[0].shape[2] > feats[1].shape[2] > feats[2].shape[2]
+
+
+def test_mobilenetv3_and_edge():
+    from noisynet_amd.timm.models import create_model
+    for name in ('mobilenetv3_large_100', 'efficientnet_es'):
+        m = create_model(name, num_classes=10)
+        m.eval()
+        out = m(torch.rand(1, 3, 64, 64))
+        assert out.shape == (1, 10)
+
+
+def test_condconv_and_mixedconv():
+    from noisynet_amd.models.conv2d_layers import (CondConv2d, MixedConv2d,
+                                                   Conv2dSame, select_conv2d)
+    x = torch.rand(3, 16, 14, 14)
+    cc = CondConv2d(16, 24, 3, padding='same', num_experts=4)
+    routing = torch.softmax(torch.rand(3, 4), dim=1)
+    out = cc(x, routing)
+    assert out.shape == (3, 24, 14, 14)
+    # per-sample experts: different routing -> different output
+    routing2 = torch.softmax(torch.rand(3, 4), dim=1)
+    out2 = cc(x, routing2)
+    assert not torch.allclose(out, out2)
+
+    mc = MixedConv2d(16, 16, [3, 5], padding='', depthwise=True)
+    assert mc(x).shape[1] == 16
+
+    cs = Conv2dSame(16, 8, 3, stride=2)
+    assert cs(x).shape == (3, 8, 7, 7)
+
+    c = select_conv2d(16, 8, 3, padding='same', stride=1)
+    assert c(x).shape == (3, 8, 14, 14)
+
+
+def test_activations_match_torch():
+    import torch.nn.functional as F
+    from noisynet_amd import ops
+    x = torch.randn(100, requires_grad=True)
+    y = ops.swish(x)
+    assert torch.allclose(y, x * torch.sigmoid(x), atol=1e-6)
+    y.sum().backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    (x2 * torch.sigmoid(x2)).sum().backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+
+    x = torch.randn(100)
+    assert torch.allclose(ops.hard_swish(x), F.hardswish(x), atol=1e-6)
+    assert torch.allclose(ops.hard_sigmoid(x), F.hardsigmoid(x), atol=1e-6)
+    assert torch.allclose(ops.mish(x), F.mish(x), atol=1e-5)
+
+
+def test_timm_optimizers():
+    from types import SimpleNamespace
+    from noisynet_amd.timm.optim import create_optimizer
+    model = torch.nn.Linear(10, 10)
+    for opt_name in ('sgd', 'adam', 'adamw', 'radam', 'nadam', 'novograd',
+                     'rmsproptf', 'lookahead_sgd'):
+        args = SimpleNamespace(opt=opt_name, lr=0.01, weight_decay=1e-4,
+                               momentum=0.9, opt_eps=1e-8)
+        opt = create_optimizer(args, model)
+        loss = model(torch.rand(4, 10)).sum()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+
+
+def test_timm_schedulers():
+    from types import SimpleNamespace
+    from noisynet_amd.timm.scheduler import create_scheduler
+    model = torch.nn.Linear(4, 4)
+    for sched in ('cosine', 'tanh', 'step', 'plateau'):
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        args = SimpleNamespace(epochs=10, sched=sched, min_lr=1e-5,
+                               decay_rate=0.1, warmup_lr=1e-4,
+                               warmup_epochs=2, cooldown_epochs=1,
+                               decay_epochs=3, patience_epochs=2)
+        s, n = create_scheduler(args, opt)
+        s.step(1, 0.5)
+        if sched == 'cosine':
+            s.step_update(5)
+
+
+def test_timm_losses():
+    from noisynet_amd.timm.loss import (LabelSmoothingCrossEntropy,
+                                        SoftTargetCrossEntropy)
+    x = torch.randn(8, 10)
+    y = torch.randint(0, 10, (8,))
+    l1 = LabelSmoothingCrossEntropy(0.1)(x, y)
+    assert torch.isfinite(l1)
+    soft = torch.softmax(torch.randn(8, 10), dim=1)
+    l2 = SoftTargetCrossEntropy()(x, soft)
+    assert torch.isfinite(l2)
+
+
+def test_timm_utils_checkpoint_saver(tmp_path):
+    from types import SimpleNamespace
+    from noisynet_amd.timm.utils import CheckpointSaver, ModelEma
+    model = torch.nn.Linear(4, 4)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    args = SimpleNamespace(model='testnet')
+    saver = CheckpointSaver(checkpoint_dir=str(tmp_path),
+                            recovery_dir=str(tmp_path), max_history=2)
+    for epoch, metric in enumerate([10.0, 30.0, 20.0]):
+        saver.save_checkpoint(model, opt, args, epoch, metric=metric)
+    import os
+    files = os.listdir(tmp_path)
+    assert 'model_best.pth.tar' in files
+    # top-2 history kept
+    assert len([f for f in files if f.startswith('checkpoint-')]) == 2
+    assert saver.best_metric == 30.0
+    saver.save_recovery(model, opt, args, 3, batch_idx=7)
+    assert saver.find_recovery()
+
+    ema = ModelEma(model, decay=0.5)
+    with torch.no_grad():
+        model.weight.add_(1.0)
+    ema.update(model)
+    assert not torch.allclose(ema.ema.weight, model.weight)
+
+
+def test_timm_mixup_and_sampler():
+    from noisynet_amd.timm.data import FastCollateMixup, mixup_target
+    from noisynet_amd.timm.data.distributed_sampler import OrderedDistributedSampler
+    t = mixup_target(torch.tensor([1, 2]), 10, lam=0.7, smoothing=0.1,
+                     device='cpu')
+    assert t.shape == (2, 10)
+    assert torch.allclose(t.sum(1), torch.ones(2))
+
+    collate = FastCollateMixup(mixup_alpha=1.0, label_smoothing=0.1,
+                               num_classes=10)
+    batch = [(np.random.randint(0, 255, (3, 8, 8), dtype=np.uint8), i % 10)
+             for i in range(4)]
+    x, y = collate(batch)
+    assert x.shape == (4, 3, 8, 8) and y.shape == (4, 10)
+
+    class DS:
+        def __len__(self):
+            return 10
+    s = OrderedDistributedSampler(DS(), num_replicas=4, rank=1)
+    assert len(list(iter(s))) == 3
+
+
+def test_timm_dataset_synthetic_loader():
+    from noisynet_amd.timm.data import Dataset, create_loader, fast_collate
+    ds = Dataset('')  # no dir -> synthetic
+    ds._synthetic.num_samples = 8
+    ds.samples = [('synthetic', 0)] * 8
+    loader = create_loader(ds, input_size=(3, 32, 32), batch_size=4,
+                           is_training=True, use_prefetcher=False,
+                           num_workers=0)
+    x, y = next(iter(loader))
+    assert x.shape[0] == 4
+
+
+def test_main_driver_efficientnet_arch(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    from noisynet_amd.drivers import imagenet
+    imagenet.main(['-a', 'efficientnet_b0', '--epochs', '1',
+                   '--batch_size', '2', '--synthetic_batches', '2',
+                   '--q_a', '0', '-p', '1'])
+
+
+def test_train_efficientnet_driver(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    from noisynet_amd.drivers import efficientnet_train
+    efficientnet_train.main(['--model', 'efficientnet_b0', '--epochs', '1',
+                             '-b', '2', '--synthetic-batches', '2',
+                             '--sched', 'step', '--cooldown-epochs', '0',
+                             '--no-prefetcher', '--workers', '0',
+                             '--output', str(tmp_path / 'out')])
+
+
+def test_chip_mnist_driver(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    from noisynet_amd.drivers import mnist
+    res = mnist.main(['--epochs', '1', '--n_train', '500', '--n_test', '100',
+                      '--L3', '0.0'])
+    assert res

@@ -446,3 +446,110 @@ def test_model_learns_gpu_noisefree():
     first = sum(losses[:5]) / 5
     last = sum(losses[-5:]) / 5
     assert last < first * 0.7, (first, last)
+
+
+def test_depthwise_conv_gpu_matches_torch():
+    torch.manual_seed(13)
+    for stride, pad, k in ((1, 1, 3), (2, 2, 5)):
+        x = to_bf16_representable(torch.randn(4, 32, 14, 14)).to(dev()).bfloat16()
+        w = to_bf16_representable(torch.randn(32, 1, k, k) * 0.2).to(dev()).bfloat16()
+        y = ops.ext().dwconv_fwd(cl(x), w.contiguous(),
+                                 torch.empty(0, device=dev(), dtype=x.dtype),
+                                 stride, pad)
+        ye = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad, 1, 32)
+        rel = (y.float().cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
+        assert rel < 0.02, rel
+        g = to_bf16_representable(torch.randn_like(ye)).to(dev()).bfloat16()
+        dx = ops.ext().dwconv_dgrad(cl(g), w.contiguous(), stride, pad, 14, 14)
+        dxe = torch.nn.grad.conv2d_input((4, 32, 14, 14), w.float().cpu(),
+                                         g.float().cpu(), stride, pad, 1, 32)
+        rel = (dx.float().cpu() - dxe).abs().max() / (dxe.abs().max() + 1e-6)
+        assert rel < 0.02, rel
+        dw = ops.ext().dwconv_wgrad(cl(g), cl(x), stride, pad, k, k)
+        dwe = torch.nn.grad.conv2d_weight(x.float().cpu(), (32, 1, k, k),
+                                          g.float().cpu(), stride, pad, 1, 32)
+        rel = (dw.float().cpu() - dwe).abs().max() / (dwe.abs().max() + 1e-6)
+        assert rel < 0.03, rel
+
+
+def test_generic_pooling_gpu():
+    torch.manual_seed(14)
+    x = torch.randn(4, 64, 56, 56, device=dev()).bfloat16()
+    y, code = ops.ext().maxpool_fwd(cl(x), 3, 2, 1)
+    ye = F.max_pool2d(x.float(), 3, 2, 1)
+    assert torch.equal(y.float().cpu(), ye.cpu())
+    g = torch.randn_like(y)
+    gx = ops.ext().maxpool_bwd(cl(g), code, 56, 56, 3, 2, 1)
+    assert torch.allclose(gx.float().sum().cpu(), g.float().sum().cpu(),
+                          rtol=1e-2)
+
+    ya = ops.ext().avgpool_fwd(cl(x), 7, 1, 0)
+    yae = F.avg_pool2d(x.float(), 7, 1)
+    assert (ya.float() - yae).abs().max().item() < 0.02
+
+
+def test_activations_gpu_match_reference():
+    torch.manual_seed(15)
+    x = torch.randn(10000, device=dev())
+    for idx, fn in ((0, lambda v: v * torch.sigmoid(v)),
+                    (2, F.hardswish), (3, F.hardsigmoid),
+                    (4, torch.sigmoid)):
+        y = ops.ext().act_fwd(x, idx)
+        assert torch.allclose(y, fn(x), atol=1e-5), idx
+    # analytic backward vs autograd (swish)
+    g = torch.randn_like(x)
+    gx = ops.ext().act_bwd(g, x, 0)
+    x2 = x.clone().requires_grad_(True)
+    (x2 * torch.sigmoid(x2)).backward(g)
+    assert torch.allclose(gx, x2.grad, atol=1e-5)
+
+
+@pytest.mark.parametrize("arch", ["resnet18", "mobilenet_v2"])
+def test_imagenet_models_step_gpu(arch):
+    from noisynet_amd.config import build_main_parser
+    from noisynet_amd import optim as native_optim
+    argv = ['-a', arch, '--q_a', '4', '--calculate_running']
+    args = build_main_parser().parse_args(argv)
+    torch.manual_seed(0)
+    if arch == 'resnet18':
+        from noisynet_amd.models.resnet import ResNet18
+        m = ResNet18(args)
+    else:
+        from noisynet_amd.models.mobilenet import mobilenet_v2
+        m = mobilenet_v2(args)
+    m = m.cuda().bfloat16()
+    for mod in m.modules():
+        if isinstance(mod, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+            mod.float()
+    m = m.to(memory_format=torch.channels_last)
+    x = cl(torch.rand(8, 3, 224, 224).cuda().bfloat16())
+    y = torch.randint(0, 1000, (8,)).cuda()
+    opt = native_optim.SGD(m.parameters(), lr=0.01, momentum=0.9)
+    m.train()
+    out = m(x)
+    loss = F.cross_entropy(out.float(), y)
+    opt.zero_grad(set_to_none=False)
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss)
+
+
+def test_efficientnet_b0_step_gpu():
+    from noisynet_amd.timm.models import create_model
+    from noisynet_amd import optim as native_optim
+    torch.manual_seed(0)
+    m = create_model('efficientnet_b0', num_classes=1000)
+    m = m.cuda().bfloat16()
+    for mod in m.modules():
+        if isinstance(mod, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+            mod.float()
+    m = m.to(memory_format=torch.channels_last)
+    x = cl(torch.rand(8, 3, 224, 224).cuda().bfloat16())
+    y = torch.randint(0, 1000, (8,)).cuda()
+    opt = native_optim.SGD(m.parameters(), lr=0.01, momentum=0.9)
+    m.train()
+    loss = F.cross_entropy(m(x).float(), y)
+    opt.zero_grad(set_to_none=False)
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss)
