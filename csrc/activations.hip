@@ -79,7 +79,7 @@ template <> struct DevT<at::Half> { using type = _Float16; };
 }  // namespace
 
 torch::Tensor act_fwd(torch::Tensor x, int64_t act) {
-  auto xc = x.contiguous();
+  auto xc = x.contiguous(x.suggest_memory_format());
   auto y = torch::empty_like(xc);
   int64_t n = xc.numel();
   NN_DISPATCH(xc.scalar_type(), "act_fwd", [&] {
@@ -93,8 +93,8 @@ torch::Tensor act_fwd(torch::Tensor x, int64_t act) {
 }
 
 torch::Tensor act_bwd(torch::Tensor g, torch::Tensor x, int64_t act) {
-  auto gc = g.contiguous();
-  auto xc = x.contiguous();
+  auto xc = x.contiguous(x.suggest_memory_format());
+  auto gc = g.contiguous(x.suggest_memory_format());
   auto gx = torch::empty_like(gc);
   int64_t n = gc.numel();
   NN_DISPATCH(gc.scalar_type(), "act_bwd", [&] {

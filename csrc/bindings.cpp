@@ -9,6 +9,8 @@ torch::Tensor ste_mask(torch::Tensor grad, torch::Tensor x, double min_value,
                        double max_value);
 torch::Tensor mult_uniform_noise(torch::Tensor x, double a, int64_t seed);
 torch::Tensor relu_clip_fwd(torch::Tensor x, bool relu, double act_max);
+torch::Tensor relu_clip_bwd(torch::Tensor g, torch::Tensor y, bool relu,
+                            double act_max);
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed);
 
 // optimizer.hip
@@ -30,6 +32,11 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x);
 torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
                          torch::Tensor invstd, torch::Tensor gamma,
                          torch::Tensor beta, bool relu, double act_max);
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
+                                      torch::Tensor y, torch::Tensor mean,
+                                      torch::Tensor invstd, torch::Tensor gamma,
+                                      bool training, bool relu,
+                                      double act_max);
 
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
@@ -65,7 +72,8 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
 std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
                                           torch::Tensor wraw, torch::Tensor bias,
                                           int64_t stride, int64_t pad,
-                                          int64_t sigma_mode, double factor,
+                                          int64_t sigma_mode,
+                                          torch::Tensor factor,
                                           int64_t seed, bool telem);
 torch::Tensor conv_dgrad(torch::Tensor gy, torch::Tensor w, int64_t stride,
                          int64_t pad, int64_t H, int64_t W);
@@ -77,16 +85,19 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x);
 std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
                                             torch::Tensor wraw,
                                             torch::Tensor bias,
-                                            int64_t sigma_mode, double factor,
+                                            int64_t sigma_mode,
+                                            torch::Tensor factor,
                                             int64_t seed, bool telem);
 std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
                                             int64_t stride, int64_t pad,
-                                            int64_t sigma_mode, double factor,
+                                            int64_t sigma_mode,
+                                            torch::Tensor factor,
                                             int64_t seed, bool telem);
 std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
                                                    torch::Tensor wraw,
                                                    int64_t sigma_mode,
-                                                   double factor, int64_t seed,
+                                                   torch::Tensor factor,
+                                                   int64_t seed,
                                                    bool telem);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -94,6 +105,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ste_mask", &ste_mask);
   m.def("mult_uniform_noise", &mult_uniform_noise);
   m.def("relu_clip_fwd", &relu_clip_fwd);
+  m.def("relu_clip_bwd", &relu_clip_bwd);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("sgd_step", &sgd_step);
   m.def("adamw_step", &adamw_step);
@@ -101,6 +113,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("bn_stats", &bn_stats);
   m.def("bn_act_fwd", &bn_act_fwd);
+  m.def("bn_act_bwd", &bn_act_bwd);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
   m.def("dwconv_fwd", &dwconv_fwd);

@@ -118,3 +118,122 @@ torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
   HIP_CHECK_LAST();
   return y;
 }
+
+// ---------------------------------------------------------------------------
+// Fused BN+act backward: (1) per-channel reduce of (g*mask) and
+// (g*mask*xhat), (2) elementwise gx. The activation mask (ReLU/clip) is
+// re-derived from the saved OUTPUT y, folding the act backward into the BN
+// backward (the reference pays separate eager kernels for each).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <typename T>
+__global__ void bn_act_bwd_reduce_kernel(
+    const T* __restrict__ g, const T* __restrict__ x, const T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ sum_g, float* __restrict__ sum_gx, int64_t rows, int C,
+    int do_relu, float act_max) {
+  int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
+  if (c >= C) return;
+  float m = mean[c], is = invstd[c];
+  float s_g = 0.0f, s_gx = 0.0f;
+  for (int64_t r = rstart; r < rows; r += gridDim.y * (kBlock / 64)) {
+    int64_t i = r * C + c;
+    float yv = to_f32(y[i]);
+    float mask = 1.0f;
+    if (do_relu && yv <= 0.0f) mask = 0.0f;
+    if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
+    float gv = to_f32(g[i]) * mask;
+    s_g += gv;
+    s_gx += gv * (to_f32(x[i]) - m) * is;
+  }
+  atomicAdd(&sum_g[c], s_g);
+  atomicAdd(&sum_gx[c], s_gx);
+}
+
+template <typename T, bool TRAIN>
+__global__ void bn_act_bwd_apply_kernel(
+    const T* __restrict__ g, const T* __restrict__ x, const T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ sum_g,
+    const float* __restrict__ sum_gx, T* __restrict__ gx, int64_t n, int C,
+    float inv_count, int do_relu, float act_max) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float yv = to_f32(y[i]);
+    float mask = 1.0f;
+    if (do_relu && yv <= 0.0f) mask = 0.0f;
+    if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
+    float gv = to_f32(g[i]) * mask;
+    float gi = gamma[c] * invstd[c];
+    if (TRAIN) {
+      float xhat = (to_f32(x[i]) - mean[c]) * invstd[c];
+      gv = gi * (gv - sum_g[c] * inv_count - xhat * sum_gx[c] * inv_count);
+    } else {
+      gv = gv * gi;
+    }
+    gx[i] = from_f32<T>(gv);
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
+                                      torch::Tensor y, torch::Tensor mean,
+                                      torch::Tensor invstd, torch::Tensor gamma,
+                                      bool training, bool relu,
+                                      double act_max) {
+  int C;
+  int64_t rows;
+  if (x.dim() == 4) {
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+    TORCH_CHECK(g.is_contiguous(at::MemoryFormat::ChannelsLast));
+    C = (int)x.size(1);
+    rows = x.size(0) * x.size(2) * x.size(3);
+  } else {
+    TORCH_CHECK(x.is_contiguous() && g.is_contiguous());
+    C = (int)x.size(1);
+    rows = x.size(0);
+  }
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sum_g = torch::zeros({C}, opts);
+  auto sum_gx = torch::zeros({C}, opts);
+  auto gx = torch::empty_like(g);
+  auto stream = c10::hip::getCurrentHIPStream();
+  int gx_blocks = (C + 63) / 64;
+  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64), 1024);
+  int64_t n = x.numel();
+  int eblocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
+  NN_DISPATCH(x.scalar_type(), "bn_act_bwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
+                       dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                       (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(),
+                       rows, C, relu ? 1 : 0, (float)act_max);
+    if (training) {
+      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
+                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                         1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
+    } else {
+      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, false>), dim3(eblocks),
+                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                         1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
+    }
+  });
+  HIP_CHECK_LAST();
+  // g_gamma = sum_gx (already masked * xhat), g_beta = sum_g
+  return {gx, sum_gx, sum_g};
+}

@@ -187,8 +187,10 @@ template <typename T, bool WANT_Y, int SIGMA_MODE, bool TELEM, bool BIAS>
 __global__ __launch_bounds__(kBlock)
 void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
                      const T* __restrict__ wraw, const float* __restrict__ bias,
-                     T* __restrict__ out, ConvGeom g, float factor,
+                     T* __restrict__ out, ConvGeom g,
+                     const float* __restrict__ factor_p,
                      uint64_t seed, float* __restrict__ telem /* [3] */) {
+  const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
   // grid: x = n-tiles, y = m-tiles
   int n0 = blockIdx.x * BN;
   int64_t m0 = (int64_t)blockIdx.y * BM;
@@ -622,7 +624,7 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
                        dim3(kBlock), lds, c10::hip::getCurrentHIPStream(),
                        (const T*)x.data_ptr(), (const T*)w.data_ptr(),
                        (const T*)w.data_ptr(), nullptr, (T*)out.data_ptr(), g,
-                       0.0f, 0, nullptr);
+                       nullptr, 0, nullptr);
   });
   HIP_CHECK_LAST();
   return out;
@@ -634,7 +636,8 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
                                                torch::Tensor bias,
                                                int64_t stride, int64_t pad,
                                                int64_t sigma_mode,
-                                               double factor, int64_t seed,
+                                               torch::Tensor factor,
+                                               int64_t seed,
                                                bool telem, bool want_y) {
   check_cl(x, "conv_fwd_fused x");
   auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
@@ -647,6 +650,7 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
   if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
   auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
   tele[2] = -std::numeric_limits<float>::infinity();
+  auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
   size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
   NN_DISPATCH(x.scalar_type(),
@@ -659,7 +663,7 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
     const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
     T* op = (T*)out.data_ptr();
     float* tp = tele.data_ptr<float>();
-    float f = (float)factor;
+    const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     auto launch = [&](auto wy, auto sm, auto tl, auto bi) {
       hipLaunchKernelGGL((conv_fwd_kernel<T, decltype(wy)::value,
@@ -780,7 +784,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w) {
                        dim3(kBlock), lds, c10::hip::getCurrentHIPStream(),
                        (const T*)x.data_ptr(), (const T*)w.data_ptr(),
                        (const T*)w.data_ptr(), nullptr, (T*)out.data_ptr(), g,
-                       0.0f, 0, nullptr);
+                       nullptr, 0, nullptr);
   });
   HIP_CHECK_LAST();
   return out;
@@ -823,7 +827,8 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
 std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
                                             torch::Tensor wraw,
                                             torch::Tensor bias,
-                                            int64_t sigma_mode, double factor,
+                                            int64_t sigma_mode,
+                                            torch::Tensor factor,
                                             int64_t seed, bool telem) {
   auto g = linear_geom(x, wraw.size(0));
   auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
@@ -832,6 +837,7 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
   if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
   auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
   tele[2] = -std::numeric_limits<float>::infinity();
+  auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
   size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
   NN_DISPATCH(x.scalar_type(),
@@ -844,7 +850,7 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
     const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
     T* op = (T*)out.data_ptr();
     float* tp = tele.data_ptr<float>();
-    float f = (float)factor;
+    const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     auto launch = [&](auto sm, auto tl, auto bi) {
       hipLaunchKernelGGL((conv_fwd_kernel<T, true, decltype(sm)::value,
@@ -867,33 +873,17 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
   return {out, tele};
 }
 
-std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
-                                          torch::Tensor wraw, torch::Tensor bias,
-                                          int64_t stride, int64_t pad,
-                                          int64_t sigma_mode, double factor,
-                                          int64_t seed, bool telem) {
-  return conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
-                             factor, seed, telem, /*want_y=*/true);
-}
-
-std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
-                                            int64_t stride, int64_t pad,
-                                            int64_t sigma_mode, double factor,
-                                            int64_t seed, bool telem) {
-  auto empty_bias = torch::empty({0}, x.options());
-  return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,
-                             sigma_mode, factor, seed, telem, /*want_y=*/false);
-}
-
 std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
                                                    torch::Tensor wraw,
                                                    int64_t sigma_mode,
-                                                   double factor, int64_t seed,
+                                                   torch::Tensor factor,
+                                                   int64_t seed,
                                                    bool telem) {
   auto g = linear_geom(x, wraw.size(0));
   auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
   auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
   tele[2] = -std::numeric_limits<float>::infinity();
+  auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
   size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
   NN_DISPATCH(x.scalar_type(),
@@ -904,7 +894,7 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
     const T* wrp = (const T*)wraw.data_ptr();
     T* op = (T*)out.data_ptr();
     float* tp = tele.data_ptr<float>();
-    float f = (float)factor;
+    const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     if (sigma_mode == 1) {
       if (telem)
@@ -928,4 +918,359 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
   });
   HIP_CHECK_LAST();
   return {out, tele};
+}
+
+// ===========================================================================
+// Image-patch conv forward: the whole (zero-padded) input image lives in LDS
+// for the block's lifetime, so the K-loop re-reads it from LDS instead of
+// re-gathering from HBM once per tap (the streaming kernel above reads each
+// input pixel R*S times). Contraction runs per filter ROW over the
+// (s, c)-contiguous span S*C_pad, which is contiguous BOTH in the padded
+// patch and in the channel-padded weights -> every fragment is an aligned
+// ds_read_b128 / 16-B weight load.
+//
+// Eligible when H_pad*W_pad*C_pad*2 fits the LDS budget -- exactly the
+// CIFAR-scale layers (conv2: 14x14x65 -> 28 KB patch) whose C=65 defeats
+// vectorized NHWC staging in the streaming kernel.
+// ===========================================================================
+
+namespace {
+
+struct PatchGeom {
+  int C_pad;   // channels padded to a multiple of 8
+  int Wp;      // W + 2*pad (patch holds the horizontal borders)
+  int Kr;      // per-row contraction length = S * C_pad
+  int MI;      // outputs per image = OH * OW
+};
+
+template <typename T>
+DEV_INLINE void stage_patch(char* patch, const T* __restrict__ x,
+                            const ConvGeom g, const PatchGeom p, int n) {
+  int total = g.H * p.Wp * p.C_pad;
+  const T* img = x + (int64_t)n * g.H * g.W * g.C;
+  bool vec = (sizeof(T) == 2) && ((g.C & 7) == 0);
+  if (vec) {
+    // 8-element vector spans within a pixel
+    int spans = total / 8;
+    for (int sidx = threadIdx.x; sidx < spans; sidx += blockDim.x) {
+      int idx = sidx * 8;
+      int c = idx % p.C_pad;
+      int pix = idx / p.C_pad;
+      int iwp = pix % p.Wp;
+      int ih = pix / p.Wp;
+      int iw = iwp - g.pad;
+      bf16x8 v;
+      if (iw >= 0 && iw < g.W && c + 8 <= g.C) {
+        v = *(const bf16x8*)(img + ((int64_t)ih * g.W + iw) * g.C + c);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ((bf16*)&v)[j] = __float2bfloat16(0.0f);
+      }
+      *(bf16x8*)(patch + (int64_t)idx * 2) = v;
+    }
+  } else {
+    for (int idx = threadIdx.x; idx < total; idx += blockDim.x) {
+      int c = idx % p.C_pad;
+      int pix = idx / p.C_pad;
+      int iwp = pix % p.Wp;
+      int ih = pix / p.Wp;
+      int iw = iwp - g.pad;
+      float v = 0.0f;
+      if (iw >= 0 && iw < g.W && c < g.C)
+        v = to_f32(img[((int64_t)ih * g.W + iw) * g.C + c]);
+      *(bf16*)(patch + (int64_t)idx * 2) = __float2bfloat16(v);
+    }
+  }
+  // zero slot at patch end for out-of-bounds fragment reads
+  if (threadIdx.x < 4)
+    *(bf16x8*)(patch + (int64_t)g.H * p.Wp * p.C_pad * 2 + threadIdx.x * 16) =
+        bf16x8{};
+}
+
+// weight tile rows from channel-padded weights [K, R, S*C_pad]
+template <typename T, bool ABS_TRANSFORM, int SIGMA_MODE>
+DEV_INLINE void stage_wpad_row(char* lds, const T* __restrict__ w,
+                               const ConvGeom g, const PatchGeom p, int n0,
+                               int r, int ck) {
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int k = n0 + row;
+  bf16 vals[8];
+  if (k < g.K) {
+    const T* pw = w + ((int64_t)k * g.R + r) * p.Kr;
+    int c0 = ck + seg * 8;
+    if (!ABS_TRANSFORM && sizeof(T) == 2 && c0 + 8 <= p.Kr) {
+      *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(pw + c0);
+      return;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float v = (c < p.Kr) ? to_f32(pw[c]) : 0.0f;
+      if (ABS_TRANSFORM) {
+        v = fabsf(v);
+        if (SIGMA_MODE == 2) v = v * v + v;
+      }
+      vals[j] = __float2bfloat16(v);
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+  }
+  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+}
+
+template <typename T, bool WANT_Y, int SIGMA_MODE, bool TELEM, bool BIAS>
+__global__ __launch_bounds__(kBlock)
+void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
+                           const T* __restrict__ wraw,
+                           const float* __restrict__ bias,
+                           T* __restrict__ out, ConvGeom g, PatchGeom p,
+                           const float* __restrict__ factor_p, uint64_t seed,
+                           float* __restrict__ telem) {
+  const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
+  int n0 = blockIdx.x * BN;   // output-channel tile
+  int n = blockIdx.y;         // image
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* patch = smem;  // H * Wp * C_pad bf16 + 64B zero slot
+  size_t patch_bytes = (size_t)g.H * p.Wp * p.C_pad * 2 + 64;
+  char* b_lds = smem + patch_bytes;
+  char* c_lds = b_lds + BN * LDS_STRIDE;
+  char* d_lds = c_lds + BN * LDS_STRIDE;
+  int zero_off = g.H * p.Wp * p.C_pad * 2;
+
+  stage_patch(patch, x, g, p, n);
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+
+  float t_sum_sigma = 0.0f, t_sum_noise = 0.0f, t_max_y = -INFINITY;
+
+  for (int m0 = 0; m0 < p.MI; m0 += BM) {
+    f32x4 acc[2][2] = {};
+    f32x4 sacc[2][2] = {};
+    f32x4 tacc[2][2] = {};
+
+    for (int r = 0; r < g.R; ++r) {
+      for (int ck = 0; ck < p.Kr; ck += BK) {
+        if (WANT_Y) stage_wpad_row<T, false, 0>(b_lds, wq, g, p, n0, r, ck);
+        if (SIGMA_MODE > 0)
+          stage_wpad_row<T, true, SIGMA_MODE>(c_lds, wraw, g, p, n0, r, ck);
+        if (TELEM && SIGMA_MODE == 2)
+          stage_wpad_row<T, true, 1>(d_lds, wraw, g, p, n0, r, ck);
+        __syncthreads();
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm) {
+          // per-lane patch fragment: 8 contiguous (s,c) at filter row r
+          int m_local = m0 + wm * 32 + fm * 16 + (lane & 15);
+          int kf = ck + (lane >> 4) * 8;
+          int off = zero_off;
+          if (m_local < p.MI) {
+            int ow = m_local % g.OW;
+            int oh = m_local / g.OW;
+            int ih = oh * g.stride - g.pad + r;
+            if (ih >= 0 && ih < g.H)
+              off = (((int64_t)ih * p.Wp + ow * g.stride) * p.C_pad + kf) * 2;
+          }
+          bf16x8 a = *(bf16x8*)(patch + off);
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn) {
+            if (WANT_Y) {
+              bf16x8 b = frag_from_lds(b_lds, wn * 32 + fn * 16);
+              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, b, acc[fm][fn], 0, 0, 0);
+            }
+            if (SIGMA_MODE > 0) {
+              bf16x8 bs = frag_from_lds(c_lds, wn * 32 + fn * 16);
+              sacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, bs, sacc[fm][fn], 0, 0, 0);
+            }
+            if (TELEM && SIGMA_MODE == 2) {
+              bf16x8 bt = frag_from_lds(d_lds, wn * 32 + fn * 16);
+              tacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a, bt, tacc[fm][fn], 0, 0, 0);
+            }
+          }
+        }
+        __syncthreads();
+      }
+    }
+
+    // epilogue for this m-subtile
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          int m_local = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
+          int k = n0 + wn * 32 + fn * 16 + (lane & 15);
+          if (m_local < p.MI && k < g.K) {
+            int64_t m = (int64_t)n * p.MI + m_local;
+            float y = WANT_Y ? acc[fm][fn][reg] : 0.0f;
+            if (BIAS) y += bias[k];
+            float v = y;
+            if (SIGMA_MODE > 0) {
+              float sig = fmaxf(sacc[fm][fn][reg], 0.0f);
+              float noise = gauss1(seed, (uint64_t)(m * g.K + k))
+                            * sqrtf(factor * sig);
+              v = y + noise;
+              if (TELEM) {
+                t_sum_noise += fabsf(noise);
+                t_max_y = fmaxf(t_max_y, y);
+                t_sum_sigma += (SIGMA_MODE == 2) ? tacc[fm][fn][reg]
+                                                 : sacc[fm][fn][reg];
+              }
+            }
+            out[m * g.K + k] = from_f32<T>(v);
+          }
+        }
+      }
+    }
+  }
+
+  if (TELEM && SIGMA_MODE > 0) {
+    __shared__ float red[4];
+    float s0 = block_sum(t_sum_sigma, red);
+    __syncthreads();
+    float s1 = block_sum(t_sum_noise, red);
+    __syncthreads();
+    float m0v = wave_max(t_max_y);
+    __shared__ float redm[4];
+    if ((threadIdx.x & (WAVE - 1)) == 0) redm[threadIdx.x / WAVE] = m0v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      atomicAdd(&telem[0], s0);
+      atomicAdd(&telem[1], s1);
+      atomic_max_f32(&telem[2],
+                     fmaxf(fmaxf(redm[0], redm[1]), fmaxf(redm[2], redm[3])));
+    }
+  }
+}
+
+inline bool patch_eligible(const ConvGeom& g) {
+  int c_pad = (g.C + 7) & ~7;
+  int wp = g.W + 2 * g.pad;
+  size_t patch_bytes = (size_t)g.H * wp * c_pad * 2 + 64;
+  return g.R * g.S > 1 && g.C > 8 && patch_bytes <= 64 * 1024;
+}
+
+// pad the raw-[K,R,S,C] weight view to [K,R,S*C_pad] (tiny tensors)
+inline torch::Tensor pad_weight_raw(const torch::Tensor& w, int c_pad) {
+  auto raw = w.permute({0, 2, 3, 1});  // logical [K,C,R,S] cl -> raw view
+  auto padded = at::constant_pad_nd(raw, {0, c_pad - (int)w.size(1)}, 0);
+  return padded.contiguous();  // [K, R, S, C_pad]
+}
+
+}  // namespace
+
+
+namespace {
+
+std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
+    torch::Tensor x, torch::Tensor wq, torch::Tensor wraw, torch::Tensor bias,
+    int64_t stride, int64_t pad, int64_t sigma_mode, torch::Tensor factor,
+    int64_t seed, bool telem, bool want_y) {
+  check_cl(x, "conv_fwd_patch x");
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
+                     (int)wraw.size(3), (int)stride, (int)pad);
+  PatchGeom p;
+  p.C_pad = (g.C + 7) & ~7;
+  p.Wp = g.W + 2 * g.pad;
+  p.Kr = g.S * p.C_pad;
+  p.MI = g.OH * g.OW;
+  auto out = torch::empty({g.N, g.K, g.OH, g.OW},
+                          x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  bool has_bias = bias.numel() > 0;
+  torch::Tensor bias_f;
+  if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
+  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+  tele[2] = -std::numeric_limits<float>::infinity();
+  auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
+  auto wq_pad = want_y ? pad_weight_raw(wq, p.C_pad) : torch::Tensor();
+  auto wraw_pad = (sigma_mode > 0) ? pad_weight_raw(wraw, p.C_pad)
+                                   : torch::Tensor();
+  size_t patch_bytes = (size_t)g.H * p.Wp * p.C_pad * 2 + 64;
+  size_t lds = patch_bytes + (size_t)3 * BN * LDS_STRIDE;
+  dim3 grid((g.K + BN - 1) / BN, g.N);
+  NN_DISPATCH(x.scalar_type(), "conv_fwd_patch", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto stream = c10::hip::getCurrentHIPStream();
+    const T* xp = (const T*)x.data_ptr();
+    const T* wqp = want_y ? (const T*)wq_pad.data_ptr()
+                          : (const T*)wraw_pad.data_ptr();
+    const T* wrp = (sigma_mode > 0) ? (const T*)wraw_pad.data_ptr() : wqp;
+    const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
+    T* op = (T*)out.data_ptr();
+    float* tp = tele.data_ptr<float>();
+    const float* f = factor_f.data_ptr<float>();
+    uint64_t sd = (uint64_t)seed;
+    auto launch = [&](auto wy, auto sm, auto tl, auto bi) {
+      hipLaunchKernelGGL((conv_fwd_patch_kernel<T, decltype(wy)::value,
+                          decltype(sm)::value, decltype(tl)::value,
+                          decltype(bi)::value>), grid, dim3(kBlock), lds,
+                         stream, xp, wqp, wrp, bp, op, g, p, f, sd, tp);
+    };
+    using TT = std::true_type; using FF = std::false_type;
+    using S0 = std::integral_constant<int, 0>;
+    using S1 = std::integral_constant<int, 1>;
+    using S2 = std::integral_constant<int, 2>;
+    if (want_y) {
+      if (sigma_mode == 0) {
+        if (has_bias) launch(TT{}, S0{}, FF{}, TT{});
+        else launch(TT{}, S0{}, FF{}, FF{});
+      } else if (sigma_mode == 1) {
+        if (telem) { if (has_bias) launch(TT{}, S1{}, TT{}, TT{}); else launch(TT{}, S1{}, TT{}, FF{}); }
+        else { if (has_bias) launch(TT{}, S1{}, FF{}, TT{}); else launch(TT{}, S1{}, FF{}, FF{}); }
+      } else {
+        if (telem) { if (has_bias) launch(TT{}, S2{}, TT{}, TT{}); else launch(TT{}, S2{}, TT{}, FF{}); }
+        else { if (has_bias) launch(TT{}, S2{}, FF{}, TT{}); else launch(TT{}, S2{}, FF{}, FF{}); }
+      }
+    } else {
+      if (sigma_mode == 1) {
+        if (telem) launch(FF{}, S1{}, TT{}, FF{}); else launch(FF{}, S1{}, FF{}, FF{});
+      } else {
+        if (telem) launch(FF{}, S2{}, TT{}, FF{}); else launch(FF{}, S2{}, FF{}, FF{});
+      }
+    }
+  });
+  HIP_CHECK_LAST();
+  return {out, tele};
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
+                                          torch::Tensor wraw, torch::Tensor bias,
+                                          int64_t stride, int64_t pad,
+                                          int64_t sigma_mode,
+                                          torch::Tensor factor,
+                                          int64_t seed, bool telem) {
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
+                     (int)wraw.size(3), (int)stride, (int)pad);
+  if (patch_eligible(g))
+    return conv_fwd_fused_patch_impl(x, wq, wraw, bias, stride, pad,
+                                     sigma_mode, factor, seed, telem, true);
+  return conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
+                             factor, seed, telem, /*want_y=*/true);
+}
+
+std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
+                                            int64_t stride, int64_t pad,
+                                            int64_t sigma_mode,
+                                            torch::Tensor factor,
+                                            int64_t seed, bool telem) {
+  auto empty_bias = torch::empty({0}, x.options());
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
+                     (int)wraw.size(3), (int)stride, (int)pad);
+  if (patch_eligible(g))
+    return conv_fwd_fused_patch_impl(x, wraw, wraw, empty_bias, stride, pad,
+                                     sigma_mode, factor, seed, telem, false);
+  return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,
+                             sigma_mode, factor, seed, telem, /*want_y=*/false);
 }

@@ -96,6 +96,21 @@ __global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ out,
   }
 }
 
+template <typename T>
+__global__ void relu_clip_bwd_kernel(const T* __restrict__ g,
+                                     const T* __restrict__ y,
+                                     T* __restrict__ out, int64_t n,
+                                     int do_relu, float act_max) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float yv = to_f32(y[i]);
+    float mask = 1.0f;
+    if (do_relu && yv <= 0.0f) mask = 0.0f;
+    if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
+    out[i] = from_f32<T>(to_f32(g[i]) * mask);
+  }
+}
+
 template <typename scalar_t> struct DevT { using type = scalar_t; };
 template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct DevT<at::Half> { using type = _Float16; };
@@ -109,7 +124,7 @@ template <> struct DevT<at::Half> { using type = _Float16; };
 torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value,
                              double max_value, double stochastic, int64_t seed) {
   TORCH_CHECK(x.is_cuda(), "fake_quant_fwd: expected GPU tensor");
-  auto xc = x.contiguous();
+  auto xc = x.contiguous(x.suggest_memory_format());
   auto out = torch::empty_like(xc);
   int64_t n = xc.numel();
   if (n == 0) return out;
@@ -138,8 +153,8 @@ torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value
 
 torch::Tensor ste_mask(torch::Tensor grad, torch::Tensor x, double min_value,
                        double max_value) {
-  auto g = grad.contiguous();
-  auto xc = x.contiguous();
+  auto g = grad.contiguous(x.suggest_memory_format());
+  auto xc = x.contiguous(x.suggest_memory_format());
   auto out = torch::empty_like(g);
   int64_t n = g.numel();
   NN_DISPATCH(g.scalar_type(),
@@ -155,7 +170,7 @@ torch::Tensor ste_mask(torch::Tensor grad, torch::Tensor x, double min_value,
 }
 
 torch::Tensor mult_uniform_noise(torch::Tensor x, double a, int64_t seed) {
-  auto xc = x.contiguous();
+  auto xc = x.contiguous(x.suggest_memory_format());
   auto out = torch::empty_like(xc);
   int64_t n = xc.numel();
   NN_DISPATCH(xc.scalar_type(),
@@ -171,7 +186,7 @@ torch::Tensor mult_uniform_noise(torch::Tensor x, double a, int64_t seed) {
 }
 
 torch::Tensor relu_clip_fwd(torch::Tensor x, bool relu, double act_max) {
-  auto xc = x.contiguous();
+  auto xc = x.contiguous(x.suggest_memory_format());
   auto out = torch::empty_like(xc);
   int64_t n = xc.numel();
   NN_DISPATCH(xc.scalar_type(),
@@ -186,7 +201,7 @@ torch::Tensor relu_clip_fwd(torch::Tensor x, bool relu, double act_max) {
 }
 
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed) {
-  auto xc = x.contiguous();
+  auto xc = x.contiguous(x.suggest_memory_format());
   auto out = torch::empty_like(xc);
   auto mask = torch::empty_like(xc);
   int64_t n = xc.numel();
@@ -201,4 +216,26 @@ std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed) 
   });
   HIP_CHECK_LAST();
   return {out, mask};
+}
+
+// grad of relu+clip re-derived from the saved OUTPUT y
+torch::Tensor relu_clip_bwd(torch::Tensor g, torch::Tensor y, bool relu,
+                            double act_max) {
+  auto yc = y.contiguous(y.suggest_memory_format());
+  auto gc = g.contiguous(y.suggest_memory_format());
+  auto out = torch::empty_like(gc);
+  int64_t n = gc.numel();
+  NN_DISPATCH(gc.scalar_type(), "relu_clip_bwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    float lo = relu ? 0.0f : -INFINITY;
+    float hi = act_max > 0 ? (float)act_max : INFINITY;
+    // reuse ste_mask semantics: zero where y <= lo or y >= hi
+    hipLaunchKernelGGL((relu_clip_bwd_kernel<T>), dim3(grid_1d(n)),
+                       dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                       (const T*)gc.data_ptr(), (const T*)yc.data_ptr(),
+                       (T*)out.data_ptr(), n, relu ? 1 : 0,
+                       act_max > 0 ? (float)act_max : 0.0f);
+  });
+  HIP_CHECK_LAST();
+  return out;
 }

@@ -66,6 +66,14 @@ def fake_quant(x, num_bits, min_value, max_value, stochastic=0.0):
     return FakeQuant.apply(x, num_bits, min_value, max_value, stochastic)
 
 
+
+def _factor_tensor(f, device):
+    """Noise factor as a device-resident f32 scalar (no host sync)."""
+    if isinstance(f, torch.Tensor):
+        return f.detach().to(device=device, dtype=torch.float32).reshape(1)
+    return torch.tensor([float(f)], device=device, dtype=torch.float32)
+
+
 # ---------------------------------------------------------------------------
 # Conv primitives: forward / dgrad / wgrad, each differentiable by composition
 # ---------------------------------------------------------------------------
@@ -75,9 +83,24 @@ def _nhwc(t):
     return t.contiguous(memory_format=torch.channels_last)
 
 
+def _patch_eligible(x, w, padding):
+    """Mirror of csrc patch_eligible: image-resident LDS conv mode."""
+    C = x.shape[1]
+    R, S = w.shape[2], w.shape[3]
+    c_pad = (C + 7) // 8 * 8
+    wp = x.shape[3] + 2 * padding
+    return R * S > 1 and C > 8 and x.shape[2] * wp * c_pad * 2 + 64 <= 65536
+
+
 def _conv_fwd_raw(x, w, bias, stride, padding):
     if use_native(x, w):
-        y = ext().conv_fwd(_nhwc(x), _nhwc(w), stride, padding)
+        if _patch_eligible(x, w, padding):
+            empty = torch.empty(0, device=x.device, dtype=x.dtype)
+            zero_f = torch.zeros(1, device=x.device, dtype=torch.float32)
+            y, _ = ext().conv_fwd_fused(_nhwc(x), _nhwc(w), _nhwc(w), empty,
+                                        stride, padding, 0, zero_f, 0, False)
+        else:
+            y = ext().conv_fwd(_nhwc(x), _nhwc(w), stride, padding)
         if bias is not None:
             y = y + bias.view(1, -1, 1, 1)
         return y
@@ -205,7 +228,8 @@ class _FusedNoisyConv(torch.autograd.Function):
                 _nhwc(x), _nhwc(wq), _nhwc(w_raw),
                 bias if bias is not None else torch.empty(0, device=x.device, dtype=x.dtype),
                 stride, padding, 1 if sigma_mode == "abs" else 2,
-                float(factor), _next_seed(x.device), bool(want_telemetry))
+                _factor_tensor(factor, x.device), _next_seed(x.device),
+                bool(want_telemetry))
             if want_telemetry:
                 sum_sigma_abs = tele[0] / x.shape[0]
                 sum_abs_noise = tele[1] / y.numel()
@@ -271,7 +295,7 @@ def sigma_noise_conv2d(x, w_raw, sigma_mode, factor, stride=1, padding=0,
     if use_native(x, w_raw):
         noise, tele = ext().sigma_noise_conv(
             _nhwc(x), _nhwc(w_raw), stride, padding,
-            1 if sigma_mode == "abs" else 2, float(factor),
+            1 if sigma_mode == "abs" else 2, _factor_tensor(factor, x.device),
             _next_seed(x.device), bool(want_sigma_abs))
         sig_mean = tele[0] / x.shape[0] if want_sigma_abs else None
         return noise, sig_mean
@@ -289,7 +313,7 @@ def sigma_noise_linear(x, w_raw, sigma_mode, factor, want_sigma_abs=False):
     if use_native(x, w_raw):
         noise, tele = ext().sigma_noise_linear(
             x.contiguous(), w_raw.contiguous(),
-            1 if sigma_mode == "abs" else 2, float(factor),
+            1 if sigma_mode == "abs" else 2, _factor_tensor(factor, x.device),
             _next_seed(x.device), bool(want_sigma_abs))
         sig_mean = tele[0] / x.shape[0] if want_sigma_abs else None
         return noise, sig_mean
@@ -391,7 +415,8 @@ class _FusedNoisyLinear(torch.autograd.Function):
             y, tele = ext().linear_fwd_fused(
                 x.contiguous(), wq.contiguous(), w_raw.contiguous(),
                 bias if bias is not None else torch.empty(0, device=x.device, dtype=x.dtype),
-                1 if sigma_mode == "abs" else 2, float(factor),
+                1 if sigma_mode == "abs" else 2,
+                _factor_tensor(factor, x.device),
                 _next_seed(x.device), bool(want_telemetry))
             if want_telemetry:
                 sum_sigma_abs = tele[0] / x.shape[0]
@@ -530,6 +555,16 @@ class BnAct(torch.autograd.Function):
     @staticmethod
     def backward(ctx, g):
         x, weight, mean, invstd, y = ctx.saved_tensors
+        if use_native(x):
+            gc = _nhwc(g) if g.dim() == 4 else g.contiguous()
+            gx, g_gamma, g_beta = ext().bn_act_bwd(
+                gc, _nhwc(x) if x.dim() == 4 else x.contiguous(),
+                _nhwc(y) if y.dim() == 4 else y.contiguous(),
+                mean.contiguous(), invstd.contiguous(),
+                weight.float().contiguous(), bool(ctx.training),
+                bool(ctx.relu), float(ctx.act_max))
+            return (gx, g_gamma.to(weight.dtype),
+                    g_beta.to(weight.dtype)) + (None,) * 7
         shape = (1, -1, 1, 1) if x.dim() == 4 else (1, -1)
         # activation mask: dy/dz = 1 where 0 < y (< act_max if clipped)
         mask = torch.ones_like(y)
@@ -800,6 +835,9 @@ class ReluClip(torch.autograd.Function):
     @staticmethod
     def backward(ctx, g):
         (y,) = ctx.saved_tensors
+        if use_native(y):
+            return (ext().relu_clip_bwd(g.contiguous(), y, bool(ctx.relu),
+                                        float(ctx.act_max)), None, None)
         mask = torch.ones_like(y)
         if ctx.relu:
             mask = mask * (y > 0).to(g.dtype)
