@@ -196,7 +196,7 @@ def test_fused_conv_noise_statistics():
     factor = 0.05
     out, tele = ops.ext().conv_fwd_fused(
         cl(x), cl(wq), cl(w), torch.empty(0, device=dev(), dtype=torch.bfloat16),
-        1, 0, 1, factor, 42, True)
+        1, 0, 1, torch.tensor([factor], device=dev()), 42, True)
     clean = F.conv2d(x.float().cpu(), w.float().cpu())
     noise = out.float().cpu() - clean
     sig = F.conv2d(x.float().cpu(), w.float().cpu().abs())
@@ -218,7 +218,7 @@ def test_fused_conv_abs2_mode():
     factor = 0.02
     out, tele = ops.ext().conv_fwd_fused(
         cl(x), cl(w), cl(w), torch.empty(0, device=dev(), dtype=torch.bfloat16),
-        1, 0, 2, factor, 7, True)
+        1, 0, 2, torch.tensor([factor], device=dev()), 7, True)
     clean = F.conv2d(x.float().cpu(), w.float().cpu())
     noise = out.float().cpu() - clean
     aw = w.float().cpu().abs()
@@ -234,7 +234,8 @@ def test_sigma_noise_only():
     torch.manual_seed(6)
     x = torch.rand(32, 8, 10, 10).to(dev()).bfloat16()
     w = (torch.randn(16, 8, 3, 3) * 0.3).to(dev()).bfloat16()
-    noise, tele = ops.ext().sigma_noise_conv(cl(x), cl(w), 1, 0, 1, 0.05, 9, False)
+    noise, tele = ops.ext().sigma_noise_conv(cl(x), cl(w), 1, 0, 1,
+        torch.tensor([0.05], device=dev()), 9, False)
     aw = w.float().cpu().abs()
     sig = F.conv2d(x.float().cpu(), aw)
     z = noise.float().cpu() / (0.05 * sig).clamp_min(1e-9).sqrt()
@@ -553,3 +554,24 @@ def test_efficientnet_b0_step_gpu():
     loss.backward()
     opt.step()
     assert torch.isfinite(loss)
+
+
+def test_patch_mode_conv_matches_torch():
+    """Shapes that trigger the image-patch LDS-resident kernel (small
+    spatial, C>8) vs torch, including the odd-C conv2 shape."""
+    torch.manual_seed(21)
+    for (N, C, H, W, K, R, stride, pad) in (
+            (8, 65, 14, 14, 120, 5, 1, 0),   # NoisyNet conv2
+            (4, 24, 16, 16, 32, 3, 1, 1),    # padded 3x3
+            (4, 16, 15, 15, 24, 3, 2, 1)):   # strided
+        x = to_bf16_representable(torch.randn(N, C, H, W)).to(dev()).bfloat16()
+        w = to_bf16_representable(torch.randn(K, C, R, R) * 0.2).to(dev()).bfloat16()
+        from noisynet_amd.ops.functional import _patch_eligible
+        assert _patch_eligible(x, w, pad)
+        empty = torch.empty(0, device=dev(), dtype=torch.bfloat16)
+        zf = torch.zeros(1, device=dev())
+        y, _ = ops.ext().conv_fwd_fused(cl(x), cl(w), cl(w), empty,
+                                        stride, pad, 0, zf, 0, False)
+        ye = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad)
+        rel = (y.float().cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
+        assert rel < 0.02, rel
