@@ -1048,10 +1048,12 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
 
   float t_sum_sigma = 0.0f, t_sum_noise = 0.0f, t_max_y = -INFINITY;
 
-  for (int m0 = 0; m0 < p.MI; m0 += BM) {
-    f32x4 acc[2][2] = {};
-    f32x4 sacc[2][2] = {};
-    f32x4 tacc[2][2] = {};
+  // two M-subtiles share each staged weight tile (halves barrier count)
+  for (int m0 = 0; m0 < p.MI; m0 += 2 * BM) {
+    f32x4 acc[2][2][2] = {};
+    f32x4 sacc[2][2][2] = {};
+    f32x4 tacc[2][2][2] = {};
+    const int nsub = (m0 + BM < p.MI) ? 2 : 1;
 
     for (int r = 0; r < g.R; ++r) {
       for (int ck = 0; ck < p.Kr; ck += BK) {
@@ -1061,36 +1063,43 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
         if (TELEM && SIGMA_MODE == 2)
           stage_wpad_row<T, true, 1>(d_lds, wraw, g, p, n0, r, ck);
         __syncthreads();
+        bf16x8 bfrag[2], sfrag[2], tfrag[2];
 #pragma unroll
-        for (int fm = 0; fm < 2; ++fm) {
-          // per-lane patch fragment: 8 contiguous (s,c) at filter row r
-          int m_local = m0 + wm * 32 + fm * 16 + (lane & 15);
-          int kf = ck + (lane >> 4) * 8;
-          int off = zero_off;
-          if (m_local < p.MI) {
-            int ow = m_local % g.OW;
-            int oh = m_local / g.OW;
-            int ih = oh * g.stride - g.pad + r;
-            if (ih >= 0 && ih < g.H)
-              off = (((int64_t)ih * p.Wp + ow * g.stride) * p.C_pad + kf) * 2;
-          }
-          bf16x8 a = *(bf16x8*)(patch + off);
+        for (int fn = 0; fn < 2; ++fn) {
+          if (WANT_Y) bfrag[fn] = frag_from_lds(b_lds, wn * 32 + fn * 16);
+          if (SIGMA_MODE > 0)
+            sfrag[fn] = frag_from_lds(c_lds, wn * 32 + fn * 16);
+          if (TELEM && SIGMA_MODE == 2)
+            tfrag[fn] = frag_from_lds(d_lds, wn * 32 + fn * 16);
+        }
 #pragma unroll
-          for (int fn = 0; fn < 2; ++fn) {
-            if (WANT_Y) {
-              bf16x8 b = frag_from_lds(b_lds, wn * 32 + fn * 16);
-              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a, b, acc[fm][fn], 0, 0, 0);
+        for (int ms = 0; ms < 2; ++ms) {
+          if (ms == 1 && nsub == 1) break;
+#pragma unroll
+          for (int fm = 0; fm < 2; ++fm) {
+            // per-lane patch fragment: 8 contiguous (s,c) at filter row r
+            int m_local = m0 + ms * BM + wm * 32 + fm * 16 + (lane & 15);
+            int kf = ck + (lane >> 4) * 8;
+            int off = zero_off;
+            if (m_local < p.MI) {
+              int ow = m_local % g.OW;
+              int oh = m_local / g.OW;
+              int ih = oh * g.stride - g.pad + r;
+              if (ih >= 0 && ih < g.H)
+                off = (((int64_t)ih * p.Wp + ow * g.stride) * p.C_pad + kf) * 2;
             }
-            if (SIGMA_MODE > 0) {
-              bf16x8 bs = frag_from_lds(c_lds, wn * 32 + fn * 16);
-              sacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a, bs, sacc[fm][fn], 0, 0, 0);
-            }
-            if (TELEM && SIGMA_MODE == 2) {
-              bf16x8 bt = frag_from_lds(d_lds, wn * 32 + fn * 16);
-              tacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a, bt, tacc[fm][fn], 0, 0, 0);
+            bf16x8 a = *(bf16x8*)(patch + off);
+#pragma unroll
+            for (int fn = 0; fn < 2; ++fn) {
+              if (WANT_Y)
+                acc[ms][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, bfrag[fn], acc[ms][fm][fn], 0, 0, 0);
+              if (SIGMA_MODE > 0)
+                sacc[ms][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, sfrag[fn], sacc[ms][fm][fn], 0, 0, 0);
+              if (TELEM && SIGMA_MODE == 2)
+                tacc[ms][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, tfrag[fn], tacc[ms][fm][fn], 0, 0, 0);
             }
           }
         }
@@ -1098,33 +1107,38 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
       }
     }
 
-    // epilogue for this m-subtile
+    // epilogue for both m-subtiles
 #pragma unroll
-    for (int fm = 0; fm < 2; ++fm) {
+    for (int ms = 0; ms < 2; ++ms) {
+      if (ms == 1 && nsub == 1) break;
 #pragma unroll
-      for (int fn = 0; fn < 2; ++fn) {
+      for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
-        for (int reg = 0; reg < 4; ++reg) {
-          int m_local = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
-          int k = n0 + wn * 32 + fn * 16 + (lane & 15);
-          if (m_local < p.MI && k < g.K) {
-            int64_t m = (int64_t)n * p.MI + m_local;
-            float y = WANT_Y ? acc[fm][fn][reg] : 0.0f;
-            if (BIAS) y += bias[k];
-            float v = y;
-            if (SIGMA_MODE > 0) {
-              float sig = fmaxf(sacc[fm][fn][reg], 0.0f);
-              float noise = gauss1(seed, (uint64_t)(m * g.K + k))
-                            * sqrtf(factor * sig);
-              v = y + noise;
-              if (TELEM) {
-                t_sum_noise += fabsf(noise);
-                t_max_y = fmaxf(t_max_y, y);
-                t_sum_sigma += (SIGMA_MODE == 2) ? tacc[fm][fn][reg]
-                                                 : sacc[fm][fn][reg];
+        for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) {
+            int m_local = m0 + ms * BM + wm * 32 + fm * 16
+                          + 4 * (lane >> 4) + reg;
+            int k = n0 + wn * 32 + fn * 16 + (lane & 15);
+            if (m_local < p.MI && k < g.K) {
+              int64_t m = (int64_t)n * p.MI + m_local;
+              float y = WANT_Y ? acc[ms][fm][fn][reg] : 0.0f;
+              if (BIAS) y += bias[k];
+              float v = y;
+              if (SIGMA_MODE > 0) {
+                float sig = fmaxf(sacc[ms][fm][fn][reg], 0.0f);
+                float noise = gauss1(seed, (uint64_t)(m * g.K + k))
+                              * sqrtf(factor * sig);
+                v = y + noise;
+                if (TELEM) {
+                  t_sum_noise += fabsf(noise);
+                  t_max_y = fmaxf(t_max_y, y);
+                  t_sum_sigma += (SIGMA_MODE == 2) ? tacc[ms][fm][fn][reg]
+                                                   : sacc[ms][fm][fn][reg];
+                }
               }
+              out[m * g.K + k] = from_f32<T>(v);
             }
-            out[m * g.K + k] = from_f32<T>(v);
           }
         }
       }
@@ -1154,7 +1168,7 @@ inline bool patch_eligible(const ConvGeom& g) {
   int c_pad = (g.C + 7) & ~7;
   int wp = g.W + 2 * g.pad;
   size_t patch_bytes = (size_t)g.H * wp * c_pad * 2 + 64;
-  return g.R * g.S > 1 && g.C > 8 && patch_bytes <= 64 * 1024;
+  return g.R * g.S > 1 && patch_bytes <= 64 * 1024;
 }
 
 // pad the raw-[K,R,S,C] weight view to [K,R,S*C_pad] (tiny tensors)

@@ -89,7 +89,7 @@ def _patch_eligible(x, w, padding):
     R, S = w.shape[2], w.shape[3]
     c_pad = (C + 7) // 8 * 8
     wp = x.shape[3] + 2 * padding
-    return R * S > 1 and C > 8 and x.shape[2] * wp * c_pad * 2 + 64 <= 65536
+    return R * S > 1 and x.shape[2] * wp * c_pad * 2 + 64 <= 65536
 
 
 def _conv_fwd_raw(x, w, bias, stride, padding):
@@ -109,6 +109,16 @@ def _conv_fwd_raw(x, w, bias, stride, padding):
 
 def _conv_dgrad_raw(g, w, stride, padding, x_shape):
     if use_native(g, w):
+        R, S = w.shape[2], w.shape[3]
+        if stride == 1 and _patch_eligible(g, w, R - 1 - padding):
+            # dgrad(stride 1) == conv of g with the flipped/transposed
+            # filter at full-correlation padding -> image-patch kernel
+            w2 = _nhwc(w.flip((2, 3)).transpose(0, 1))
+            empty = torch.empty(0, device=g.device, dtype=g.dtype)
+            zero_f = torch.zeros(1, device=g.device, dtype=torch.float32)
+            dx, _ = ext().conv_fwd_fused(_nhwc(g), w2, w2, empty, 1,
+                                         R - 1 - padding, 0, zero_f, 0, False)
+            return dx
         return ext().conv_dgrad(_nhwc(g), _nhwc(w), stride, padding,
                                 x_shape[2], x_shape[3])
     return torch.nn.grad.conv2d_input(x_shape, w, g, stride, padding)
