@@ -37,6 +37,16 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
                                       torch::Tensor invstd, torch::Tensor gamma,
                                       bool training, bool relu,
                                       double act_max);
+std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
+                                             torch::Tensor y, torch::Tensor mean,
+                                             torch::Tensor invstd, bool relu,
+                                             double act_max);
+torch::Tensor bn_act_bwd_apply(torch::Tensor g, torch::Tensor x,
+                               torch::Tensor y, torch::Tensor mean,
+                               torch::Tensor invstd, torch::Tensor gamma,
+                               torch::Tensor sum_g, torch::Tensor sum_gx,
+                               double count, bool training, bool relu,
+                               double act_max);
 
 // pool.hip
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x);
@@ -114,6 +124,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_stats", &bn_stats);
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("bn_act_bwd", &bn_act_bwd);
+  m.def("bn_act_bwd_reduce", &bn_act_bwd_reduce);
+  m.def("bn_act_bwd_apply", &bn_act_bwd_apply);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
   m.def("dwconv_fwd", &dwconv_fwd);

@@ -237,3 +237,71 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
   // g_gamma = sum_gx (already masked * xhat), g_beta = sum_g
   return {gx, sum_gx, sum_g};
 }
+
+
+// Split backward for SyncBN: reduce -> (RCCL all-reduce in Python) -> apply.
+std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
+                                             torch::Tensor y, torch::Tensor mean,
+                                             torch::Tensor invstd, bool relu,
+                                             double act_max) {
+  int C;
+  int64_t rows;
+  if (x.dim() == 4) {
+    C = (int)x.size(1);
+    rows = x.size(0) * x.size(2) * x.size(3);
+  } else {
+    C = (int)x.size(1);
+    rows = x.size(0);
+  }
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sum_g = torch::zeros({C}, opts);
+  auto sum_gx = torch::zeros({C}, opts);
+  int gx_blocks = (C + 63) / 64;
+  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64), 1024);
+  NN_DISPATCH(x.scalar_type(), "bn_act_bwd_reduce", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
+                       dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                       (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                       (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), sum_g.data_ptr<float>(),
+                       sum_gx.data_ptr<float>(), rows, C, relu ? 1 : 0,
+                       (float)act_max);
+  });
+  HIP_CHECK_LAST();
+  return {sum_g, sum_gx};
+}
+
+torch::Tensor bn_act_bwd_apply(torch::Tensor g, torch::Tensor x,
+                               torch::Tensor y, torch::Tensor mean,
+                               torch::Tensor invstd, torch::Tensor gamma,
+                               torch::Tensor sum_g, torch::Tensor sum_gx,
+                               double count, bool training, bool relu,
+                               double act_max) {
+  int C = (int)x.size(1);
+  auto gx = torch::empty_like(g);
+  int64_t n = x.numel();
+  int eblocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
+  NN_DISPATCH(x.scalar_type(), "bn_act_bwd_apply", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (training)
+      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
+                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                         1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+    else
+      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, false>), dim3(eblocks),
+                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                         1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+  });
+  HIP_CHECK_LAST();
+  return gx;
+}

@@ -56,8 +56,9 @@ def build_model(args):
     if device.type == 'cuda':
         model = model.to(memory_format=torch.channels_last)
 
-    if args.sync_bn and distributed:
-        model = nn.SyncBatchNorm.convert_sync_batchnorm(model)
+    # --sync-bn uses the FUSED SyncBN: models pass sync=args.sync_bn into
+    # ops.bn_act, which all-reduces (mean, E[x^2]) across ranks inside the
+    # fused BN kernel path (SURVEY.md §5 distributed row).
 
     dp = dist_mod.DataParallel(model) if distributed else None
     return model, dp, device, distributed

@@ -69,7 +69,8 @@ class ConvBNReLU(nn.Module):
             x = ops.bn_act(x, self.bn.weight, self.bn.bias,
                            self.bn.running_mean, self.bn.running_var,
                            self.training, self.bn.momentum, self.bn.eps,
-                           relu=True, act_max=6.0)
+                           relu=True, act_max=6.0,
+                           sync=getattr(args, 'sync_bn', False))
         return x
 
 
@@ -116,7 +117,8 @@ class InvertedResidual(nn.Module):
             x = ops.bn_act(x, self.bn.weight, self.bn.bias,
                            self.bn.running_mean, self.bn.running_var,
                            self.training, self.bn.momentum, self.bn.eps,
-                           relu=False, act_max=0.0)
+                           relu=False, act_max=0.0,
+                           sync=getattr(args, 'sync_bn', False))
         if self.use_res_connect:
             return x + input
         return x

@@ -137,7 +137,8 @@ class Net(nn.Module):
         return ops.bn_act(x, bn.weight, bn.bias, bn.running_mean,
                           bn.running_var,
                           self.training or not self.args.track_running_stats,
-                          bn.momentum, bn.eps, relu=True, act_max=act_max)
+                          bn.momentum, bn.eps, relu=True, act_max=act_max,
+                          sync=getattr(self.args, 'sync_bn', False))
 
     # ------------------------------------------------------------------
     def forward(self, input, epoch=0, i=0, s=0, acc=0.0):

@@ -86,7 +86,8 @@ class BasicBlock(nn.Module):
         return ops.bn_act(x, bn.weight, bn.bias, bn.running_mean,
                           bn.running_var,
                           self.training or not args.track_running_stats,
-                          bn.momentum, bn.eps, relu=relu, act_max=act_max)
+                          bn.momentum, bn.eps, relu=relu, act_max=act_max,
+                          sync=getattr(args, 'sync_bn', False))
 
     def forward(self, x):
         args = self.args
@@ -223,7 +224,7 @@ class ResNet(nn.Module):
                            self.bn1.running_mean, self.bn1.running_var,
                            self.training or not args.track_running_stats,
                            self.bn1.momentum, self.bn1.eps, relu=False,
-                           act_max=0.0)
+                           act_max=0.0, sync=getattr(args, 'sync_bn', False))
         x = ops.relu_clip(x, act_max if act_max > 0 else 0.0)
         x = ops.maxpool_nhwc(x, 3, 2, 1)
 

@@ -526,20 +526,34 @@ def add_weight_noise(w, a):
 class BnAct(torch.autograd.Function):
     """BN (train: batch stats; eval: running stats) fused with ReLU and an
     optional upper clip. Backward folds the activation mask into the BN
-    gradient. Native path: csrc/bn_act.hip (Welford partials + one fused
-    normalize/act pass)."""
+    gradient. Native path: csrc/bn_act.hip (per-channel partials + one fused
+    normalize/act pass).
+
+    ``sync=True`` is the fused SyncBN (main.py:786-796 equivalent): the
+    per-rank (mean, E[x^2]) pairs are combined with ONE RCCL all-reduce
+    before normalization, and the backward's per-channel sums are likewise
+    all-reduced before the apply pass."""
 
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu, act_max):
+                momentum, eps, relu, act_max, sync=False):
+        import torch.distributed as dist
+        ws = dist.get_world_size() if (sync and dist.is_initialized()) else 1
+        ctx.sync_ws = ws if training else 1
         if training:
             if use_native(x):
                 mean, var = ext().bn_stats(_nhwc(x) if x.dim() == 4 else x.contiguous())
             else:
                 mean, var = ref.bn_stats(x)
+            if ws > 1:
+                stats = torch.stack([mean, var + mean * mean])
+                dist.all_reduce(stats)
+                stats /= ws
+                mean = stats[0]
+                var = (stats[1] - mean * mean).clamp_min(0)
             if running_mean is not None:
                 with torch.no_grad():
-                    n = x.numel() / x.shape[1]
+                    n = x.numel() / x.shape[1] * ws
                     unbiased = var * (n / max(n - 1.0, 1.0))
                     running_mean.mul_(1 - momentum).add_(momentum * mean.to(running_mean.dtype))
                     running_var.mul_(1 - momentum).add_(momentum * unbiased.to(running_var.dtype))
@@ -564,17 +578,35 @@ class BnAct(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, g):
+        import torch.distributed as dist
         x, weight, mean, invstd, y = ctx.saved_tensors
+        ws = getattr(ctx, 'sync_ws', 1)
         if use_native(x):
             gc = _nhwc(g) if g.dim() == 4 else g.contiguous()
+            xc = _nhwc(x) if x.dim() == 4 else x.contiguous()
+            yc = _nhwc(y) if y.dim() == 4 else y.contiguous()
+            if ws > 1:
+                sum_g_loc, sum_gx_loc = ext().bn_act_bwd_reduce(
+                    gc, xc, yc, mean.contiguous(), invstd.contiguous(),
+                    bool(ctx.relu), float(ctx.act_max))
+                pair = torch.stack([sum_g_loc, sum_gx_loc])
+                dist.all_reduce(pair)
+                sum_g, sum_gx = pair[0].contiguous(), pair[1].contiguous()
+                count = x.numel() / x.shape[1] * ws
+                gx = ext().bn_act_bwd_apply(
+                    gc, xc, yc, mean.contiguous(), invstd.contiguous(),
+                    weight.float().contiguous(), sum_g, sum_gx, float(count),
+                    bool(ctx.training), bool(ctx.relu), float(ctx.act_max))
+                # gamma/beta grads stay LOCAL sums: the data-parallel bucket
+                # all-reduce averages them like torch SyncBatchNorm expects
+                return (gx, sum_gx_loc.to(weight.dtype),
+                        sum_g_loc.to(weight.dtype)) + (None,) * 8
             gx, g_gamma, g_beta = ext().bn_act_bwd(
-                gc, _nhwc(x) if x.dim() == 4 else x.contiguous(),
-                _nhwc(y) if y.dim() == 4 else y.contiguous(),
-                mean.contiguous(), invstd.contiguous(),
+                gc, xc, yc, mean.contiguous(), invstd.contiguous(),
                 weight.float().contiguous(), bool(ctx.training),
                 bool(ctx.relu), float(ctx.act_max))
             return (gx, g_gamma.to(weight.dtype),
-                    g_beta.to(weight.dtype)) + (None,) * 7
+                    g_beta.to(weight.dtype)) + (None,) * 8
         shape = (1, -1, 1, 1) if x.dim() == 4 else (1, -1)
         # activation mask: dy/dz = 1 where 0 < y (< act_max if clipped)
         mask = torch.ones_like(y)
@@ -587,21 +619,27 @@ class BnAct(torch.autograd.Function):
         xhat = (x.float() - mean.view(shape)) * invstd.view(shape)
         g_gamma = (g * xhat).sum(dims)
         g_beta = g.sum(dims)
+        if ws > 1:
+            pair = torch.stack([g_beta, g_gamma])
+            dist.all_reduce(pair)
+            g_beta_r, g_gamma_r = pair[0], pair[1]
+        else:
+            g_beta_r, g_gamma_r = g_beta, g_gamma
         wf = weight.float()
         if ctx.training:
-            n = x.numel() / x.shape[1]
+            n = x.numel() / x.shape[1] * ws
             gx = (wf.view(shape) * invstd.view(shape)) * (
-                g - g_beta.view(shape) / n - xhat * g_gamma.view(shape) / n)
+                g - g_beta_r.view(shape) / n - xhat * g_gamma_r.view(shape) / n)
         else:
             gx = g * (wf.view(shape) * invstd.view(shape))
         return (gx.to(x.dtype), g_gamma.to(weight.dtype),
-                g_beta.to(weight.dtype)) + (None,) * 7
+                g_beta.to(weight.dtype)) + (None,) * 8
 
 
 def bn_act(x, weight, bias, running_mean, running_var, training, momentum,
-           eps, relu=True, act_max=0.0):
+           eps, relu=True, act_max=0.0, sync=False):
     return BnAct.apply(x, weight, bias, running_mean, running_var, training,
-                       momentum, eps, relu, act_max)
+                       momentum, eps, relu, act_max, sync)
 
 
 # ---------------------------------------------------------------------------

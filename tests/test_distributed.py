@@ -114,3 +114,64 @@ def test_bench_multiprocess_cpu():
     assert rec["scaling"] == "weak"
     assert rec["value"] > 0
     assert outs[1].strip() == "" or "{" not in outs[1]
+
+
+def _syncbn_worker(rank, world_size, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    from noisynet_amd import distributed as dist_mod
+    from noisynet_amd import ops
+
+    dist_mod.init_distributed(backend="gloo")
+    torch.manual_seed(7)
+    full_x = torch.randn(8, 6, 5, 5)
+    gamma = torch.randn(6, requires_grad=True)
+    beta = torch.randn(6, requires_grad=True)
+    rm = torch.zeros(6)
+    rv = torch.ones(6)
+
+    # sync path on this rank's shard
+    shard = full_x[rank * 4:(rank + 1) * 4].clone().requires_grad_(True)
+    y = ops.bn_act(shard, gamma, beta, rm.clone(), rv.clone(), True, 0.1,
+                   1e-5, relu=True, act_max=2.0, sync=True)
+    g_out = torch.ones_like(y)
+    y.backward(g_out)
+
+    # reference: single-process BN over the FULL batch
+    gamma2 = gamma.detach().clone().requires_grad_(True)
+    beta2 = beta.detach().clone().requires_grad_(True)
+    full = full_x.clone().requires_grad_(True)
+    y2 = ops.bn_act(full, gamma2, beta2, rm.clone(), rv.clone(), True, 0.1,
+                    1e-5, relu=True, act_max=2.0, sync=False)
+    y2.backward(torch.ones_like(y2))
+
+    ok_fwd = torch.allclose(y, y2[rank * 4:(rank + 1) * 4], atol=1e-5)
+    ok_gx = torch.allclose(shard.grad, full.grad[rank * 4:(rank + 1) * 4],
+                           atol=1e-5)
+    # gamma grad: local sums average to global/ws; sum across ranks == full
+    gsum = gamma.grad.clone()
+    dist.all_reduce(gsum)
+    ok_gamma = torch.allclose(gsum, gamma2.grad, atol=1e-4)
+    q.put((rank, bool(ok_fwd and ok_gx and ok_gamma),
+           (ok_fwd, ok_gx, ok_gamma)))
+    dist.destroy_process_group()
+
+
+def test_syncbn_matches_full_batch_gloo():
+    """Fused SyncBN == single-process BN over the concatenated batch."""
+    world_size = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_syncbn_worker, args=(r, world_size, 29531, q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world_size)]
+    for p in procs:
+        p.join(timeout=120)
+    assert all(ok for _, ok, _ in results), results
