@@ -36,14 +36,21 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=8)
-    p.add_argument("--batch", type=int, default=2048,
-                   help="per-GPU batch size (weak scaling)")
+    p.add_argument("--model", type=str, default="noisynet",
+                   choices=["noisynet", "resnet18", "mobilenet_v2",
+                            "efficientnet_b0"],
+                   help="BASELINE.json configs 2-5; default = flagship")
+    p.add_argument("--batch", type=int, default=0,
+                   help="per-GPU batch size (weak scaling); 0 = per-model default")
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32"])
     p.add_argument("--no-noise", action="store_true",
-                   help="disable analog noise (noise-free baseline config)")
+                   help="disable analog/weight noise (noise-free baseline)")
     p.add_argument("--seed", type=int, default=42)
-    return p.parse_args()
+    args = p.parse_args()
+    if args.batch == 0:
+        args.batch = 2048 if args.model == "noisynet" else 256
+    return args
 
 
 def flagship_args(bench):
@@ -57,6 +64,50 @@ def flagship_args(bench):
     args = build_noisynet_parser().parse_args(argv)
     broadcast_per_layer(args)
     return args
+
+
+def build_secondary(bench, device, dtype):
+    """BASELINE.json configs 3-5: ResNet-18 CIFAR-shaped 4-bit + weight
+    noise; MobileNetV2 / EfficientNet-B0 ImageNet-shaped quant-aware."""
+    from noisynet_amd.config import build_main_parser
+
+    if bench.model == "resnet18":
+        argv = ["-a", "resnet18", "--q_a", "4", "--q_w", "4",
+                "--calculate_running"]
+        if not bench.no_noise:
+            argv += ["--n_w", "0.1"]
+        image_size, num_classes = 32, 10   # CIFAR-10-shaped (config 3)
+    else:
+        argv = ["-a", bench.model, "--q_a", "4", "--calculate_running"]
+        image_size, num_classes = 224, 1000
+    args = build_main_parser().parse_args(argv)
+
+    if bench.model == "resnet18":
+        from noisynet_amd.models.resnet import ResNet18
+        model = ResNet18(args)
+        # CIFAR-10 head
+        from noisynet_amd.hardware_model import NoisyLinear
+        model.fc = NoisyLinear(512, num_classes, bias=True, num_bits=0,
+                               num_bits_weight=args.q_w, noise=args.n_w,
+                               test_noise=args.n_w_test,
+                               stochastic=args.stochastic)
+    elif bench.model == "mobilenet_v2":
+        from noisynet_amd.models.mobilenet import mobilenet_v2
+        model = mobilenet_v2(args)
+    else:
+        from noisynet_amd.timm.models import create_model
+        model = create_model("efficientnet_b0", num_classes=num_classes)
+        model.args = args
+
+    model = model.to(device)
+    if dtype is torch.bfloat16:
+        model = model.bfloat16()
+        for m in model.modules():
+            if isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+                m.float()
+    if device.type == "cuda":
+        model = model.to(memory_format=torch.channels_last)
+    return model, args, image_size, num_classes
 
 
 def main():
@@ -75,50 +126,60 @@ def main():
         device = torch.device("cpu")
 
     dtype = torch.bfloat16 if bench.dtype == "bf16" else torch.float32
-    args = flagship_args(bench)
 
-    model = Net(args)
-    utils.init_model(model, args)
-    model = model.to(device)
-    if dtype is torch.bfloat16:
-        model = model.bfloat16()
-        # BN statistics stay fp32 for stability
-        for m in model.modules():
-            if isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
-                m.float()
-    if device.type == "cuda":
-        model = model.to(memory_format=torch.channels_last)
+    if bench.model == "noisynet":
+        args = flagship_args(bench)
+        model = Net(args)
+        utils.init_model(model, args)
+        model = model.to(device)
+        if dtype is torch.bfloat16:
+            model = model.bfloat16()
+            # BN statistics stay fp32 for stability
+            for m in model.modules():
+                if isinstance(m, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+                    m.float()
+        if device.type == "cuda":
+            model = model.to(memory_format=torch.channels_last)
+        image_size, num_classes = 32, 10
+    else:
+        model, args, image_size, num_classes = build_secondary(bench, device,
+                                                               dtype)
 
     dp = dist_mod.DataParallel(model) if distributed else None
 
-    # synthetic 4-bit CIFAR-shaped data, resident on device
+    # synthetic 4-bit-grid data, resident on device
     g = torch.Generator().manual_seed(bench.seed + rank)
-    data = (torch.randint(0, 16, (bench.batch, 3, 32, 32), generator=g)
-            .to(device=device, dtype=dtype) / 15.0)
+    data = (torch.randint(0, 16, (bench.batch, 3, image_size, image_size),
+                          generator=g).to(device=device, dtype=dtype) / 15.0)
     if device.type == "cuda":
         data = data.contiguous(memory_format=torch.channels_last)
-    labels = torch.randint(0, 10, (bench.batch,), generator=g).to(device)
+    labels = torch.randint(0, num_classes, (bench.batch,), generator=g).to(device)
 
-    param_groups = [
-        {"params": model.conv1.parameters(), "weight_decay": args.L2_1,
-         "lr": args.LR, "clamp": (-args.w_max1, args.w_max1)},
-        {"params": model.conv2.parameters(), "weight_decay": args.L2_2,
-         "lr": args.LR},
-        {"params": model.linear1.parameters(), "weight_decay": args.L2_3,
-         "lr": args.LR},
-        {"params": model.linear2.parameters(), "weight_decay": args.L2_4,
-         "lr": args.LR},
-        {"params": [p for m in (model.bn1, model.bn2, model.bn3, model.bn4)
-                    for p in m.parameters()], "weight_decay": 0.0,
-         "lr": args.LR},
-    ]
-    optimizer = native_optim.SGD(param_groups, lr=args.LR,
-                                 momentum=args.momentum, nesterov=True)
+    if bench.model == "noisynet":
+        param_groups = [
+            {"params": model.conv1.parameters(), "weight_decay": args.L2_1,
+             "lr": args.LR, "clamp": (-args.w_max1, args.w_max1)},
+            {"params": model.conv2.parameters(), "weight_decay": args.L2_2,
+             "lr": args.LR},
+            {"params": model.linear1.parameters(), "weight_decay": args.L2_3,
+             "lr": args.LR},
+            {"params": model.linear2.parameters(), "weight_decay": args.L2_4,
+             "lr": args.LR},
+            {"params": [p for m in (model.bn1, model.bn2, model.bn3, model.bn4)
+                        for p in m.parameters()], "weight_decay": 0.0,
+             "lr": args.LR},
+        ]
+        optimizer = native_optim.SGD(param_groups, lr=args.LR,
+                                     momentum=args.momentum, nesterov=True)
+    else:
+        optimizer = native_optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                                     weight_decay=1e-4, nesterov=False)
     criterion = torch.nn.CrossEntropyLoss()
 
     def step(i):
         model.train()
-        out = model(data, 0, i)
+        out = model(data, 0, i) if bench.model != "efficientnet_b0" \
+            else model(data)
         loss = criterion(out.float(), labels)
         optimizer.zero_grad(set_to_none=False)
         loss.backward()
@@ -175,12 +236,16 @@ def main():
             "dtype": bench.dtype,
             "data": "synthetic",
             "config": {
-                "model": "noisynet-4layer-cifar",
+                "model": ("noisynet-4layer-cifar" if bench.model == "noisynet"
+                          else bench.model),
                 "global_batch": n_gpus * bench.batch,
                 "seq_len": None,
-                "image_size": 32,
-                "q_a": 4, "q_w": 0, "current_nA": 0 if bench.no_noise else 1,
-                "act_max": 5,
+                "image_size": image_size,
+                "q_a": 4,
+                "q_w": 4 if bench.model == "resnet18" else 0,
+                "current_nA": (0 if (bench.no_noise or bench.model != "noisynet")
+                               else 1),
+                "act_max": 5 if bench.model == "noisynet" else 0,
                 "parallelism": "dp%d" % n_gpus,
             },
         }))

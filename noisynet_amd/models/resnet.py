@@ -233,7 +233,8 @@ class ResNet(nn.Module):
         x = self.layer3(x)
         x = self.layer4(x)
 
-        x = ops.avgpool_nhwc(x, 7, 1, 0)
+        k = min(7, x.size(2))  # CIFAR-sized inputs reach layer4 at < 7x7
+        x = ops.avgpool_nhwc(x, k, 1, 0)
         x = x.reshape(x.size(0), -1)
 
         if args.distort_pre_act and self.offset > 0:
