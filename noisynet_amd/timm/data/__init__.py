@@ -1,3 +1,5 @@
+from .auto_augment import (AutoAugment, RandAugment,  # noqa: F401
+                           auto_augment_transform, rand_augment_transform)
 from .config import resolve_data_config  # noqa: F401
 from .dataset import Dataset, SyntheticImageDataset  # noqa: F401
 from .distributed_sampler import OrderedDistributedSampler  # noqa: F401

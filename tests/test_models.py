@@ -275,3 +275,35 @@ def test_chip_mnist_driver(tmp_path, monkeypatch):
     res = mnist.main(['--epochs', '1', '--n_train', '500', '--n_test', '100',
                       '--L3', '0.0'])
     assert res
+
+
+def test_auto_augment_and_randaugment():
+    from noisynet_amd.timm.data import (AutoAugment, rand_augment_transform)
+    img = np.random.randint(0, 255, (3, 32, 32), dtype=np.uint8)
+    aa = AutoAugment()
+    out = aa(img.copy())
+    assert out.shape == img.shape and out.dtype == np.uint8
+    ra = rand_augment_transform('rand-m9-mstd0.5-n2')
+    out = ra(img.copy())
+    assert out.shape == img.shape and out.dtype == np.uint8
+    # individual ops sanity
+    from noisynet_amd.timm.data.auto_augment import (equalize, posterize,
+                                                     rotate, solarize)
+    assert solarize(img, 128).shape == img.shape
+    assert posterize(img, 4).max() <= 255
+    assert rotate(img, 15.0).shape == img.shape
+    assert equalize(img).shape == img.shape
+
+
+def test_median_pool_and_feature_hooks():
+    from noisynet_amd.timm.models.median_pool import MedianPool2d
+    x = torch.rand(1, 3, 8, 8)
+    mp = MedianPool2d(3, 1, 1)
+    assert mp(x).shape == x.shape
+
+    from noisynet_amd.timm.models.feature_hooks import FeatureHooks
+    net = torch.nn.Sequential(torch.nn.Conv2d(3, 4, 3), torch.nn.ReLU())
+    hooks = FeatureHooks([{'name': '0'}], net.named_modules())
+    y = net(x)
+    feats = hooks.get_output(x.device)
+    assert len(feats) == 1 and feats[0].shape[1] == 4
