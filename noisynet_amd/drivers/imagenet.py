@@ -212,6 +212,31 @@ def main(argv=None):
     optimizer = native_optim.SGD(params, lr=args.lr, momentum=args.momentum,
                                  weight_decay=args.weight_decay, nesterov=False)
 
+    # --var_name eval sweep (reference main.py:1011-1087; gated dead there,
+    # functional here): re-validate a restored model over a hyperparam grid.
+    if args.var_name and args.resume:
+        grids = {
+            'pctl': [99.99, 99.992, 99.994, 99.996, 99.998, 99.999, 99.9995],
+            'q_scale': [0.87, 0.88, 0.89, 0.90, 0.91, 0.92, 0.93, 0.94,
+                        0.95, 0.96],
+            'selected_weights': [1, 2, 5, 10],
+        }
+        var_list = grids.get(args.var_name, [getattr(args, args.var_name)])
+        load_from_checkpoint(args, model, optimizer, device)
+        total_list = []
+        for var in var_list:
+            setattr(args, args.var_name, var)
+            accs = [validate(val_loader, model, args)
+                    for _ in range(args.num_sims)]
+            total_list.append((float(np.mean(accs)), float(np.min(accs)),
+                               float(np.max(accs))))
+            print('{:d} runs:  {} {} {:.2f} ({:.2f}/{:.2f})'.format(
+                args.num_sims, args.var_name, var, *total_list[-1]))
+        for var, (mean_, min_, max_) in zip(var_list, total_list):
+            print('{} {} acc {:.2f} ({:.2f}/{:.2f})'.format(
+                args.var_name, var, mean_, min_, max_))
+        return total_list
+
     start_epoch, best_acc = 0, 0.0
     if args.resume:
         start_epoch, best_acc = load_from_checkpoint(args, model, optimizer,
