@@ -81,33 +81,43 @@ __global__ void dwconv_dgrad_kernel(const T* __restrict__ gy,
   }
 }
 
-template <typename T>
+// Each thread owns ONE channel and a slice of the output pixels,
+// accumulating all R*S taps in registers; one atomicAdd per tap per
+// thread-slice (instead of one per OUTPUT ELEMENT, which serialized on
+// the same dw[c][r][s] word ~N*OH*OW deep).
+template <typename T, int MAXTAPS>
 __global__ void dwconv_wgrad_kernel(const T* __restrict__ gy,
                                     const T* __restrict__ x,
-                                    float* __restrict__ dw, int64_t n_out,
-                                    int C, int H, int W, int OH, int OW, int R,
-                                    int S, int stride, int pad) {
-  // each thread accumulates its output element's contribution to each tap
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    int64_t t = i / C;
+                                    float* __restrict__ dw, int C,
+                                    int H, int W, int OH, int OW, int R,
+                                    int S, int stride, int pad,
+                                    int64_t npix) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float acc[MAXTAPS];
+  int taps = R * S;
+#pragma unroll
+  for (int t = 0; t < MAXTAPS; ++t) acc[t] = 0.0f;
+  // pixel slice: grid.y slices over N*OH*OW
+  for (int64_t pix = blockIdx.y; pix < npix; pix += gridDim.y) {
+    int64_t t = pix;
     int ow = (int)(t % OW);
     t /= OW;
     int oh = (int)(t % OH);
     int64_t nb = t / OH;
-    float g = to_f32(gy[i]);
+    float g = to_f32(gy[pix * C + c]);
     for (int r = 0; r < R; ++r) {
       int ih = oh * stride - pad + r;
       if (ih < 0 || ih >= H) continue;
       for (int s = 0; s < S; ++s) {
         int iw = ow * stride - pad + s;
         if (iw < 0 || iw >= W) continue;
-        float xv = to_f32(x[((nb * H + ih) * W + iw) * C + c]);
-        atomicAdd(&dw[((int64_t)c * R + r) * S + s], g * xv);
+        acc[r * S + s] += g * to_f32(x[((nb * H + ih) * W + iw) * C + c]);
       }
     }
   }
+  for (int t = 0; t < taps; ++t)
+    if (acc[t] != 0.0f) atomicAdd(&dw[(int64_t)c * taps + t], acc[t]);
 }
 
 template <typename scalar_t> struct DevT { using type = scalar_t; };
@@ -173,15 +183,32 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   int N = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2), W = (int)x.size(3);
   int OH = (int)gy.size(2), OW = (int)gy.size(3);
   auto dw_f = torch::zeros({C, (int)R, (int)S}, x.options().dtype(torch::kFloat32));
-  int64_t n_out = (int64_t)N * C * OH * OW;
-  int blocks = (int)std::min<int64_t>((n_out + kBlock - 1) / kBlock, 8192);
+  int64_t npix = (int64_t)N * OH * OW;
+  int cblocks = (C + kBlock - 1) / kBlock;
+  int mslices = (int)std::min<int64_t>(npix,
+      std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
+  TORCH_CHECK(R * S <= 64, "dwconv_wgrad: filter too large");
   NN_DISPATCH(gy.scalar_type(), "dwconv_wgrad", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((dwconv_wgrad_kernel<T>), dim3(blocks), dim3(kBlock), 0,
-                       c10::hip::getCurrentHIPStream(),
-                       (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
-                       dw_f.data_ptr<float>(), n_out, C, H, W, OH, OW, (int)R,
-                       (int)S, (int)stride, (int)pad);
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (R * S <= 9)
+      hipLaunchKernelGGL((dwconv_wgrad_kernel<T, 9>), dim3(cblocks, mslices),
+                         dim3(kBlock), 0, stream, (const T*)gy.data_ptr(),
+                         (const T*)x.data_ptr(), dw_f.data_ptr<float>(), C, H,
+                         W, OH, OW, (int)R, (int)S, (int)stride, (int)pad,
+                         npix);
+    else if (R * S <= 25)
+      hipLaunchKernelGGL((dwconv_wgrad_kernel<T, 25>), dim3(cblocks, mslices),
+                         dim3(kBlock), 0, stream, (const T*)gy.data_ptr(),
+                         (const T*)x.data_ptr(), dw_f.data_ptr<float>(), C, H,
+                         W, OH, OW, (int)R, (int)S, (int)stride, (int)pad,
+                         npix);
+    else
+      hipLaunchKernelGGL((dwconv_wgrad_kernel<T, 64>), dim3(cblocks, mslices),
+                         dim3(kBlock), 0, stream, (const T*)gy.data_ptr(),
+                         (const T*)x.data_ptr(), dw_f.data_ptr<float>(), C, H,
+                         W, OH, OW, (int)R, (int)S, (int)stride, (int)pad,
+                         npix);
   });
   HIP_CHECK_LAST();
   return dw_f.view({C, 1, (int)R, (int)S}).to(x.scalar_type());
