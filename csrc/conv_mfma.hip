@@ -1288,3 +1288,90 @@ std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
   return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,
                              sigma_mode, factor, seed, telem, /*want_y=*/false);
 }
+
+// ===========================================================================
+// im2col materialization + GEMM wgrad.
+//
+// Wgrad's contraction runs over N*OH*OW with a scattered per-tap input
+// gather; on 288 GB HBM it is cheaper to materialize the im2col matrix
+// once ([M, R*S*C] padded to a multiple of 8 columns, one coalesced write
+// pass) and run the wgrad as a dense GEMM with fully vectorized staging
+// than to re-gather per K-chunk. Each thread copies one C-span per
+// (pixel, tap), so the row/tap decode happens once per span, not per
+// element.
+// ===========================================================================
+
+namespace {
+
+template <typename T>
+__global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
+                              ConvGeom g, int c_pad, int64_t nspans) {
+  // span = (m, r, s): copies C channels (zero-padding to c_pad)
+  int taps = g.R * g.S;
+  for (int64_t span = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       span < nspans; span += (int64_t)gridDim.x * blockDim.x) {
+    int rs = (int)(span % taps);
+    int64_t m = span / taps;
+    int s = rs % g.S;
+    int r = rs / g.S;
+    int64_t t = m;
+    int ow = (int)(t % g.OW);
+    t /= g.OW;
+    int oh = (int)(t % g.OH);
+    int n = (int)t;
+    int ih = oh * g.stride - g.pad + r;
+    int iw = ow * g.stride - g.pad + s;
+    T* dst = out + (m * taps + rs) * (int64_t)c_pad;
+    if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
+      const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
+      int c = 0;
+      if (sizeof(T) == 2 && (g.C & 7) == 0) {
+        for (; c + 8 <= g.C; c += 8)
+          *(bf16x8*)(dst + c) = *(const bf16x8*)(src + c);
+      }
+      for (; c < g.C; ++c) dst[c] = src[c];
+      for (; c < c_pad; ++c) dst[c] = from_f32<T>(0.0f);
+    } else {
+      for (int c = 0; c < c_pad; ++c) dst[c] = from_f32<T>(0.0f);
+    }
+  }
+}
+
+}  // namespace
+
+// conv wgrad through materialized im2col (called from Python when the
+// buffer fits; falls back to conv_wgrad otherwise).
+torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
+                                int64_t stride, int64_t pad, int64_t R,
+                                int64_t S) {
+  check_cl(gy, "conv_wgrad_im2col gy");
+  check_cl(x, "conv_wgrad_im2col x");
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)gy.size(1), (int)R, (int)S,
+                     (int)stride, (int)pad);
+  int taps = g.R * g.S;
+  // pad the flattened (r,s,c) ... per-tap c_pad keeps spans aligned
+  int c_pad = (g.C + 7) & ~7;
+  int cols = taps * c_pad;
+  auto col = torch::empty({g.M, cols}, x.options());
+  int64_t nspans = g.M * taps;
+  int blocks = (int)std::min<int64_t>((nspans + 255) / 256, 8192);
+  NN_DISPATCH(x.scalar_type(), "im2col", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((im2col_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)x.data_ptr(), (T*)col.data_ptr(), g, c_pad,
+                       nspans);
+  });
+  HIP_CHECK_LAST();
+
+  // dw[k, taps*c_pad] = gy^T @ col via the dense transposed-staging wgrad
+  auto gy2 = gy.permute({0, 2, 3, 1}).reshape({g.M, g.K});  // raw view, free
+  auto dw_flat = linear_wgrad(gy2, col);                    // [K, cols]
+  // un-pad: [K, R, S, c_pad] -> [K, R, S, C] -> logical [K, C, R, S] cl
+  auto dw = dw_flat.view({g.K, g.R, g.S, c_pad})
+                .narrow(3, 0, g.C)
+                .permute({0, 3, 1, 2})
+                .contiguous(at::MemoryFormat::ChannelsLast);
+  return dw;
+}

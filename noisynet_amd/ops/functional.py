@@ -126,8 +126,17 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
 
 def _conv_wgrad_raw(g, x, stride, padding, w_shape):
     if use_native(g, x):
-        return ext().conv_wgrad(_nhwc(g), _nhwc(x), stride, padding,
-                                w_shape[2], w_shape[3])
+        R, S = w_shape[2], w_shape[3]
+        if R * S > 1:
+            # im2col-GEMM wgrad when the buffer is affordable (<2 GB):
+            # one coalesced materialization pass beats per-K-chunk gather
+            c_pad = (x.shape[1] + 7) // 8 * 8
+            m = g.shape[0] * g.shape[2] * g.shape[3]
+            col_bytes = m * R * S * c_pad * x.element_size()
+            if col_bytes <= 2 << 30:
+                return ext().conv_wgrad_im2col(_nhwc(g), _nhwc(x), stride,
+                                               padding, R, S)
+        return ext().conv_wgrad(_nhwc(g), _nhwc(x), stride, padding, R, S)
     return torch.nn.grad.conv2d_weight(x, w_shape, g, stride, padding)
 
 
