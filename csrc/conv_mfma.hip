@@ -1318,6 +1318,7 @@ __global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
     int ow = (int)(t % g.OW);
     t /= g.OW;
     int oh = (int)(t % g.OH);
+    t /= g.OH;
     int n = (int)t;
     int ih = oh * g.stride - g.pad + r;
     int iw = ow * g.stride - g.pad + s;
@@ -1338,6 +1339,30 @@ __global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
 }
 
 }  // namespace
+
+// exposed for testing: the materialization pass alone
+torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
+                                 int64_t pad, int64_t R, int64_t S) {
+  check_cl(x, "im2col x");
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)K, (int)R, (int)S, (int)stride,
+                     (int)pad);
+  int taps = g.R * g.S;
+  int c_pad = (g.C + 7) & ~7;
+  int cols = taps * c_pad;
+  auto col = torch::empty({g.M, cols}, x.options());
+  int64_t nspans = g.M * taps;
+  int blocks = (int)std::min<int64_t>((nspans + 255) / 256, 8192);
+  NN_DISPATCH(x.scalar_type(), "im2col", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((im2col_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)x.data_ptr(), (T*)col.data_ptr(), g, c_pad,
+                       nspans);
+  });
+  HIP_CHECK_LAST();
+  return col;
+}
 
 // conv wgrad through materialized im2col (called from Python when the
 // buffer fits; falls back to conv_wgrad otherwise).
