@@ -1168,7 +1168,7 @@ inline bool patch_eligible(const ConvGeom& g) {
   int c_pad = (g.C + 7) & ~7;
   int wp = g.W + 2 * g.pad;
   size_t patch_bytes = (size_t)g.H * wp * c_pad * 2 + 64;
-  return g.R * g.S > 1 && patch_bytes <= 64 * 1024;
+  return g.R * g.S > 1 && g.C > 8 && patch_bytes <= 64 * 1024;
 }
 
 // pad the raw-[K,R,S,C] weight view to [K,R,S*C_pad] (tiny tensors)
@@ -1323,17 +1323,30 @@ __global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
     int ih = oh * g.stride - g.pad + r;
     int iw = ow * g.stride - g.pad + s;
     T* dst = out + (m * taps + rs) * (int64_t)c_pad;
-    if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
-      const T* src = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
-      int c = 0;
-      if (sizeof(T) == 2 && (g.C & 7) == 0) {
-        for (; c + 8 <= g.C; c += 8)
-          *(bf16x8*)(dst + c) = *(const bf16x8*)(src + c);
+    bool inb = (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W);
+    const T* src = inb ? x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C
+                       : nullptr;
+    if (sizeof(T) == 2) {
+      // destination rows are 16B-aligned (c_pad % 8 == 0): build each
+      // 8-element chunk in registers and emit one vector store
+      for (int c0 = 0; c0 < c_pad; c0 += 8) {
+        bf16x8 v;
+        if (inb && (g.C & 7) == 0 && c0 + 8 <= g.C) {
+          v = *(const bf16x8*)(src + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            int c = c0 + j;
+            ((bf16*)&v)[j] = (inb && c < g.C)
+                ? __float2bfloat16(to_f32(src[c]))
+                : __float2bfloat16(0.0f);
+          }
+        }
+        *(bf16x8*)(dst + c0) = v;
       }
-      for (; c < g.C; ++c) dst[c] = src[c];
-      for (; c < c_pad; ++c) dst[c] = from_f32<T>(0.0f);
     } else {
-      for (int c = 0; c < c_pad; ++c) dst[c] = from_f32<T>(0.0f);
+      for (int c = 0; c < c_pad; ++c)
+        dst[c] = (inb && c < g.C) ? src[c] : from_f32<T>(0.0f);
     }
   }
 }
