@@ -33,11 +33,107 @@ constexpr int kBlock = 256;
 constexpr int BM = 64;
 constexpr int BN = 64;
 constexpr int BK = 32;
-constexpr int LDS_STRIDE = 80;  // bytes per tile row: 64B data + 16B skew
 
 using bf16 = __hip_bfloat16;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f16x8 = __attribute__((ext_vector_type(8))) _Float16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
+using f32x8 = __attribute__((ext_vector_type(8))) float;
+
+// ---------------------------------------------------------------------------
+// Compute-type traits: one MFMA tile abstraction per input dtype.
+//   bf16 -> v_mfma_f32_16x16x32_bf16 (8 bf16/lane fragments)
+//   fp16 -> v_mfma_f32_16x16x32_f16
+//   fp32 -> v_mfma_f32_16x16x4_f32 x8 (EXACT f32 at the f32 vector rate --
+//           gfx950 has no xf32; cdna_hip_programming.md §3). The f32 LDS
+//           image stores k at permuted position (k&3)*8 + (k>>2) so each
+//           lane's 8 needed elements (k = q + 4*kk, q = lane>>4) are the
+//           contiguous span [q*8, q*8+8) -> one 32-B read.
+// All three share the C/D fragment map (dtype-independent on gfx950), so
+// accumulators and epilogues are identical.
+// ---------------------------------------------------------------------------
+
+template <typename T> struct Mma;
+
+template <> struct Mma<bf16> {
+  using frag = bf16x8;
+  static constexpr int STRIDE = 80;   // 32 el * 2 B + 16 B skew
+  static DEV_INLINE void store8(char* lds, int row, int k0, const float* v) {
+    bf16 tmp[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) tmp[j] = __float2bfloat16(v[j]);
+    *(bf16x8*)(lds + row * STRIDE + k0 * 2) = *(bf16x8*)tmp;
+  }
+  static DEV_INLINE frag load(char* lds, int row, int lane) {
+    return *(frag*)(lds + row * STRIDE + (lane >> 4) * 16);
+  }
+  static DEV_INLINE void mma(const frag& a, const frag& b, f32x4& acc) {
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  static DEV_INLINE void store1(char* lds, int row, int k, float v) {
+    *(bf16*)(lds + row * STRIDE + k * 2) = __float2bfloat16(v);
+  }
+  // read the lane's fragment from a RAW 32-element span (no tile rows)
+  static DEV_INLINE frag load_span(const char* base, int lane) {
+    return *(frag*)(base + (lane >> 4) * 16);
+  }
+};
+
+template <> struct Mma<_Float16> {
+  using frag = f16x8;
+  static constexpr int STRIDE = 80;
+  static DEV_INLINE void store8(char* lds, int row, int k0, const float* v) {
+    _Float16 tmp[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) tmp[j] = (_Float16)v[j];
+    *(f16x8*)(lds + row * STRIDE + k0 * 2) = *(f16x8*)tmp;
+  }
+  static DEV_INLINE frag load(char* lds, int row, int lane) {
+    return *(frag*)(lds + row * STRIDE + (lane >> 4) * 16);
+  }
+  static DEV_INLINE void mma(const frag& a, const frag& b, f32x4& acc) {
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, acc, 0, 0, 0);
+  }
+  static DEV_INLINE void store1(char* lds, int row, int k, float v) {
+    *(_Float16*)(lds + row * STRIDE + k * 2) = (_Float16)v;
+  }
+  static DEV_INLINE frag load_span(const char* base, int lane) {
+    return *(frag*)(base + (lane >> 4) * 16);
+  }
+};
+
+template <> struct Mma<float> {
+  using frag = f32x8;
+  static constexpr int STRIDE = 144;  // 32 el * 4 B + 16 B skew
+  static DEV_INLINE void store8(char* lds, int row, int k0, const float* v) {
+    float* base = (float*)(lds + row * STRIDE);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = k0 + j;
+      base[((k & 3) << 3) + (k >> 2)] = v[j];
+    }
+  }
+  static DEV_INLINE frag load(char* lds, int row, int lane) {
+    return *(frag*)(lds + row * STRIDE + (lane >> 4) * 32);
+  }
+  static DEV_INLINE void mma(const frag& a, const frag& b, f32x4& acc) {
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a[kk], b[kk], acc, 0, 0, 0);
+  }
+  static DEV_INLINE void store1(char* lds, int row, int k, float v) {
+    ((float*)(lds + row * STRIDE))[((k & 3) << 3) + (k >> 2)] = v;
+  }
+  static DEV_INLINE frag load_span(const char* base, int lane) {
+    // raw span is k-linear: gather the lane's strided k = q + 4*kk
+    frag f;
+    int q = lane >> 4;
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk)
+      f[kk] = ((const float*)base)[q + 4 * kk];
+    return f;
+  }
+};
 
 DEV_INLINE float atomic_max_f32(float* addr, float val) {
   // monotone int mapping for IEEE floats
@@ -66,10 +162,6 @@ struct ConvGeom {
 //   row = tid >> 2, seg = tid & 3  (4 segs * 8 el * 2B = 64B per row)
 // --------------------------------------------------------------------------
 
-DEV_INLINE void* lds_at(char* base, int row, int byte_in_row) {
-  return base + row * LDS_STRIDE + byte_in_row;
-}
-
 // stage activation tile: rows are output pixels m0+row, cols are the
 // contraction slice ck+seg*8 .. +8 (input channels of tap (r,s), or the
 // flattened (r,s,c) index when g.flat). bf16 inputs with C%8==0 take one
@@ -80,7 +172,7 @@ DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
   int row = threadIdx.x >> 2;
   int seg = threadIdx.x & 3;
   int64_t m = m0 + row;
-  bf16 vals[8];
+  float vals[8];
   bool zero = m >= g.M;
   int n = 0, oh = 0, ow = 0;
   if (!zero) {
@@ -93,7 +185,7 @@ DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
   if (g.flat) {
     if (zero) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+      for (int j = 0; j < 8; ++j) vals[j] = 0.0f;
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -107,10 +199,10 @@ DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
         float v = 0.0f;
         if (rr < g.R && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
           v = to_f32(x[(((int64_t)n * g.H + ih) * g.W + iw) * g.C + c]);
-        vals[j] = __float2bfloat16(v);
+        vals[j] = v;
       }
     }
-    *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+    Mma<T>::store8(lds, row, seg * 8, vals);
     return;
   }
   int ih = oh * g.stride - g.pad + r;
@@ -118,20 +210,20 @@ DEV_INLINE void stage_x_tap(char* lds, const T* __restrict__ x,
   if (!zero && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
     const T* px = x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C;
     if (sizeof(T) == 2 && (g.C & 7) == 0 && c0 + 8 <= g.C) {
-      *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(px + c0);
+      *(bf16x8*)(lds + row * Mma<T>::STRIDE + seg * 16) =
+          *(const bf16x8*)(px + c0);
       return;
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = c0 + j;
-      vals[j] = (c < g.C) ? __float2bfloat16(to_f32(px[c]))
-                          : __float2bfloat16(0.0f);
+      vals[j] = (c < g.C) ? to_f32(px[c]) : 0.0f;
     }
   } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+    for (int j = 0; j < 8; ++j) vals[j] = 0.0f;
   }
-  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+  Mma<T>::store8(lds, row, seg * 8, vals);
 }
 
 // stage weight tile for tap (r,s): rows are output channels n0+row,
@@ -143,7 +235,7 @@ DEV_INLINE void stage_w_tap(char* lds, const T* __restrict__ w,
   int row = threadIdx.x >> 2;
   int seg = threadIdx.x & 3;
   int k = n0 + row;
-  bf16 vals[8];
+  float vals[8];
   int span = g.flat ? g.R * g.S * g.C : g.C;
   if (k < g.K) {
     const T* pw = g.flat ? (w + (int64_t)k * span)
@@ -157,21 +249,13 @@ DEV_INLINE void stage_w_tap(char* lds, const T* __restrict__ w,
         v = fabsf(v);
         if (SIGMA_MODE == 2) v = v * v + v;
       }
-      vals[j] = __float2bfloat16(v);
+      vals[j] = v;
     }
   } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+    for (int j = 0; j < 8; ++j) vals[j] = 0.0f;
   }
-  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
-}
-
-// read one 16x16x32 fragment (8 bf16) for this lane from an LDS tile
-DEV_INLINE bf16x8 frag_from_lds(char* lds, int row_base) {
-  int lane = threadIdx.x & (WAVE - 1);
-  int row = row_base + (lane & 15);
-  int byte = (lane >> 4) * 16;
-  return *(bf16x8*)lds_at(lds, row, byte);
+  Mma<T>::store8(lds, row, seg * 8, vals);
 }
 
 // --------------------------------------------------------------------------
@@ -195,11 +279,12 @@ void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
   int n0 = blockIdx.x * BN;
   int64_t m0 = (int64_t)blockIdx.y * BM;
 
+  constexpr int STR = Mma<T>::STRIDE;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* a_lds = smem;                                   // BM rows
-  char* b_lds = smem + BM * LDS_STRIDE;                 // BN rows (wq or wraw)
-  char* c_lds = smem + (BM + BN) * LDS_STRIDE;          // BN rows (wraw sigma)
-  char* d_lds = smem + (BM + 2 * BN) * LDS_STRIDE;      // BN rows (|w| telem)
+  char* b_lds = smem + BM * STR;                        // BN rows (wq or wraw)
+  char* c_lds = smem + (BM + BN) * STR;                 // BN rows (wraw sigma)
+  char* d_lds = smem + (BM + 2 * BN) * STR;             // BN rows (|w| telem)
 
   int wid = threadIdx.x / WAVE;
   int wm = wid >> 1, wn = wid & 1;
@@ -224,23 +309,21 @@ void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
         __syncthreads();
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm) {
-          bf16x8 a = frag_from_lds(a_lds, wm * 32 + fm * 16);
+          auto a = Mma<T>::load(a_lds, wm * 32 + fm * 16 + (lane & 15), lane);
 #pragma unroll
           for (int fn = 0; fn < 2; ++fn) {
+            int brow = wn * 32 + fn * 16 + (lane & 15);
             if (WANT_Y) {
-              bf16x8 b = frag_from_lds(b_lds, wn * 32 + fn * 16);
-              acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a, b, acc[fm][fn], 0, 0, 0);
+              auto b = Mma<T>::load(b_lds, brow, lane);
+              Mma<T>::mma(a, b, acc[fm][fn]);
             }
             if (SIGMA_MODE > 0) {
-              bf16x8 bs = frag_from_lds(c_lds, wn * 32 + fn * 16);
-              sacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a, bs, sacc[fm][fn], 0, 0, 0);
+              auto bs = Mma<T>::load(c_lds, brow, lane);
+              Mma<T>::mma(a, bs, sacc[fm][fn]);
             }
             if (TELEM && SIGMA_MODE == 2) {
-              bf16x8 bt = frag_from_lds(d_lds, wn * 32 + fn * 16);
-              tacc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a, bt, tacc[fm][fn], 0, 0, 0);
+              auto bt = Mma<T>::load(d_lds, brow, lane);
+              Mma<T>::mma(a, bt, tacc[fm][fn]);
             }
           }
         }
@@ -316,7 +399,7 @@ DEV_INLINE void stage_g_tap_dgrad(char* lds, const T* __restrict__ gy,
   int seg = threadIdx.x & 3;
   int64_t m = m0 + row;
   int64_t MI = (int64_t)g.N * g.H * g.W;
-  bf16 vals[8];
+  float vals[8];
   bool ok = false;
   const T* pg = nullptr;
   if (m < MI) {
@@ -336,16 +419,16 @@ DEV_INLINE void stage_g_tap_dgrad(char* lds, const T* __restrict__ gy,
   }
   int k0 = kk + seg * 8;
   if (ok && sizeof(T) == 2 && (g.K & 7) == 0 && k0 + 8 <= g.K) {
-    *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(pg + k0);
+    *(bf16x8*)(lds + row * Mma<T>::STRIDE + seg * 16) =
+        *(const bf16x8*)(pg + k0);
     return;
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     int k = k0 + j;
-    vals[j] = (ok && k < g.K) ? __float2bfloat16(to_f32(pg[k]))
-                              : __float2bfloat16(0.0f);
+    vals[j] = (ok && k < g.K) ? to_f32(pg[k]) : 0.0f;
   }
-  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+  Mma<T>::store8(lds, row, seg * 8, vals);
 }
 
 template <typename T>
@@ -355,25 +438,25 @@ DEV_INLINE void stage_wt_tap(char* lds, const T* __restrict__ wt,
   int row = threadIdx.x >> 2;
   int seg = threadIdx.x & 3;
   int c = n0 + row;
-  bf16 vals[8];
+  float vals[8];
   if (c < g.C) {
     const T* pw = wt + (((int64_t)r * g.S + s) * g.C + c) * g.K;
     int k0 = kk + seg * 8;
     if (sizeof(T) == 2 && (g.K & 7) == 0 && k0 + 8 <= g.K) {
-      *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(pw + k0);
+      *(bf16x8*)(lds + row * Mma<T>::STRIDE + seg * 16) =
+          *(const bf16x8*)(pw + k0);
       return;
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int k = k0 + j;
-      vals[j] = (k < g.K) ? __float2bfloat16(to_f32(pw[k]))
-                          : __float2bfloat16(0.0f);
+      vals[j] = (k < g.K) ? to_f32(pw[k]) : 0.0f;
     }
   } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+    for (int j = 0; j < 8; ++j) vals[j] = 0.0f;
   }
-  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+  Mma<T>::store8(lds, row, seg * 8, vals);
 }
 
 template <typename T>
@@ -384,9 +467,10 @@ void conv_dgrad_kernel(const T* __restrict__ gy, const T* __restrict__ wt,
   int64_t m0 = (int64_t)blockIdx.y * BM;  // over input pixels
   int64_t MI = (int64_t)g.N * g.H * g.W;
 
+  constexpr int STR = Mma<T>::STRIDE;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* a_lds = smem;
-  char* b_lds = smem + BM * LDS_STRIDE;
+  char* b_lds = smem + BM * STR;
 
   int wid = threadIdx.x / WAVE;
   int wm = wid >> 1, wn = wid & 1;
@@ -401,12 +485,11 @@ void conv_dgrad_kernel(const T* __restrict__ gy, const T* __restrict__ wt,
         __syncthreads();
 #pragma unroll
         for (int fm = 0; fm < 2; ++fm) {
-          bf16x8 a = frag_from_lds(a_lds, wm * 32 + fm * 16);
+          auto a = Mma<T>::load(a_lds, wm * 32 + fm * 16 + (lane & 15), lane);
 #pragma unroll
           for (int fn = 0; fn < 2; ++fn) {
-            bf16x8 b = frag_from_lds(b_lds, wn * 32 + fn * 16);
-            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a, b, acc[fm][fn], 0, 0, 0);
+            auto b = Mma<T>::load(b_lds, wn * 32 + fn * 16 + (lane & 15), lane);
+            Mma<T>::mma(a, b, acc[fm][fn]);
           }
         }
         __syncthreads();
@@ -440,8 +523,9 @@ DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
                                     const T* __restrict__ x, const ConvGeom g,
                                     int64_t m0, int k0, int c0, int r, int s) {
   // 256 threads load a [BK=32 m] x [64 col] slab of G and X each, writing
-  // transposed into LDS rows [col][m]. Thread: mi = tid&31, colseg = tid>>5
-  // (8 segs of 8 cols).
+  // transposed into LDS rows [col][m] through the compute-type store
+  // (which also applies the f32 k-permutation). Thread: mi = tid&31,
+  // colseg = tid>>5 (8 segs of 8 cols).
   int mi = threadIdx.x & 31;
   int colseg = threadIdx.x >> 5;
   int64_t m = m0 + mi;
@@ -465,7 +549,7 @@ DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
     for (int j = 0; j < 8; ++j) {
       int kcol = k0 + colseg * 8 + j;
       float gv = (mok && kcol < g.K) ? to_f32(pg[kcol]) : 0.0f;
-      *(bf16*)lds_at(g_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(gv);
+      Mma<T>::store1(g_lds, colseg * 8 + j, mi, gv);
       // flat column -> (r', s', c) decoded per element
       int fcol = c0 + colseg * 8 + j;
       float xv = 0.0f;
@@ -482,38 +566,27 @@ DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
         if (ih2 >= 0 && ih2 < g.H && iw2 >= 0 && iw2 < g.W)
           xv = to_f32(x[(((int64_t)n * g.H + ih2) * g.W + iw2) * g.C + c]);
       }
-      *(bf16*)lds_at(x_lds, colseg * 8 + j, mi * 2) = __float2bfloat16(xv);
+      Mma<T>::store1(x_lds, colseg * 8 + j, mi, xv);
     }
     return;
   }
-  bf16 gvals[8], xvals[8];
+  float gvals[8], xvals[8];
   int kc0 = k0 + colseg * 8;
-  if (mok && sizeof(T) == 2 && (g.K & 7) == 0 && kc0 + 8 <= g.K) {
-    *(bf16x8*)gvals = *(const bf16x8*)(pg + kc0);
-  } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int kcol = kc0 + j;
-      gvals[j] = (mok && kcol < g.K) ? __float2bfloat16(to_f32(pg[kcol]))
-                                     : __float2bfloat16(0.0f);
-    }
+  for (int j = 0; j < 8; ++j) {
+    int kcol = kc0 + j;
+    gvals[j] = (mok && kcol < g.K) ? to_f32(pg[kcol]) : 0.0f;
   }
   int cc0 = c0 + colseg * 8;
-  if (px != nullptr && sizeof(T) == 2 && (g.C & 7) == 0 && cc0 + 8 <= g.C) {
-    *(bf16x8*)xvals = *(const bf16x8*)(px + cc0);
-  } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int ccol = cc0 + j;
-      xvals[j] = (px != nullptr && ccol < g.C)
-                     ? __float2bfloat16(to_f32(px[ccol]))
-                     : __float2bfloat16(0.0f);
-    }
+  for (int j = 0; j < 8; ++j) {
+    int ccol = cc0 + j;
+    xvals[j] = (px != nullptr && ccol < g.C) ? to_f32(px[ccol]) : 0.0f;
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    *(bf16*)lds_at(g_lds, colseg * 8 + j, mi * 2) = gvals[j];
-    *(bf16*)lds_at(x_lds, colseg * 8 + j, mi * 2) = xvals[j];
+    Mma<T>::store1(g_lds, colseg * 8 + j, mi, gvals[j]);
+    Mma<T>::store1(x_lds, colseg * 8 + j, mi, xvals[j]);
   }
 }
 
@@ -530,9 +603,10 @@ void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
   int c0 = blockIdx.x * BN;
   int k0 = blockIdx.y * BM;
 
+  constexpr int STR = Mma<T>::STRIDE;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* g_lds = smem;                        // [64 k rows][32 m]
-  char* x_lds = smem + BM * LDS_STRIDE;      // [64 c rows][32 m]
+  char* x_lds = smem + BM * STR;             // [64 c rows][32 m]
 
   int wid = threadIdx.x / WAVE;
   int wm = wid >> 1, wn = wid & 1;
@@ -548,12 +622,11 @@ void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
     __syncthreads();
 #pragma unroll
     for (int fm = 0; fm < 2; ++fm) {
-      bf16x8 a = frag_from_lds(g_lds, wm * 32 + fm * 16);  // A: [k][m]
+      auto a = Mma<T>::load(g_lds, wm * 32 + fm * 16 + (lane & 15), lane);
 #pragma unroll
       for (int fn = 0; fn < 2; ++fn) {
-        bf16x8 b = frag_from_lds(x_lds, wn * 32 + fn * 16);  // B: [c][m]
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, b, acc[fm][fn], 0, 0, 0);
+        auto b = Mma<T>::load(x_lds, wn * 32 + fn * 16 + (lane & 15), lane);
+        Mma<T>::mma(a, b, acc[fm][fn]);
       }
     }
     __syncthreads();
@@ -616,10 +689,11 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w, int64_t stride,
   auto out = torch::empty({g.N, g.K, g.OH, g.OW},
                           x.options().memory_format(at::MemoryFormat::ChannelsLast));
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
-  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(x.scalar_type(),
                                   "conv_fwd", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + BN) * Mma<T>::STRIDE;
     hipLaunchKernelGGL((conv_fwd_kernel<T, true, 0, false, false>), grid,
                        dim3(kBlock), lds, c10::hip::getCurrentHIPStream(),
                        (const T*)x.data_ptr(), (const T*)w.data_ptr(),
@@ -652,10 +726,11 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
   tele[2] = -std::numeric_limits<float>::infinity();
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
-  size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(x.scalar_type(),
                                   "conv_fwd_fused", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + 3 * BN) * Mma<T>::STRIDE;
     auto stream = c10::hip::getCurrentHIPStream();
     const T* xp = (const T*)x.data_ptr();
     const T* wqp = (const T*)wq.data_ptr();
@@ -709,10 +784,11 @@ torch::Tensor conv_dgrad(torch::Tensor gy, torch::Tensor w, int64_t stride,
                          gy.options().memory_format(at::MemoryFormat::ChannelsLast));
   int64_t MI = (int64_t)g.N * g.H * g.W;
   dim3 grid((g.C + BN - 1) / BN, (int)((MI + BM - 1) / BM));
-  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(gy.scalar_type(),
                                   "conv_dgrad", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + BN) * Mma<T>::STRIDE;
     hipLaunchKernelGGL((conv_dgrad_kernel<T>), grid, dim3(kBlock), lds,
                        c10::hip::getCurrentHIPStream(),
                        (const T*)gy.data_ptr(), (const T*)wt.data_ptr(),
@@ -742,10 +818,11 @@ torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       mtotal, std::max<int64_t>(1, target_z / taps));
   int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
   dim3 grid(ctiles, ktiles, taps * mslices);
-  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(gy.scalar_type(),
                                   "conv_wgrad", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + BN) * Mma<T>::STRIDE;
     hipLaunchKernelGGL((conv_wgrad_kernel<T>), grid, dim3(kBlock), lds,
                        c10::hip::getCurrentHIPStream(),
                        (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
@@ -776,10 +853,11 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w) {
   auto g = linear_geom(x, w.size(0));
   auto out = torch::empty({x.size(0), w.size(0)}, x.options());
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
-  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(x.scalar_type(),
                                   "linear_fwd", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + BN) * Mma<T>::STRIDE;
     hipLaunchKernelGGL((conv_fwd_kernel<T, true, 0, false, false>), grid,
                        dim3(kBlock), lds, c10::hip::getCurrentHIPStream(),
                        (const T*)x.data_ptr(), (const T*)w.data_ptr(),
@@ -811,10 +889,11 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
   if (mslices < 1) mslices = 1;
   int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
   dim3 grid(ctiles, ktiles, mslices);
-  size_t lds = (size_t)(BM + BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(gy.scalar_type(),
                                   "linear_wgrad", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + BN) * Mma<T>::STRIDE;
     hipLaunchKernelGGL((conv_wgrad_kernel<T>), grid, dim3(kBlock), lds,
                        c10::hip::getCurrentHIPStream(),
                        (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
@@ -839,10 +918,11 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
   tele[2] = -std::numeric_limits<float>::infinity();
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
-  size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(x.scalar_type(),
                                   "linear_fwd_fused", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + 3 * BN) * Mma<T>::STRIDE;
     auto stream = c10::hip::getCurrentHIPStream();
     const T* xp = (const T*)x.data_ptr();
     const T* wqp = (const T*)wq.data_ptr();
@@ -885,10 +965,11 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
   tele[2] = -std::numeric_limits<float>::infinity();
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
-  size_t lds = (size_t)(BM + 3 * BN) * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(x.scalar_type(),
                                   "sigma_noise_linear", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)(BM + 3 * BN) * Mma<T>::STRIDE;
     auto stream = c10::hip::getCurrentHIPStream();
     const T* xp = (const T*)x.data_ptr();
     const T* wrp = (const T*)wraw.data_ptr();
@@ -946,6 +1027,7 @@ struct PatchGeom {
 template <typename T>
 DEV_INLINE void stage_patch(char* patch, const T* __restrict__ x,
                             const ConvGeom g, const PatchGeom p, int n) {
+  constexpr int EB = (int)sizeof(T) == 4 ? 4 : 2;
   int total = g.H * p.Wp * p.C_pad;
   const T* img = x + (int64_t)n * g.H * g.W * g.C;
   bool vec = (sizeof(T) == 2) && ((g.C & 7) == 0);
@@ -978,13 +1060,15 @@ DEV_INLINE void stage_patch(char* patch, const T* __restrict__ x,
       float v = 0.0f;
       if (iw >= 0 && iw < g.W && c < g.C)
         v = to_f32(img[((int64_t)ih * g.W + iw) * g.C + c]);
-      *(bf16*)(patch + (int64_t)idx * 2) = __float2bfloat16(v);
+      ((T*)patch)[idx] = from_f32<T>(v);
     }
   }
-  // zero slot at patch end for out-of-bounds fragment reads
-  if (threadIdx.x < 4)
-    *(bf16x8*)(patch + (int64_t)g.H * p.Wp * p.C_pad * 2 + threadIdx.x * 16) =
-        bf16x8{};
+  // zero slot at patch end for out-of-bounds fragment reads (one full
+  // 32-element span of the compute type)
+  if (threadIdx.x < 8) {
+    char* z = patch + (int64_t)g.H * p.Wp * p.C_pad * EB + threadIdx.x * 16;
+    *(bf16x8*)z = bf16x8{};
+  }
 }
 
 // weight tile rows from channel-padded weights [K, R, S*C_pad]
@@ -995,12 +1079,13 @@ DEV_INLINE void stage_wpad_row(char* lds, const T* __restrict__ w,
   int row = threadIdx.x >> 2;
   int seg = threadIdx.x & 3;
   int k = n0 + row;
-  bf16 vals[8];
+  float vals[8];
   if (k < g.K) {
     const T* pw = w + ((int64_t)k * g.R + r) * p.Kr;
     int c0 = ck + seg * 8;
     if (!ABS_TRANSFORM && sizeof(T) == 2 && c0 + 8 <= p.Kr) {
-      *(bf16x8*)lds_at(lds, row, seg * 16) = *(const bf16x8*)(pw + c0);
+      *(bf16x8*)(lds + row * Mma<T>::STRIDE + seg * 16) =
+          *(const bf16x8*)(pw + c0);
       return;
     }
 #pragma unroll
@@ -1011,13 +1096,13 @@ DEV_INLINE void stage_wpad_row(char* lds, const T* __restrict__ w,
         v = fabsf(v);
         if (SIGMA_MODE == 2) v = v * v + v;
       }
-      vals[j] = __float2bfloat16(v);
+      vals[j] = v;
     }
   } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.0f);
+    for (int j = 0; j < 8; ++j) vals[j] = 0.0f;
   }
-  *(bf16x8*)lds_at(lds, row, seg * 16) = *(bf16x8*)vals;
+  Mma<T>::store8(lds, row, seg * 8, vals);
 }
 
 template <typename T, bool WANT_Y, int SIGMA_MODE, bool TELEM, bool BIAS>
@@ -1029,16 +1114,18 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
                            const float* __restrict__ factor_p, uint64_t seed,
                            float* __restrict__ telem) {
   const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
+  constexpr int EB = (int)sizeof(T) == 4 ? 4 : 2;
+  constexpr int STR = Mma<T>::STRIDE;
   int n0 = blockIdx.x * BN;   // output-channel tile
   int n = blockIdx.y;         // image
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* patch = smem;  // H * Wp * C_pad bf16 + 64B zero slot
-  size_t patch_bytes = (size_t)g.H * p.Wp * p.C_pad * 2 + 64;
+  char* patch = smem;  // H * Wp * C_pad elements + one zero span
+  size_t patch_bytes = (size_t)g.H * p.Wp * p.C_pad * EB + 128;
   char* b_lds = smem + patch_bytes;
-  char* c_lds = b_lds + BN * LDS_STRIDE;
-  char* d_lds = c_lds + BN * LDS_STRIDE;
-  int zero_off = g.H * p.Wp * p.C_pad * 2;
+  char* c_lds = b_lds + BN * STR;
+  char* d_lds = c_lds + BN * STR;
+  int64_t zero_off = (int64_t)g.H * p.Wp * p.C_pad * EB;
 
   stage_patch(patch, x, g, p, n);
 
@@ -1063,14 +1150,14 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
         if (TELEM && SIGMA_MODE == 2)
           stage_wpad_row<T, true, 1>(d_lds, wraw, g, p, n0, r, ck);
         __syncthreads();
-        bf16x8 bfrag[2], sfrag[2], tfrag[2];
+        typename Mma<T>::frag bfrag[2], sfrag[2], tfrag[2];
 #pragma unroll
         for (int fn = 0; fn < 2; ++fn) {
-          if (WANT_Y) bfrag[fn] = frag_from_lds(b_lds, wn * 32 + fn * 16);
-          if (SIGMA_MODE > 0)
-            sfrag[fn] = frag_from_lds(c_lds, wn * 32 + fn * 16);
+          int brow = wn * 32 + fn * 16 + (lane & 15);
+          if (WANT_Y) bfrag[fn] = Mma<T>::load(b_lds, brow, lane);
+          if (SIGMA_MODE > 0) sfrag[fn] = Mma<T>::load(c_lds, brow, lane);
           if (TELEM && SIGMA_MODE == 2)
-            tfrag[fn] = frag_from_lds(d_lds, wn * 32 + fn * 16);
+            tfrag[fn] = Mma<T>::load(d_lds, brow, lane);
         }
 #pragma unroll
         for (int ms = 0; ms < 2; ++ms) {
@@ -1079,27 +1166,23 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
           for (int fm = 0; fm < 2; ++fm) {
             // per-lane patch fragment: 8 contiguous (s,c) at filter row r
             int m_local = m0 + ms * BM + wm * 32 + fm * 16 + (lane & 15);
-            int kf = ck + (lane >> 4) * 8;
-            int off = zero_off;
+            // span start for this row's 32 contraction elements at tap row r
+            int64_t off = zero_off;
             if (m_local < p.MI) {
               int ow = m_local % g.OW;
               int oh = m_local / g.OW;
               int ih = oh * g.stride - g.pad + r;
               if (ih >= 0 && ih < g.H)
-                off = (((int64_t)ih * p.Wp + ow * g.stride) * p.C_pad + kf) * 2;
+                off = (((int64_t)ih * p.Wp + ow * g.stride) * p.C_pad + ck)
+                      * EB;
             }
-            bf16x8 a = *(bf16x8*)(patch + off);
+            auto a = Mma<T>::load_span(patch + off, lane);
 #pragma unroll
             for (int fn = 0; fn < 2; ++fn) {
-              if (WANT_Y)
-                acc[ms][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a, bfrag[fn], acc[ms][fm][fn], 0, 0, 0);
-              if (SIGMA_MODE > 0)
-                sacc[ms][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a, sfrag[fn], sacc[ms][fm][fn], 0, 0, 0);
+              if (WANT_Y) Mma<T>::mma(a, bfrag[fn], acc[ms][fm][fn]);
+              if (SIGMA_MODE > 0) Mma<T>::mma(a, sfrag[fn], sacc[ms][fm][fn]);
               if (TELEM && SIGMA_MODE == 2)
-                tacc[ms][fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a, tfrag[fn], tacc[ms][fm][fn], 0, 0, 0);
+                Mma<T>::mma(a, tfrag[fn], tacc[ms][fm][fn]);
             }
           }
         }
@@ -1164,11 +1247,11 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
   }
 }
 
-inline bool patch_eligible(const ConvGeom& g) {
+inline bool patch_eligible(const ConvGeom& g, int elem_bytes) {
   int c_pad = (g.C + 7) & ~7;
   int wp = g.W + 2 * g.pad;
-  size_t patch_bytes = (size_t)g.H * wp * c_pad * 2 + 64;
-  return g.R * g.S > 1 && g.C > 8 && patch_bytes <= 64 * 1024;
+  size_t patch_bytes = (size_t)g.H * wp * c_pad * elem_bytes + 128;
+  return g.R * g.S > 1 && g.C > 8 && patch_bytes <= 96 * 1024;
 }
 
 // pad the raw-[K,R,S,C] weight view to [K,R,S*C_pad] (tiny tensors)
@@ -1207,11 +1290,12 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
   auto wq_pad = want_y ? pad_weight_raw(wq, p.C_pad) : torch::Tensor();
   auto wraw_pad = (sigma_mode > 0) ? pad_weight_raw(wraw, p.C_pad)
                                    : torch::Tensor();
-  size_t patch_bytes = (size_t)g.H * p.Wp * p.C_pad * 2 + 64;
-  size_t lds = patch_bytes + (size_t)3 * BN * LDS_STRIDE;
+  size_t lds = 0;  // per-dtype, set inside the dispatch
   dim3 grid((g.K + BN - 1) / BN, g.N);
   NN_DISPATCH(x.scalar_type(), "conv_fwd_patch", [&] {
     using T = typename DevT<scalar_t>::type;
+    lds = (size_t)g.H * p.Wp * p.C_pad * sizeof(T) + 128
+          + (size_t)3 * BN * Mma<T>::STRIDE;
     auto stream = c10::hip::getCurrentHIPStream();
     const T* xp = (const T*)x.data_ptr();
     const T* wqp = want_y ? (const T*)wq_pad.data_ptr()
@@ -1266,7 +1350,7 @@ std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
   auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
                      (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
                      (int)wraw.size(3), (int)stride, (int)pad);
-  if (patch_eligible(g))
+  if (patch_eligible(g, (int)x.element_size()))
     return conv_fwd_fused_patch_impl(x, wq, wraw, bias, stride, pad,
                                      sigma_mode, factor, seed, telem, true);
   return conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
@@ -1282,7 +1366,7 @@ std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
   auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
                      (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
                      (int)wraw.size(3), (int)stride, (int)pad);
-  if (patch_eligible(g))
+  if (patch_eligible(g, (int)x.element_size()))
     return conv_fwd_fused_patch_impl(x, wraw, wraw, empty_bias, stride, pad,
                                      sigma_mode, factor, seed, telem, false);
   return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,

@@ -89,7 +89,9 @@ def _patch_eligible(x, w, padding):
     R, S = w.shape[2], w.shape[3]
     c_pad = (C + 7) // 8 * 8
     wp = x.shape[3] + 2 * padding
-    return R * S > 1 and C > 8 and x.shape[2] * wp * c_pad * 2 + 64 <= 65536
+    eb = x.element_size()
+    return (R * S > 1 and C > 8
+            and x.shape[2] * wp * c_pad * eb + 128 <= 96 * 1024)
 
 
 def _conv_fwd_raw(x, w, bias, stride, padding):

@@ -660,3 +660,63 @@ def test_cifar_driver_short_gpu(tmp_path, monkeypatch):
                       '--q_a', '4', '--calculate_running', '--LR', '0.005',
                       '--bf16', '--keep_bn_fp32'])
     assert out
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float16])
+def test_conv_exact_dtype_paths(dtype):
+    """fp32 convs use the exact f32-input MFMA (bitwise f32 fmaf chain,
+    cdna_hip_programming.md §3); fp16 uses the f16 MFMA. Tight tolerances
+    prove the compute really runs at the input precision (a bf16-internal
+    path would miss by ~1e-2)."""
+    torch.manual_seed(31)
+    for (N, C, H, W, K, R, stride, pad) in (
+            (4, 65, 14, 14, 120, 5, 1, 0),
+            (2, 16, 16, 16, 32, 3, 1, 1),
+            (4, 3, 32, 32, 65, 5, 1, 0)):
+        x = torch.randn(N, C, H, W, dtype=dtype, device=dev())
+        w = (torch.randn(K, C, R, R, dtype=dtype, device=dev()) * 0.2)
+        out = ops.ext().conv_fwd(cl(x), cl(w), stride, pad)
+        expect = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad)
+        rel = (out.float().cpu() - expect).abs().max() / (expect.abs().max() + 1e-6)
+        tol = 1e-5 if dtype == torch.float32 else 4e-3
+        assert rel < tol, (dtype, (N, C, H, W, K, R), rel)
+
+        # dgrad/wgrad too
+        OH = (H + 2 * pad - R) // stride + 1
+        g = torch.randn(N, K, OH, OH, dtype=dtype, device=dev())
+        dx = ops.ext().conv_dgrad(cl(g), cl(w), stride, pad, H, W)
+        dxe = torch.nn.grad.conv2d_input((N, C, H, W), w.float().cpu(),
+                                         g.float().cpu(), stride, pad)
+        rel = (dx.float().cpu() - dxe).abs().max() / (dxe.abs().max() + 1e-6)
+        assert rel < (1e-5 if dtype == torch.float32 else 5e-3), (dtype, rel)
+
+        dw = ops.ext().conv_wgrad(cl(g), cl(x), stride, pad, R, R)
+        dwe = torch.nn.grad.conv2d_weight(x.float().cpu(), (K, C, R, R),
+                                          g.float().cpu(), stride, pad)
+        rel = (dw.float().cpu() - dwe).abs().max() / (dwe.abs().max() + 1e-6)
+        assert rel < (1e-4 if dtype == torch.float32 else 2e-2), (dtype, rel)
+
+
+def test_linear_fp32_exact():
+    torch.manual_seed(32)
+    x = torch.randn(64, 3000, device=dev())
+    w = torch.randn(390, 3000, device=dev()) * 0.05
+    y = ops.ext().linear_fwd(x, w)
+    ye = x.cpu() @ w.cpu().t()
+    rel = (y.cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
+    assert rel < 1e-5, rel
+
+
+def test_patch_conv_fp32_exact():
+    torch.manual_seed(33)
+    x = torch.randn(4, 24, 14, 14, device=dev())
+    w = torch.randn(32, 24, 3, 3, device=dev()) * 0.2
+    from noisynet_amd.ops.functional import _patch_eligible
+    assert _patch_eligible(x, w, 1)
+    empty = torch.empty(0, device=dev())
+    zf = torch.zeros(1, device=dev())
+    y, _ = ops.ext().conv_fwd_fused(cl(x), cl(w), cl(w), empty, 1, 1, 0, zf,
+                                    0, False)
+    ye = F.conv2d(x.cpu(), w.cpu(), None, 1, 1)
+    rel = (y.cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
+    assert rel < 1e-5, rel
