@@ -73,6 +73,9 @@ template <> struct Mma<bf16> {
   static DEV_INLINE void store1(char* lds, int row, int k, float v) {
     *(bf16*)(lds + row * STRIDE + k * 2) = __float2bfloat16(v);
   }
+  static DEV_INLINE void store1n(char* lds, int row, int k, bf16 v) {
+    *(bf16*)(lds + row * STRIDE + k * 2) = v;
+  }
   // read the lane's fragment from a RAW 32-element span (no tile rows)
   static DEV_INLINE frag load_span(const char* base, int lane) {
     return *(frag*)(base + (lane >> 4) * 16);
@@ -96,6 +99,9 @@ template <> struct Mma<_Float16> {
   }
   static DEV_INLINE void store1(char* lds, int row, int k, float v) {
     *(_Float16*)(lds + row * STRIDE + k * 2) = (_Float16)v;
+  }
+  static DEV_INLINE void store1n(char* lds, int row, int k, _Float16 v) {
+    *(_Float16*)(lds + row * STRIDE + k * 2) = v;
   }
   static DEV_INLINE frag load_span(const char* base, int lane) {
     return *(frag*)(base + (lane >> 4) * 16);
@@ -123,6 +129,9 @@ template <> struct Mma<float> {
   }
   static DEV_INLINE void store1(char* lds, int row, int k, float v) {
     ((float*)(lds + row * STRIDE))[((k & 3) << 3) + (k >> 2)] = v;
+  }
+  static DEV_INLINE void store1n(char* lds, int row, int k, float v) {
+    store1(lds, row, k, v);
   }
   static DEV_INLINE frag load_span(const char* base, int lane) {
     // raw span is k-linear: gather the lane's strided k = q + 4*kk
@@ -570,23 +579,34 @@ DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
     }
     return;
   }
-  float gvals[8], xvals[8];
+  // Keep the values in the NATIVE element type: for 16-bit dtypes the
+  // aligned case is one 16-B global vector load per operand (this kernel is
+  // global-load bound -- 8 scalar u16 loads here cost ~2.5x on the bench).
+  T gvals[8], xvals[8];
   int kc0 = k0 + colseg * 8;
+  if (mok && sizeof(T) == 2 && (g.K & 7) == 0 && kc0 + 8 <= g.K) {
+    *(bf16x8*)gvals = *(const bf16x8*)(pg + kc0);  // raw 16-B copy (punned)
+  } else {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    int kcol = kc0 + j;
-    gvals[j] = (mok && kcol < g.K) ? to_f32(pg[kcol]) : 0.0f;
+    for (int j = 0; j < 8; ++j) {
+      int kcol = kc0 + j;
+      gvals[j] = (mok && kcol < g.K) ? pg[kcol] : from_f32<T>(0.0f);
+    }
   }
   int cc0 = c0 + colseg * 8;
+  if (px != nullptr && sizeof(T) == 2 && (g.C & 7) == 0 && cc0 + 8 <= g.C) {
+    *(bf16x8*)xvals = *(const bf16x8*)(px + cc0);
+  } else {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    int ccol = cc0 + j;
-    xvals[j] = (px != nullptr && ccol < g.C) ? to_f32(px[ccol]) : 0.0f;
+    for (int j = 0; j < 8; ++j) {
+      int ccol = cc0 + j;
+      xvals[j] = (px != nullptr && ccol < g.C) ? px[ccol] : from_f32<T>(0.0f);
+    }
   }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    Mma<T>::store1(g_lds, colseg * 8 + j, mi, gvals[j]);
-    Mma<T>::store1(x_lds, colseg * 8 + j, mi, xvals[j]);
+    Mma<T>::store1n(g_lds, colseg * 8 + j, mi, gvals[j]);
+    Mma<T>::store1n(x_lds, colseg * 8 + j, mi, xvals[j]);
   }
 }
 
