@@ -1361,6 +1361,50 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
 
 }  // namespace
 
+// defined below (im2col section); used by the flat/small-C fused route
+torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
+                                 int64_t pad, int64_t R, int64_t S);
+
+namespace {
+
+// Flat/small-C convs (C < 16, R*S > 1, e.g. a 5x5 conv over RGB): the
+// streaming kernel's per-element (r,s,c) decode gathers scalars; one
+// coalesced im2col pass + the 1x1-GEMM fused kernel over the padded
+// [M, R*S*c_pad] rows (16-B vectorized staging) is ~3x faster. The
+// padded weight columns are zero, so sigma over |W| / |W|^2+|W| and the
+// Philox noise index m*K+k are unchanged.
+std::vector<torch::Tensor> conv_fwd_fused_col_impl(
+    torch::Tensor x, torch::Tensor wq, torch::Tensor wraw, torch::Tensor bias,
+    int64_t stride, int64_t pad, int64_t sigma_mode, torch::Tensor factor,
+    int64_t seed, bool telem, bool want_y) {
+  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
+                     (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
+                     (int)wraw.size(3), (int)stride, (int)pad);
+  int rsc = g.R * g.S * g.C;
+  int cols_p = (rsc + 7) & ~7;
+  auto col = im2col_materialize(x, g.K, stride, pad, g.R, g.S);  // [M,cols_p]
+  // flat weight rows: raw [K,R,S,C] view reshaped to [K, R*S*C], zero-padded
+  auto flat_w = [&](const torch::Tensor& w) {
+    auto raw = w.permute({0, 2, 3, 1}).reshape({(int64_t)g.K, (int64_t)rsc});
+    return at::constant_pad_nd(raw, {0, cols_p - rsc}, 0).contiguous();
+  };
+  auto wraw_flat = flat_w(wraw);
+  auto wq_flat = want_y ? flat_w(wq) : wraw_flat;
+  std::vector<torch::Tensor> r;
+  if (want_y) {
+    r = linear_fwd_fused(col, wq_flat, wraw_flat, bias, sigma_mode, factor,
+                         seed, telem);
+  } else {
+    r = sigma_noise_linear_impl(col, wraw_flat, sigma_mode, factor, seed,
+                                telem);
+  }
+  // [M, K] row-major IS NHWC: reinterpret as a channels_last 4-D view
+  auto out = r[0].view({g.N, g.OH, g.OW, g.K}).permute({0, 3, 1, 2});
+  return {out, r[1]};
+}
+
+}  // namespace
+
 std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
                                           torch::Tensor wraw, torch::Tensor bias,
                                           int64_t stride, int64_t pad,
@@ -1373,6 +1417,9 @@ std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
   if (patch_eligible(g, (int)x.element_size()))
     return conv_fwd_fused_patch_impl(x, wq, wraw, bias, stride, pad,
                                      sigma_mode, factor, seed, telem, true);
+  if (g.flat)
+    return conv_fwd_fused_col_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
+                                   factor, seed, telem, /*want_y=*/true);
   return conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
                              factor, seed, telem, /*want_y=*/true);
 }
@@ -1389,6 +1436,10 @@ std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
   if (patch_eligible(g, (int)x.element_size()))
     return conv_fwd_fused_patch_impl(x, wraw, wraw, empty_bias, stride, pad,
                                      sigma_mode, factor, seed, telem, false);
+  if (g.flat)
+    return conv_fwd_fused_col_impl(x, wraw, wraw, empty_bias, stride, pad,
+                                   sigma_mode, factor, seed, telem,
+                                   /*want_y=*/false);
   return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,
                              sigma_mode, factor, seed, telem, /*want_y=*/false);
 }
@@ -1409,8 +1460,9 @@ namespace {
 
 template <typename T>
 __global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
-                              ConvGeom g, int c_pad, int64_t nspans) {
-  // span = (m, r, s): copies C channels (zero-padding to c_pad)
+                              ConvGeom g, int64_t nspans) {
+  // C % 8 == 0 fast path: span = (m, r, s) copies the C channels of one
+  // tap as aligned 16-B vectors into the flat row at column rs*C.
   int taps = g.R * g.S;
   for (int64_t span = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        span < nspans; span += (int64_t)gridDim.x * blockDim.x) {
@@ -1426,38 +1478,80 @@ __global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
     int n = (int)t;
     int ih = oh * g.stride - g.pad + r;
     int iw = ow * g.stride - g.pad + s;
-    T* dst = out + (m * taps + rs) * (int64_t)c_pad;
+    T* dst = out + (m * taps + rs) * (int64_t)g.C;
     bool inb = (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W);
     const T* src = inb ? x + (((int64_t)n * g.H + ih) * g.W + iw) * g.C
                        : nullptr;
     if (sizeof(T) == 2) {
-      // destination rows are 16B-aligned (c_pad % 8 == 0): build each
-      // 8-element chunk in registers and emit one vector store
-      for (int c0 = 0; c0 < c_pad; c0 += 8) {
+      for (int c0 = 0; c0 < g.C; c0 += 8) {
         bf16x8 v;
-        if (inb && (g.C & 7) == 0 && c0 + 8 <= g.C) {
+        if (inb) {
           v = *(const bf16x8*)(src + c0);
         } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            int c = c0 + j;
-            ((bf16*)&v)[j] = (inb && c < g.C)
-                ? __float2bfloat16(to_f32(src[c]))
-                : __float2bfloat16(0.0f);
-          }
+          v = bf16x8{};
         }
         *(bf16x8*)(dst + c0) = v;
       }
     } else {
-      for (int c = 0; c < c_pad; ++c)
-        dst[c] = (inb && c < g.C) ? src[c] : from_f32<T>(0.0f);
+      for (int c = 0; c < g.C; ++c)
+        dst[c] = inb ? src[c] : from_f32<T>(0.0f);
+    }
+  }
+}
+
+// C % 8 != 0 (small-C convs): flat row of R*S*C columns padded to a
+// multiple of 8; each thread fills one aligned 8-column chunk, decoding
+// the (r, s, c) of each column. The decode cost is paid ONCE here instead
+// of on every LDS staging pass of the conv/wgrad kernels.
+template <typename T>
+__global__ void im2col_flat_kernel(const T* __restrict__ x,
+                                   T* __restrict__ out, ConvGeom g,
+                                   int cols_p, int64_t nchunks_total) {
+  int nchunks = cols_p >> 3;
+  int rsc = g.R * g.S * g.C;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < nchunks_total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int seg = (int)(idx % nchunks);
+    int64_t m = idx / nchunks;
+    int64_t t = m;
+    int ow = (int)(t % g.OW);
+    t /= g.OW;
+    int oh = (int)(t % g.OH);
+    t /= g.OH;
+    int n = (int)t;
+    T vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = seg * 8 + j;
+      float v = 0.0f;
+      if (k < rsc) {
+        int c = k % g.C;
+        int rs = k / g.C;
+        int s = rs % g.S;
+        int r = rs / g.S;
+        int ih = oh * g.stride - g.pad + r;
+        int iw = ow * g.stride - g.pad + s;
+        if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
+          v = to_f32(x[(((int64_t)n * g.H + ih) * g.W + iw) * g.C + c]);
+      }
+      vals[j] = from_f32<T>(v);
+    }
+    T* dst = out + m * cols_p + seg * 8;
+    if (sizeof(T) == 2) {
+      *(bf16x8*)dst = *(bf16x8*)vals;  // raw 16 B
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[j] = vals[j];
     }
   }
 }
 
 }  // namespace
 
-// exposed for testing: the materialization pass alone
+// Materialize the im2col matrix [M, cols_p] with the FLAT column layout:
+// column k = flattened (r, s, c), row padded with zeros to cols_p =
+// round8(R*S*C). (For C % 8 == 0 that equals the per-tap layout.)
+// Exposed for testing.
 torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
                                  int64_t pad, int64_t R, int64_t S) {
   check_cl(x, "im2col x");
@@ -1465,17 +1559,24 @@ torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
                      (int)x.size(1), (int)K, (int)R, (int)S, (int)stride,
                      (int)pad);
   int taps = g.R * g.S;
-  int c_pad = (g.C + 7) & ~7;
-  int cols = taps * c_pad;
-  auto col = torch::empty({g.M, cols}, x.options());
-  int64_t nspans = g.M * taps;
-  int blocks = (int)std::min<int64_t>((nspans + 255) / 256, 8192);
+  int cols_p = (taps * g.C + 7) & ~7;
+  auto col = torch::empty({g.M, cols_p}, x.options());
   NN_DISPATCH(x.scalar_type(), "im2col", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((im2col_kernel<T>), dim3(blocks), dim3(256), 0,
-                       c10::hip::getCurrentHIPStream(),
-                       (const T*)x.data_ptr(), (T*)col.data_ptr(), g, c_pad,
-                       nspans);
+    auto stream = c10::hip::getCurrentHIPStream();
+    if ((g.C & 7) == 0) {
+      int64_t nspans = g.M * taps;
+      int blocks = (int)std::min<int64_t>((nspans + 255) / 256, 8192);
+      hipLaunchKernelGGL((im2col_kernel<T>), dim3(blocks), dim3(256), 0,
+                         stream, (const T*)x.data_ptr(), (T*)col.data_ptr(),
+                         g, nspans);
+    } else {
+      int64_t nchunks_total = g.M * (cols_p >> 3);
+      int blocks = (int)std::min<int64_t>((nchunks_total + 255) / 256, 8192);
+      hipLaunchKernelGGL((im2col_flat_kernel<T>), dim3(blocks), dim3(256), 0,
+                         stream, (const T*)x.data_ptr(), (T*)col.data_ptr(),
+                         g, cols_p, nchunks_total);
+    }
   });
   HIP_CHECK_LAST();
   return col;
@@ -1491,28 +1592,23 @@ torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
   auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
                      (int)x.size(1), (int)gy.size(1), (int)R, (int)S,
                      (int)stride, (int)pad);
-  int taps = g.R * g.S;
-  // pad the flattened (r,s,c) ... per-tap c_pad keeps spans aligned
-  int c_pad = (g.C + 7) & ~7;
-  int cols = taps * c_pad;
-  auto col = torch::empty({g.M, cols}, x.options());
-  int64_t nspans = g.M * taps;
-  int blocks = (int)std::min<int64_t>((nspans + 255) / 256, 8192);
-  NN_DISPATCH(x.scalar_type(), "im2col", [&] {
-    using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((im2col_kernel<T>), dim3(blocks), dim3(256), 0,
-                       c10::hip::getCurrentHIPStream(),
-                       (const T*)x.data_ptr(), (T*)col.data_ptr(), g, c_pad,
-                       nspans);
-  });
-  HIP_CHECK_LAST();
+  int rsc = g.R * g.S * g.C;
+  int cols_p = (rsc + 7) & ~7;
+  auto col = im2col_materialize(x, gy.size(1), stride, pad, R, S);
 
-  // dw[k, taps*c_pad] = gy^T @ col via the dense transposed-staging wgrad
+  // dw[k, cols_p] = gy^T @ col. Tall-skinny contractions into a tiny
+  // output (conv1-type: M ~1e6, K*cols ~1e4) favor the split-M atomic
+  // wgrad kernel (2.5x over hipBLASLt there); wide outputs favor the
+  // library GEMM (measured: conv2-type 0.87 ms blas vs 1.19 ms ours).
   auto gy2 = gy.permute({0, 2, 3, 1}).reshape({g.M, g.K});  // raw view, free
-  auto dw_flat = linear_wgrad(gy2, col);                    // [K, cols]
-  // un-pad: [K, R, S, c_pad] -> [K, R, S, C] -> logical [K, C, R, S] cl
-  auto dw = dw_flat.view({g.K, g.R, g.S, c_pad})
-                .narrow(3, 0, g.C)
+  torch::Tensor dw_flat;
+  if ((int64_t)g.K * cols_p <= 65536 && g.M >= 262144)
+    dw_flat = linear_wgrad(gy2, col);                       // [K, cols_p]
+  else
+    dw_flat = at::matmul(gy2.t(), col).to(x.scalar_type());
+  // un-pad the flat row: [K, cols_p] -> [K, R*S*C] -> logical [K,C,R,S] cl
+  auto dw = dw_flat.narrow(1, 0, rsc)
+                .view({g.K, g.R, g.S, g.C})
                 .permute({0, 3, 1, 2})
                 .contiguous(at::MemoryFormat::ChannelsLast);
   return dw;

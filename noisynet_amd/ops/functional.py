@@ -132,9 +132,9 @@ def _conv_wgrad_raw(g, x, stride, padding, w_shape):
         if R * S > 1:
             # im2col-GEMM wgrad when the buffer is affordable (<2 GB):
             # one coalesced materialization pass beats per-K-chunk gather
-            c_pad = (x.shape[1] + 7) // 8 * 8
+            cols_p = (R * S * x.shape[1] + 7) // 8 * 8
             m = g.shape[0] * g.shape[2] * g.shape[3]
-            col_bytes = m * R * S * c_pad * x.element_size()
+            col_bytes = m * cols_p * x.element_size()
             if col_bytes <= 2 << 30:
                 return ext().conv_wgrad_im2col(_nhwc(g), _nhwc(x), stride,
                                                padding, R, S)
@@ -351,11 +351,11 @@ def sigma_noise_linear(x, w_raw, sigma_mode, factor, want_sigma_abs=False):
 
 
 def _linear_fwd_raw(x, w, bias):
-    if use_native(x, w):
-        y = ext().linear_fwd(x.contiguous(), w.contiguous())
-        if bias is not None:
-            y = y + bias
-        return y
+    # Plain (un-fused) linear IS a library GEMM: hipBLASLt via F.linear
+    # beats our MFMA kernel on the fc shapes (0.023 vs 0.172 ms for
+    # 2048x3000 @ 390x3000 bf16 on MI355X). The custom kernel stays for
+    # the FUSED noisy path (linear_fwd_fused) where sigma+noise ride the
+    # same tiles, and is still exercised directly in tests/test_ops_gpu.
     return F.linear(x, w, bias)
 
 
@@ -385,9 +385,7 @@ class LinearDgrad(torch.autograd.Function):
     @staticmethod
     def forward(ctx, g, w):
         ctx.save_for_backward(g, w)
-        if use_native(g, w):
-            return ext().linear_dgrad(g.contiguous(), w.contiguous())
-        return g.matmul(w)
+        return g.matmul(w)  # plain GEMM -> hipBLASLt
 
     @staticmethod
     def backward(ctx, gg):
@@ -406,9 +404,7 @@ class LinearWgrad(torch.autograd.Function):
     @staticmethod
     def forward(ctx, g, x):
         ctx.save_for_backward(g, x)
-        if use_native(g, x):
-            return ext().linear_wgrad(g.contiguous(), x.contiguous())
-        return g.t().matmul(x)
+        return g.t().matmul(x)  # plain GEMM -> hipBLASLt
 
     @staticmethod
     def backward(ctx, gw):

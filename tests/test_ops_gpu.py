@@ -720,3 +720,24 @@ def test_patch_conv_fp32_exact():
     ye = F.conv2d(x.cpu(), w.cpu(), None, 1, 1)
     rel = (y.cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
     assert rel < 1e-5, rel
+
+
+@pytest.mark.gpu
+def test_im2col_materialize_flat_layout():
+    """im2col produces the FLAT (r,s,c)-flattened row layout, zero-padded
+    to a multiple of 8 columns, for both the aligned (C%8==0) and the
+    decode (C%8!=0) kernels."""
+    torch.manual_seed(0)
+    for C in (3, 8, 13, 16):
+        x = torch.randn(4, C, 9, 9, device="cuda", dtype=torch.bfloat16)
+        xc = x.contiguous(memory_format=torch.channels_last)
+        col = ops.ext().im2col_materialize(xc, 7, 1, 2, 3, 3)
+        ref_u = F.unfold(x.float(), 3, padding=2)         # [N, C*9, L]
+        N, _, L = ref_u.shape
+        ref_u = (ref_u.view(N, C, 9, L).permute(0, 3, 2, 1)
+                 .reshape(N * L, 9 * C))                  # [(m), (r,s,c)]
+        rsc = 9 * C
+        assert col.shape == (N * L, (rsc + 7) // 8 * 8)
+        assert torch.allclose(col[:, :rsc].float().cpu(), ref_u.cpu(),
+                              atol=1e-1, rtol=1e-2), C
+        assert (col[:, rsc:] == 0).all()
