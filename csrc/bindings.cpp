@@ -91,6 +91,8 @@ torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
                          int64_t pad, int64_t R, int64_t S);
 torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
                                  int64_t pad, int64_t R, int64_t S);
+torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
+                                  int64_t C, int64_t R, int64_t S);
 torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
                                 int64_t stride, int64_t pad, int64_t R,
                                 int64_t S);
@@ -148,6 +150,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);
   m.def("conv_wgrad_im2col", &conv_wgrad_im2col);
+  m.def("conv_wgrad_from_col", &conv_wgrad_from_col);
   m.def("im2col_materialize", &im2col_materialize);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_dgrad", &linear_dgrad);

@@ -742,8 +742,15 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
   bool has_bias = bias.numel() > 0;
   torch::Tensor bias_f;
   if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
-  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
-  tele[2] = -std::numeric_limits<float>::infinity();
+  // telemetry buffer only when requested (steady-state training has
+  // telem=false; the zeros+fill launches were ~12 tiny kernels/step)
+  auto tele = telem
+      ? [&] {
+          auto t = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+          t[2] = -std::numeric_limits<float>::infinity();
+          return t;
+        }()
+      : torch::empty({0}, x.options().dtype(torch::kFloat32));
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
   size_t lds = 0;  // per-dtype, set inside the dispatch
@@ -757,7 +764,7 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
     const T* wrp = (const T*)wraw.data_ptr();
     const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
     T* op = (T*)out.data_ptr();
-    float* tp = tele.data_ptr<float>();
+    float* tp = telem ? tele.data_ptr<float>() : nullptr;
     const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     auto launch = [&](auto wy, auto sm, auto tl, auto bi) {
@@ -934,8 +941,15 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
   bool has_bias = bias.numel() > 0;
   torch::Tensor bias_f;
   if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
-  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
-  tele[2] = -std::numeric_limits<float>::infinity();
+  // telemetry buffer only when requested (steady-state training has
+  // telem=false; the zeros+fill launches were ~12 tiny kernels/step)
+  auto tele = telem
+      ? [&] {
+          auto t = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+          t[2] = -std::numeric_limits<float>::infinity();
+          return t;
+        }()
+      : torch::empty({0}, x.options().dtype(torch::kFloat32));
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
   size_t lds = 0;  // per-dtype, set inside the dispatch
@@ -949,7 +963,7 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
     const T* wrp = (const T*)wraw.data_ptr();
     const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
     T* op = (T*)out.data_ptr();
-    float* tp = tele.data_ptr<float>();
+    float* tp = telem ? tele.data_ptr<float>() : nullptr;
     const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     auto launch = [&](auto sm, auto tl, auto bi) {
@@ -981,8 +995,15 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
                                                    bool telem) {
   auto g = linear_geom(x, wraw.size(0));
   auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
-  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
-  tele[2] = -std::numeric_limits<float>::infinity();
+  // telemetry buffer only when requested (steady-state training has
+  // telem=false; the zeros+fill launches were ~12 tiny kernels/step)
+  auto tele = telem
+      ? [&] {
+          auto t = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+          t[2] = -std::numeric_limits<float>::infinity();
+          return t;
+        }()
+      : torch::empty({0}, x.options().dtype(torch::kFloat32));
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));
   size_t lds = 0;  // per-dtype, set inside the dispatch
@@ -994,7 +1015,7 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
     const T* xp = (const T*)x.data_ptr();
     const T* wrp = (const T*)wraw.data_ptr();
     T* op = (T*)out.data_ptr();
-    float* tp = tele.data_ptr<float>();
+    float* tp = telem ? tele.data_ptr<float>() : nullptr;
     const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     if (sigma_mode == 1) {
@@ -1304,8 +1325,15 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
   bool has_bias = bias.numel() > 0;
   torch::Tensor bias_f;
   if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
-  auto tele = torch::zeros({3}, x.options().dtype(torch::kFloat32));
-  tele[2] = -std::numeric_limits<float>::infinity();
+  // telemetry buffer only when requested (steady-state training has
+  // telem=false; the zeros+fill launches were ~12 tiny kernels/step)
+  auto tele = telem
+      ? [&] {
+          auto t = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+          t[2] = -std::numeric_limits<float>::infinity();
+          return t;
+        }()
+      : torch::empty({0}, x.options().dtype(torch::kFloat32));
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
   auto wq_pad = want_y ? pad_weight_raw(wq, p.C_pad) : torch::Tensor();
   auto wraw_pad = (sigma_mode > 0) ? pad_weight_raw(wraw, p.C_pad)
@@ -1323,7 +1351,7 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
     const T* wrp = (sigma_mode > 0) ? (const T*)wraw_pad.data_ptr() : wqp;
     const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
     T* op = (T*)out.data_ptr();
-    float* tp = tele.data_ptr<float>();
+    float* tp = telem ? tele.data_ptr<float>() : nullptr;
     const float* f = factor_f.data_ptr<float>();
     uint64_t sd = (uint64_t)seed;
     auto launch = [&](auto wy, auto sm, auto tl, auto bi) {
@@ -1400,11 +1428,14 @@ std::vector<torch::Tensor> conv_fwd_fused_col_impl(
   }
   // [M, K] row-major IS NHWC: reinterpret as a channels_last 4-D view
   auto out = r[0].view({g.N, g.OH, g.OW, g.K}).permute({0, 3, 1, 2});
-  return {out, r[1]};
+  return {out, r[1], col};
 }
 
 }  // namespace
 
+// Returns {out, telemetry, col}: col is the flat im2col matrix when the
+// small-C route materialized one (callers pass it back to
+// conv_wgrad_from_col so backward skips the materialization), else empty.
 std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
                                           torch::Tensor wraw, torch::Tensor bias,
                                           int64_t stride, int64_t pad,
@@ -1414,14 +1445,18 @@ std::vector<torch::Tensor> conv_fwd_fused(torch::Tensor x, torch::Tensor wq,
   auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
                      (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
                      (int)wraw.size(3), (int)stride, (int)pad);
-  if (patch_eligible(g, (int)x.element_size()))
-    return conv_fwd_fused_patch_impl(x, wq, wraw, bias, stride, pad,
-                                     sigma_mode, factor, seed, telem, true);
+  auto empty = torch::empty({0}, x.options());
+  if (patch_eligible(g, (int)x.element_size())) {
+    auto r = conv_fwd_fused_patch_impl(x, wq, wraw, bias, stride, pad,
+                                       sigma_mode, factor, seed, telem, true);
+    return {r[0], r[1], empty};
+  }
   if (g.flat)
     return conv_fwd_fused_col_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
                                    factor, seed, telem, /*want_y=*/true);
-  return conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
-                             factor, seed, telem, /*want_y=*/true);
+  auto r = conv_fwd_fused_impl(x, wq, wraw, bias, stride, pad, sigma_mode,
+                               factor, seed, telem, /*want_y=*/true);
+  return {r[0], r[1], empty};
 }
 
 std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
@@ -1436,10 +1471,12 @@ std::vector<torch::Tensor> sigma_noise_conv(torch::Tensor x, torch::Tensor wraw,
   if (patch_eligible(g, (int)x.element_size()))
     return conv_fwd_fused_patch_impl(x, wraw, wraw, empty_bias, stride, pad,
                                      sigma_mode, factor, seed, telem, false);
-  if (g.flat)
-    return conv_fwd_fused_col_impl(x, wraw, wraw, empty_bias, stride, pad,
-                                   sigma_mode, factor, seed, telem,
-                                   /*want_y=*/false);
+  if (g.flat) {
+    auto r = conv_fwd_fused_col_impl(x, wraw, wraw, empty_bias, stride, pad,
+                                     sigma_mode, factor, seed, telem,
+                                     /*want_y=*/false);
+    return {r[0], r[1]};
+  }
   return conv_fwd_fused_impl(x, wraw, wraw, empty_bias, stride, pad,
                              sigma_mode, factor, seed, telem, /*want_y=*/false);
 }
@@ -1582,6 +1619,9 @@ torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
   return col;
 }
 
+torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
+                                  int64_t C, int64_t R, int64_t S);
+
 // conv wgrad through materialized im2col (called from Python when the
 // buffer fits; falls back to conv_wgrad otherwise).
 torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
@@ -1589,26 +1629,36 @@ torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
                                 int64_t S) {
   check_cl(gy, "conv_wgrad_im2col gy");
   check_cl(x, "conv_wgrad_im2col x");
-  auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
-                     (int)x.size(1), (int)gy.size(1), (int)R, (int)S,
-                     (int)stride, (int)pad);
-  int rsc = g.R * g.S * g.C;
-  int cols_p = (rsc + 7) & ~7;
   auto col = im2col_materialize(x, gy.size(1), stride, pad, R, S);
+  return conv_wgrad_from_col(gy, col, x.size(1), R, S);
+}
+
+// wgrad when the flat im2col matrix already exists (shared from the
+// fused forward's col route -- the input does not change between the
+// forward and its backward, so the materialization is paid once).
+torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
+                                  int64_t C, int64_t R, int64_t S) {
+  check_cl(gy, "conv_wgrad_from_col gy");
+  TORCH_CHECK(col.dim() == 2 && col.is_contiguous());
+  int64_t M = col.size(0);
+  int64_t cols_p = col.size(1);
+  int64_t K = gy.size(1);
+  int64_t rsc = R * S * C;
+  TORCH_CHECK((int64_t)gy.size(0) * gy.size(2) * gy.size(3) == M);
 
   // dw[k, cols_p] = gy^T @ col. Tall-skinny contractions into a tiny
   // output (conv1-type: M ~1e6, K*cols ~1e4) favor the split-M atomic
   // wgrad kernel (2.5x over hipBLASLt there); wide outputs favor the
   // library GEMM (measured: conv2-type 0.87 ms blas vs 1.19 ms ours).
-  auto gy2 = gy.permute({0, 2, 3, 1}).reshape({g.M, g.K});  // raw view, free
+  auto gy2 = gy.permute({0, 2, 3, 1}).reshape({M, K});  // raw view, free
   torch::Tensor dw_flat;
-  if ((int64_t)g.K * cols_p <= 65536 && g.M >= 262144)
+  if (K * cols_p <= 65536 && M >= 262144)
     dw_flat = linear_wgrad(gy2, col);                       // [K, cols_p]
   else
-    dw_flat = at::matmul(gy2.t(), col).to(x.scalar_type());
+    dw_flat = at::matmul(gy2.t(), col).to(col.scalar_type());
   // un-pad the flat row: [K, cols_p] -> [K, R*S*C] -> logical [K,C,R,S] cl
   auto dw = dw_flat.narrow(1, 0, rsc)
-                .view({g.K, g.R, g.S, g.C})
+                .view({K, R, S, C})
                 .permute({0, 3, 1, 2})
                 .contiguous(at::MemoryFormat::ChannelsLast);
   return dw;

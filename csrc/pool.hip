@@ -87,8 +87,12 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
                              int64_t W) {
   TORCH_CHECK(g.dim() == 4 && g.is_contiguous(at::MemoryFormat::ChannelsLast));
   int N = (int)g.size(0), C = (int)g.size(1), OH = (int)g.size(2), OW = (int)g.size(3);
-  auto gx = torch::zeros({N, C, (int)H, (int)W},
+  // NOTE: the zeros() factory drops memory_format from TensorOptions (the
+  // returned grad then forces an NHWC re-copy in every conv backward);
+  // empty()+zero_() keeps the channels_last layout.
+  auto gx = torch::empty({N, C, (int)H, (int)W},
                          g.options().memory_format(at::MemoryFormat::ChannelsLast));
+  gx.zero_();
   int64_t n_out = (int64_t)N * C * OH * OW;
   int blocks = (int)std::min<int64_t>((n_out + kBlock - 1) / kBlock, 8192);
   NN_DISPATCH(g.scalar_type(),

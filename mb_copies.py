@@ -1,5 +1,10 @@
-"""Attribute aten::copy_ calls in one flagship training step to Python stacks."""
+"""Find which ops copy big activations in one flagship step (TorchDispatchMode)."""
+import traceback
+from collections import Counter
+
 import torch
+from torch.utils._python_dispatch import TorchDispatchMode
+
 import bench as bench_mod
 from noisynet_amd.models.noisynet import Net
 from noisynet_amd import optim as native_optim, utils
@@ -37,23 +42,34 @@ def step(i):
     opt.step()
 
 
-for i in range(5):
+for i in range(3):
     step(i + 100)
 torch.cuda.synchronize()
 
-from torch.profiler import profile, ProfilerActivity
-with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
-             with_stack=True) as prof:
-    for i in range(2):
-        step(i + 1000)
-    torch.cuda.synchronize()
+log = Counter()
 
-evs = prof.key_averages(group_by_stack_n=7)
-rows = [e for e in evs if 'copy_' in e.key or 'contiguous' in e.key or 'clone' in e.key]
-rows.sort(key=lambda e: -e.self_device_time_total)
-for e in rows[:12]:
-    print('=' * 80)
-    print(f"{e.key}  calls={e.count}  device={e.self_device_time_total/1000:.3f}ms")
-    if e.stack:
-        for line in e.stack[:7]:
-            print('   ', line)
+
+class CopySpy(TorchDispatchMode):
+    def __torch_dispatch__(self, func, types, args=(), kwargs=None):
+        kwargs = kwargs or {}
+        name = str(func)
+        if any(s in name for s in ('copy_', 'clone', '_to_copy', 'cat', 'mul', 'add', 'fill', 'zero')):
+            tens = [a for a in args if isinstance(a, torch.Tensor)]
+            big = max((a.numel() for a in tens), default=0)
+            if big >= (1 << 19):
+                shp = tuple(tuple(a.shape) for a in tens[:2])
+                st = traceback.extract_stack()
+                src = '<-'.join(
+                    f"{f.filename.split('/')[-1]}:{f.lineno}" for f in reversed(st)
+                    if ('noisynet_amd' in f.filename or 'bench' in f.filename
+                        or 'mb_copies' in f.filename))[:140] or 'backward-thread'
+                log[(name, shp, src)] += 1
+        return func(*args, **kwargs)
+
+
+with CopySpy():
+    step(1000)
+torch.cuda.synchronize()
+
+for (name, shp, src), cnt in log.most_common(30):
+    print(f"x{cnt:<3} {name:<28} {str(shp):<50} {src}")

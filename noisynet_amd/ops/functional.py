@@ -56,7 +56,9 @@ class FakeQuant(torch.autograd.Function):
     def backward(ctx, grad_output):
         (x,) = ctx.saved_tensors
         if use_native(x):
-            g = ext().ste_mask(grad_output.contiguous(), x, ctx.min_value, ctx.max_value)
+            # ste_mask aligns layouts internally (x.suggest_memory_format);
+            # a plain .contiguous() here would force an NCHW round-trip
+            g = ext().ste_mask(grad_output, x, ctx.min_value, ctx.max_value)
         else:
             g = ref.fake_quant_backward(grad_output, x, ctx.min_value, ctx.max_value)
         return g, None, None, None, None
@@ -99,8 +101,8 @@ def _conv_fwd_raw(x, w, bias, stride, padding):
         if _patch_eligible(x, w, padding):
             empty = torch.empty(0, device=x.device, dtype=x.dtype)
             zero_f = torch.zeros(1, device=x.device, dtype=torch.float32)
-            y, _ = ext().conv_fwd_fused(_nhwc(x), _nhwc(w), _nhwc(w), empty,
-                                        stride, padding, 0, zero_f, 0, False)
+            y = ext().conv_fwd_fused(_nhwc(x), _nhwc(w), _nhwc(w), empty,
+                                     stride, padding, 0, zero_f, 0, False)[0]
         else:
             y = ext().conv_fwd(_nhwc(x), _nhwc(w), stride, padding)
         if bias is not None:
@@ -118,17 +120,20 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
             w2 = _nhwc(w.flip((2, 3)).transpose(0, 1))
             empty = torch.empty(0, device=g.device, dtype=g.dtype)
             zero_f = torch.zeros(1, device=g.device, dtype=torch.float32)
-            dx, _ = ext().conv_fwd_fused(_nhwc(g), w2, w2, empty, 1,
-                                         R - 1 - padding, 0, zero_f, 0, False)
+            dx = ext().conv_fwd_fused(_nhwc(g), w2, w2, empty, 1,
+                                      R - 1 - padding, 0, zero_f, 0, False)[0]
             return dx
         return ext().conv_dgrad(_nhwc(g), _nhwc(w), stride, padding,
                                 x_shape[2], x_shape[3])
     return torch.nn.grad.conv2d_input(x_shape, w, g, stride, padding)
 
 
-def _conv_wgrad_raw(g, x, stride, padding, w_shape):
+def _conv_wgrad_raw(g, x, stride, padding, w_shape, col=None):
     if use_native(g, x):
         R, S = w_shape[2], w_shape[3]
+        if col is not None and R * S > 1:
+            # flat im2col matrix shared from the forward pass
+            return ext().conv_wgrad_from_col(_nhwc(g), col, x.shape[1], R, S)
         if R * S > 1:
             # im2col-GEMM wgrad when the buffer is affordable (<2 GB):
             # one coalesced materialization pass beats per-K-chunk gather
@@ -157,7 +162,7 @@ class ConvFwd(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             gx = ConvDgrad.apply(g, w, ctx.stride, ctx.padding, x.shape)
         if ctx.needs_input_grad[1]:
-            gw = ConvWgrad.apply(g, x, ctx.stride, ctx.padding, w.shape)
+            gw = ConvWgrad.apply(g, x, ctx.stride, ctx.padding, w.shape, None)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             gb = g.sum(dim=(0, 2, 3))
         return gx, gw, gb, None, None
@@ -178,17 +183,17 @@ class ConvDgrad(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             d_g = ConvFwd.apply(gg, w, None, ctx.stride, ctx.padding)
         if ctx.needs_input_grad[1]:
-            d_w = ConvWgrad.apply(g, gg, ctx.stride, ctx.padding, w.shape)
+            d_w = ConvWgrad.apply(g, gg, ctx.stride, ctx.padding, w.shape, None)
         return d_g, d_w, None, None, None
 
 
 class ConvWgrad(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, g, x, stride, padding, w_shape):
+    def forward(ctx, g, x, stride, padding, w_shape, col=None):
         ctx.stride, ctx.padding = stride, padding
         ctx.w_shape = tuple(w_shape)
         ctx.save_for_backward(g, x)
-        return _conv_wgrad_raw(g, x, stride, padding, w_shape)
+        return _conv_wgrad_raw(g, x, stride, padding, w_shape, col)
 
     @staticmethod
     def backward(ctx, gw):
@@ -198,7 +203,7 @@ class ConvWgrad(torch.autograd.Function):
             d_g = ConvFwd.apply(x, gw, None, ctx.stride, ctx.padding)
         if ctx.needs_input_grad[1]:
             d_x = ConvDgrad.apply(g, gw, ctx.stride, ctx.padding, x.shape)
-        return d_g, d_x, None, None, None
+        return d_g, d_x, None, None, None, None
 
 
 def conv2d(x, w, bias=None, stride=1, padding=0):
@@ -242,10 +247,10 @@ class _FusedNoisyConv(torch.autograd.Function):
     def forward(ctx, x, wq, w_raw, bias, stride, padding, sigma_mode, factor,
                 want_telemetry, telemetry_out, current, power_denom):
         ctx.stride, ctx.padding = stride, padding
-        ctx.save_for_backward(x, wq)
         ctx.has_bias = bias is not None
+        col = None
         if use_native(x, wq):
-            y, tele = ext().conv_fwd_fused(
+            y, tele, col = ext().conv_fwd_fused(
                 _nhwc(x), _nhwc(wq), _nhwc(w_raw),
                 bias if bias is not None else torch.empty(0, device=x.device, dtype=x.dtype),
                 stride, padding, 1 if sigma_mode == "abs" else 2,
@@ -274,16 +279,23 @@ class _FusedNoisyConv(torch.autograd.Function):
                 telemetry_out.power = 1.2e-6 * current * sum_sigma_abs / power_denom
                 telemetry_out.nsr = sum_abs_noise / max_y
                 telemetry_out.input_sparsity = (x.detach() > 0).sum() / x.numel()
+        if col is not None and col.numel():
+            # the small-C route materialized the flat im2col matrix; keep
+            # it so backward's wgrad skips the re-materialization
+            ctx.save_for_backward(x, wq, col)
+        else:
+            ctx.save_for_backward(x, wq)
         return y
 
     @staticmethod
     def backward(ctx, g):
-        x, wq = ctx.saved_tensors
+        x, wq = ctx.saved_tensors[0], ctx.saved_tensors[1]
+        col = ctx.saved_tensors[2] if len(ctx.saved_tensors) > 2 else None
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
             gx = ConvDgrad.apply(g, wq, ctx.stride, ctx.padding, x.shape)
         if ctx.needs_input_grad[1]:
-            gw = ConvWgrad.apply(g, x, ctx.stride, ctx.padding, wq.shape)
+            gw = ConvWgrad.apply(g, x, ctx.stride, ctx.padding, wq.shape, col)
         if ctx.has_bias and ctx.needs_input_grad[3]:
             gb = g.sum(dim=(0, 2, 3))
         return (gx, gw, None, gb) + (None,) * 8

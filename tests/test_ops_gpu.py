@@ -196,7 +196,7 @@ def test_fused_conv_noise_statistics():
     factor = 0.05
     out, tele = ops.ext().conv_fwd_fused(
         cl(x), cl(wq), cl(w), torch.empty(0, device=dev(), dtype=torch.bfloat16),
-        1, 0, 1, torch.tensor([factor], device=dev()), 42, True)
+        1, 0, 1, torch.tensor([factor], device=dev()), 42, True)[:2]
     clean = F.conv2d(x.float().cpu(), w.float().cpu())
     noise = out.float().cpu() - clean
     sig = F.conv2d(x.float().cpu(), w.float().cpu().abs())
@@ -218,7 +218,7 @@ def test_fused_conv_abs2_mode():
     factor = 0.02
     out, tele = ops.ext().conv_fwd_fused(
         cl(x), cl(w), cl(w), torch.empty(0, device=dev(), dtype=torch.bfloat16),
-        1, 0, 2, torch.tensor([factor], device=dev()), 7, True)
+        1, 0, 2, torch.tensor([factor], device=dev()), 7, True)[:2]
     clean = F.conv2d(x.float().cpu(), w.float().cpu())
     noise = out.float().cpu() - clean
     aw = w.float().cpu().abs()
@@ -570,8 +570,8 @@ def test_patch_mode_conv_matches_torch():
         assert _patch_eligible(x, w, pad)
         empty = torch.empty(0, device=dev(), dtype=torch.bfloat16)
         zf = torch.zeros(1, device=dev())
-        y, _ = ops.ext().conv_fwd_fused(cl(x), cl(w), cl(w), empty,
-                                        stride, pad, 0, zf, 0, False)
+        y = ops.ext().conv_fwd_fused(cl(x), cl(w), cl(w), empty,
+                                     stride, pad, 0, zf, 0, False)[0]
         ye = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad)
         rel = (y.float().cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
         assert rel < 0.02, rel
@@ -715,8 +715,8 @@ def test_patch_conv_fp32_exact():
     assert _patch_eligible(x, w, 1)
     empty = torch.empty(0, device=dev())
     zf = torch.zeros(1, device=dev())
-    y, _ = ops.ext().conv_fwd_fused(cl(x), cl(w), cl(w), empty, 1, 1, 0, zf,
-                                    0, False)
+    y = ops.ext().conv_fwd_fused(cl(x), cl(w), cl(w), empty, 1, 1, 0, zf,
+                                 0, False)[0]
     ye = F.conv2d(x.cpu(), w.cpu(), None, 1, 1)
     rel = (y.cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
     assert rel < 1e-5, rel
