@@ -24,14 +24,26 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
   int c = blockIdx.x * 64 + (threadIdx.x & 63);
   int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
   if (c >= C) return;
-  float s = 0.0f, sq = 0.0f;
-  for (int64_t r = rstart; r < rows; r += gridDim.y * (kBlock / 64)) {
-    float v = to_f32(x[r * C + c]);
-    s += v;
-    sq += v * v;
+  // 4 independent accumulator pairs keep 4 loads in flight per thread
+  // (the single-accumulator loop was latency-bound at ~1/6 of HBM peak)
+  const int64_t st = (int64_t)gridDim.y * (kBlock / 64);
+  float s[4] = {}, sq[4] = {};
+  int64_t r = rstart;
+  for (; r + 3 * st < rows; r += 4 * st) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float v = to_f32(x[(r + j * st) * C + c]);
+      s[j] += v;
+      sq[j] += v * v;
+    }
   }
-  atomicAdd(&sum[c], s);
-  atomicAdd(&sumsq[c], sq);
+  for (; r < rows; r += st) {
+    float v = to_f32(x[r * C + c]);
+    s[0] += v;
+    sq[0] += v * v;
+  }
+  atomicAdd(&sum[c], (s[0] + s[1]) + (s[2] + s[3]));
+  atomicAdd(&sumsq[c], (sq[0] + sq[1]) + (sq[2] + sq[3]));
 }
 
 template <typename T>
@@ -138,19 +150,27 @@ __global__ void bn_act_bwd_reduce_kernel(
   int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
   if (c >= C) return;
   float m = mean[c], is = invstd[c];
-  float s_g = 0.0f, s_gx = 0.0f;
-  for (int64_t r = rstart; r < rows; r += gridDim.y * (kBlock / 64)) {
+  // unrolled x4: see bn_stats_kernel
+  const int64_t st = (int64_t)gridDim.y * (kBlock / 64);
+  float s_g[4] = {}, s_gx[4] = {};
+  auto body = [&](int64_t r, int j) {
     int64_t i = r * C + c;
     float yv = to_f32(y[i]);
     float mask = 1.0f;
     if (do_relu && yv <= 0.0f) mask = 0.0f;
     if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
     float gv = to_f32(g[i]) * mask;
-    s_g += gv;
-    s_gx += gv * (to_f32(x[i]) - m) * is;
+    s_g[j] += gv;
+    s_gx[j] += gv * (to_f32(x[i]) - m) * is;
+  };
+  int64_t r = rstart;
+  for (; r + 3 * st < rows; r += 4 * st) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) body(r + j * st, j);
   }
-  atomicAdd(&sum_g[c], s_g);
-  atomicAdd(&sum_gx[c], s_gx);
+  for (; r < rows; r += st) body(r, 0);
+  atomicAdd(&sum_g[c], (s_g[0] + s_g[1]) + (s_g[2] + s_g[3]));
+  atomicAdd(&sum_gx[c], (s_gx[0] + s_gx[1]) + (s_gx[2] + s_gx[3]));
 }
 
 template <typename T, bool TRAIN>

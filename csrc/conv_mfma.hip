@@ -1652,10 +1652,20 @@ torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
   // library GEMM (measured: conv2-type 0.87 ms blas vs 1.19 ms ours).
   auto gy2 = gy.permute({0, 2, 3, 1}).reshape({M, K});  // raw view, free
   torch::Tensor dw_flat;
-  if (K * cols_p <= 65536 && M >= 262144)
-    dw_flat = linear_wgrad(gy2, col);                       // [K, cols_p]
-  else
+  if (K * cols_p <= 65536 && M >= 262144) {
+    int64_t K8 = (K + 7) & ~7;
+    if (K8 != K && gy.scalar_type() != torch::kFloat32) {
+      // odd K (e.g. 65 output channels) leaves gy rows 2-byte-misaligned,
+      // forcing scalar staging loads; one zero-pad pass restores 16-B
+      // vector loads (the extra dw rows are sliced off below)
+      auto gy2p = at::constant_pad_nd(gy2, {0, K8 - K}, 0);
+      dw_flat = linear_wgrad(gy2p, col).narrow(0, 0, K);
+    } else {
+      dw_flat = linear_wgrad(gy2, col);                     // [K, cols_p]
+    }
+  } else {
     dw_flat = at::matmul(gy2.t(), col).to(col.scalar_type());
+  }
   // un-pad the flat row: [K, cols_p] -> [K, R*S*C] -> logical [K,C,R,S] cl
   auto dw = dw_flat.narrow(1, 0, rsc)
                 .view({K, R, S, C})

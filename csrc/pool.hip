@@ -37,11 +37,15 @@ __global__ void maxpool2x2_fwd_kernel(const T* __restrict__ x, T* __restrict__ y
   }
 }
 
+// Windows are non-overlapping, so each thread owns one (window, channel)
+// and writes ALL four input cells (grad to the argmax cell, zero to the
+// rest): full coverage, no separate zero-fill pass over gx.
 template <typename T>
 __global__ void maxpool2x2_bwd_kernel(const T* __restrict__ g,
                                       const uint8_t* __restrict__ code,
                                       T* __restrict__ gx, int64_t n_out, int C,
                                       int H, int W, int OH, int OW) {
+  const T zero = from_f32<T>(0.0f);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_out;
        i += (int64_t)gridDim.x * blockDim.x) {
     int c = (int)(i % C);
@@ -51,8 +55,22 @@ __global__ void maxpool2x2_bwd_kernel(const T* __restrict__ g,
     int oh = (int)(t % OH);
     int64_t nb = t / OH;
     int k = code[i];
-    int dh = k >> 1, dw = k & 1;
-    gx[((nb * H + 2 * oh + dh) * W + 2 * ow + dw) * C + c] = g[i];
+    T gv = g[i];
+    T* base = gx + ((nb * H + 2 * oh) * W + 2 * ow) * C + c;
+    base[0] = (k == 0) ? gv : zero;
+    base[C] = (k == 1) ? gv : zero;
+    base[(int64_t)W * C] = (k == 2) ? gv : zero;
+    base[(int64_t)W * C + C] = (k == 3) ? gv : zero;
+    // odd input dims: the last column/row belong to no window; zero them
+    if ((W & 1) && ow == OW - 1) {
+      base[2 * C] = zero;
+      base[(int64_t)W * C + 2 * C] = zero;
+    }
+    if ((H & 1) && oh == OH - 1) {
+      base[2 * (int64_t)W * C] = zero;
+      base[2 * (int64_t)W * C + C] = zero;
+      if ((W & 1) && ow == OW - 1) base[2 * (int64_t)W * C + 2 * C] = zero;
+    }
   }
 }
 
@@ -87,12 +105,10 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
                              int64_t W) {
   TORCH_CHECK(g.dim() == 4 && g.is_contiguous(at::MemoryFormat::ChannelsLast));
   int N = (int)g.size(0), C = (int)g.size(1), OH = (int)g.size(2), OW = (int)g.size(3);
-  // NOTE: the zeros() factory drops memory_format from TensorOptions (the
-  // returned grad then forces an NHWC re-copy in every conv backward);
-  // empty()+zero_() keeps the channels_last layout.
+  // empty() (not zeros(): that factory drops memory_format from
+  // TensorOptions) -- the kernel writes every element, no fill needed.
   auto gx = torch::empty({N, C, (int)H, (int)W},
                          g.options().memory_format(at::MemoryFormat::ChannelsLast));
-  gx.zero_();
   int64_t n_out = (int64_t)N * C * OH * OW;
   int blocks = (int)std::min<int64_t>((n_out + kBlock - 1) / kBlock, 8192);
   NN_DISPATCH(g.scalar_type(),
