@@ -393,6 +393,186 @@ void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
 }
 
 // --------------------------------------------------------------------------
+// Small-K fused GEMM: out[M, K] = x2[M, Kc] @ w[K, Kc]^T (+sigma +noise),
+// for Kc <= 128 (contraction fits LDS whole) and K <= 96 (one block tile
+// covers every output channel). This is the conv1-as-im2col shape
+// ([1.6M, 80] @ [65, 80]^T at batch 2048): the generic kernel pays a
+// sync per 32-deep k-step and re-reads the A rows once per 32-wide
+// n-tile; here the whole weight set is staged ONCE per block (it stays
+// LDS-resident across the block's grid-stride m-tiles) and each A tile
+// is read exactly once, giving one global pass over x2.
+// Wave layout: 4 waves, wm = wave>>1 covers m 0..63, wn = wave&1 owns
+// output fragments f in {wn, wn+2, wn+4} (16 channels each, K<=96).
+// 16-bit dtypes only (fp32 LDS image would not fit).
+// --------------------------------------------------------------------------
+
+template <typename T, bool WANT_Y, int SIGMA_MODE, bool TELEM, bool BIAS>
+__global__ __launch_bounds__(kBlock)
+void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
+                            const T* __restrict__ wraw,
+                            const float* __restrict__ bias,
+                            T* __restrict__ out, int64_t M, int K, int Kc,
+                            const float* __restrict__ factor_p, uint64_t seed,
+                            float* __restrict__ telem) {
+  const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
+  constexpr int STR = Mma<T>::STRIDE;
+  const int CH = (Kc + 31) >> 5;     // 32-wide contraction chunks (<= 4),
+                                     // last one zero-padded when Kc % 32
+  const int KROWS = 96;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* a_lds = smem;                              // CH * 64 rows
+  char* b_lds = smem + (size_t)CH * BM * STR;      // CH * 96 rows (wq)
+  char* c_lds = b_lds + (size_t)CH * KROWS * STR;  // CH * 96 rows (sigma w)
+  char* d_lds = c_lds + (size_t)CH * KROWS * STR;  // CH * 96 rows (|w|)
+
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+  const int nfrag = (K + 15) >> 4;                 // <= 6
+  const int nf_w = (nfrag - wn + 1) >> 1;          // fragments of this wave
+
+  // stage the full weight set once (rows k < K, zero-padded to 96)
+  for (int rb = 0; rb < KROWS; rb += BM) {
+    int k = rb + row;
+    if (k >= KROWS) continue;  // rows 96..127 of the second pass
+    for (int ch = 0; ch < CH; ++ch) {
+      int col0 = ch * 32 + seg * 8;
+      float vals[8];
+      const T* pw = wq + (int64_t)k * Kc + col0;
+      const T* pr = wraw + (int64_t)k * Kc + col0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = (WANT_Y && k < K && col0 + j < Kc) ? to_f32(pw[j]) : 0.0f;
+        vals[j] = v;
+      }
+      if (WANT_Y)
+        Mma<T>::store8(b_lds + (size_t)ch * KROWS * STR, k, seg * 8, vals);
+      if (SIGMA_MODE > 0) {
+        float sv[8], tv[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = (k < K && col0 + j < Kc) ? fabsf(to_f32(pr[j])) : 0.0f;
+          tv[j] = v;
+          sv[j] = (SIGMA_MODE == 2) ? v * v + v : v;
+        }
+        Mma<T>::store8(c_lds + (size_t)ch * KROWS * STR, k, seg * 8, sv);
+        if (TELEM && SIGMA_MODE == 2)
+          Mma<T>::store8(d_lds + (size_t)ch * KROWS * STR, k, seg * 8, tv);
+      }
+    }
+  }
+  __syncthreads();
+
+  float t_sum_sigma = 0.0f, t_sum_noise = 0.0f, t_max_y = -INFINITY;
+  int64_t mtiles = (M + BM - 1) / BM;
+
+  for (int64_t mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
+    int64_t m0 = mt * BM;
+    // stage A tile: one vector load per (row, chunk, seg)
+    {
+      int64_t m = m0 + row;
+      for (int ch = 0; ch < CH; ++ch) {
+        int col0 = ch * 32 + seg * 8;
+        if (m < M && col0 + 8 <= Kc && sizeof(T) == 2) {
+          *(bf16x8*)(a_lds + (size_t)ch * BM * STR + row * STR + seg * 16) =
+              *(const bf16x8*)(x2 + m * Kc + col0);
+        } else {
+          float vals[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vals[j] = (m < M && col0 + j < Kc)
+                          ? to_f32(x2[m * Kc + col0 + j]) : 0.0f;
+          Mma<T>::store8(a_lds + (size_t)ch * BM * STR, row, seg * 8, vals);
+        }
+      }
+    }
+    __syncthreads();
+
+    f32x4 acc[2][3] = {};
+    f32x4 sacc[2][3] = {};
+    f32x4 tacc[2][3] = {};
+    for (int ch = 0; ch < CH; ++ch) {
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        auto a = Mma<T>::load(a_lds + (size_t)ch * BM * STR,
+                              wm * 32 + fm * 16 + (lane & 15), lane);
+#pragma unroll
+        for (int fi = 0; fi < 3; ++fi) {
+          if (fi >= nf_w) break;  // constant-trip unroll, predicated tail
+          int brow = (wn + 2 * fi) * 16 + (lane & 15);
+          if (WANT_Y) {
+            auto b = Mma<T>::load(b_lds + (size_t)ch * KROWS * STR, brow, lane);
+            Mma<T>::mma(a, b, acc[fm][fi]);
+          }
+          if (SIGMA_MODE > 0) {
+            auto c = Mma<T>::load(c_lds + (size_t)ch * KROWS * STR, brow, lane);
+            Mma<T>::mma(a, c, sacc[fm][fi]);
+          }
+          if (TELEM && SIGMA_MODE == 2) {
+            auto d = Mma<T>::load(d_lds + (size_t)ch * KROWS * STR, brow, lane);
+            Mma<T>::mma(a, d, tacc[fm][fi]);
+          }
+        }
+      }
+    }
+
+    // epilogue
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int fi = 0; fi < 3; ++fi) {
+        if (fi >= nf_w) break;
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          int64_t m = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
+          int k = (wn + 2 * fi) * 16 + (lane & 15);
+          if (m < M && k < K) {
+            float y = WANT_Y ? acc[fm][fi][reg] : 0.0f;
+            if (BIAS) y += bias[k];
+            float v = y;
+            if (SIGMA_MODE > 0) {
+              float sig = fmaxf(sacc[fm][fi][reg], 0.0f);
+              float noise = gauss1(seed, (uint64_t)(m * K + k))
+                            * sqrtf(factor * sig);
+              v = y + noise;
+              if (TELEM) {
+                t_sum_noise += fabsf(noise);
+                t_max_y = fmaxf(t_max_y, y);
+                t_sum_sigma += (SIGMA_MODE == 2) ? tacc[fm][fi][reg]
+                                                 : sacc[fm][fi][reg];
+              }
+            }
+            out[m * K + k] = from_f32<T>(v);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (TELEM && SIGMA_MODE > 0) {
+    __shared__ float red[4];
+    float s0 = block_sum(t_sum_sigma, red);
+    __syncthreads();
+    float s1 = block_sum(t_sum_noise, red);
+    __syncthreads();
+    float m0v = wave_max(t_max_y);
+    __shared__ float redm[4];
+    if ((threadIdx.x & (WAVE - 1)) == 0) redm[threadIdx.x / WAVE] = m0v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      atomicAdd(&telem[0], s0);
+      atomicAdd(&telem[1], s1);
+      atomic_max_f32(&telem[2],
+                     fmaxf(fmaxf(redm[0], redm[1]), fmaxf(redm[2], redm[3])));
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
 // Dgrad: dx[n,ih,iw,c] = sum_{r,s,k} g[n,oh,ow,k] * w[k,r,s,c]
 // with oh = (ih + pad - r)/stride when divisible. Implicit GEMM over taps:
 // A rows = input pixels, contraction = output channels K, B = wt[r,s,c,k]
@@ -930,12 +1110,99 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
   return dw_f.to(x.scalar_type());
 }
 
+namespace {
+
+// host side of the small-K fused GEMM (see conv_fwd_smallk_kernel)
+bool smallk_eligible(const torch::Tensor& x, int64_t K) {
+  return x.size(1) <= 128 && (x.size(1) & 7) == 0 && K <= 96 &&
+         x.element_size() == 2;
+}
+
+std::vector<torch::Tensor> linear_fwd_fused_smallk(
+    torch::Tensor x, torch::Tensor wq, torch::Tensor wraw, torch::Tensor bias,
+    int64_t sigma_mode, torch::Tensor factor, int64_t seed, bool telem,
+    bool want_y) {
+  int64_t M = x.size(0);
+  int K = (int)wraw.size(0);
+  int Kc = (int)x.size(1);
+  int CH = (Kc + 31) >> 5;
+  auto out = torch::empty({M, (int64_t)K}, x.options());
+  bool has_bias = bias.numel() > 0;
+  torch::Tensor bias_f;
+  if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
+  auto tele = telem
+      ? [&] {
+          auto t = torch::zeros({3}, x.options().dtype(torch::kFloat32));
+          t[2] = -std::numeric_limits<float>::infinity();
+          return t;
+        }()
+      : torch::empty({0}, x.options().dtype(torch::kFloat32));
+  auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
+  int64_t mtiles = (M + BM - 1) / BM;
+  int blocks = (int)std::min<int64_t>(mtiles, 8192);
+  NN_DISPATCH(x.scalar_type(), "linear_fwd_fused_smallk", [&] {
+    using T = typename DevT<scalar_t>::type;
+    bool d_live = telem && sigma_mode == 2;
+    size_t lds = (size_t)CH * Mma<T>::STRIDE * (BM + 96 * (d_live ? 3 : 2));
+    auto stream = c10::hip::getCurrentHIPStream();
+    const T* xp = (const T*)x.data_ptr();
+    const T* wqp = (const T*)wq.data_ptr();
+    const T* wrp = (const T*)wraw.data_ptr();
+    const float* bp = has_bias ? bias_f.data_ptr<float>() : nullptr;
+    T* op = (T*)out.data_ptr();
+    float* tp = telem ? tele.data_ptr<float>() : nullptr;
+    const float* f = factor_f.data_ptr<float>();
+    uint64_t sd = (uint64_t)seed;
+    auto launch = [&](auto wy, auto sm, auto tl, auto bi) {
+      auto* kfn = &conv_fwd_smallk_kernel<T, decltype(wy)::value,
+                                          decltype(sm)::value,
+                                          decltype(tl)::value,
+                                          decltype(bi)::value>;
+      if (lds > 64 * 1024)
+        hipFuncSetAttribute((const void*)kfn,
+                            hipFuncAttributeMaxDynamicSharedMemorySize,
+                            (int)lds);
+      hipLaunchKernelGGL(kfn, dim3(blocks), dim3(kBlock), lds, stream, xp,
+                         wqp, wrp, bp, op, M, K, Kc, f, sd, tp);
+    };
+    using TT = std::true_type; using FF = std::false_type;
+    using S0 = std::integral_constant<int, 0>;
+    using S1 = std::integral_constant<int, 1>;
+    using S2 = std::integral_constant<int, 2>;
+    if (want_y) {
+      if (sigma_mode == 0) {
+        if (has_bias) launch(TT{}, S0{}, FF{}, TT{});
+        else launch(TT{}, S0{}, FF{}, FF{});
+      } else if (sigma_mode == 1) {
+        if (telem) { if (has_bias) launch(TT{}, S1{}, TT{}, TT{}); else launch(TT{}, S1{}, TT{}, FF{}); }
+        else { if (has_bias) launch(TT{}, S1{}, FF{}, TT{}); else launch(TT{}, S1{}, FF{}, FF{}); }
+      } else {
+        if (telem) { if (has_bias) launch(TT{}, S2{}, TT{}, TT{}); else launch(TT{}, S2{}, TT{}, FF{}); }
+        else { if (has_bias) launch(TT{}, S2{}, FF{}, TT{}); else launch(TT{}, S2{}, FF{}, FF{}); }
+      }
+    } else {
+      if (sigma_mode == 1) {
+        if (telem) launch(FF{}, S1{}, TT{}, FF{}); else launch(FF{}, S1{}, FF{}, FF{});
+      } else {
+        if (telem) launch(FF{}, S2{}, TT{}, FF{}); else launch(FF{}, S2{}, FF{}, FF{});
+      }
+    }
+  });
+  HIP_CHECK_LAST();
+  return {out, tele};
+}
+
+}  // namespace
+
 std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
                                             torch::Tensor wraw,
                                             torch::Tensor bias,
                                             int64_t sigma_mode,
                                             torch::Tensor factor,
                                             int64_t seed, bool telem) {
+  if (smallk_eligible(x, wraw.size(0)))
+    return linear_fwd_fused_smallk(x, wq, wraw, bias, sigma_mode, factor,
+                                   seed, telem, /*want_y=*/true);
   auto g = linear_geom(x, wraw.size(0));
   auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
   bool has_bias = bias.numel() > 0;
@@ -993,6 +1260,11 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
                                                    torch::Tensor factor,
                                                    int64_t seed,
                                                    bool telem) {
+  if (smallk_eligible(x, wraw.size(0))) {
+    auto empty_bias = torch::empty({0}, x.options());
+    return linear_fwd_fused_smallk(x, wraw, wraw, empty_bias, sigma_mode,
+                                   factor, seed, telem, /*want_y=*/false);
+  }
   auto g = linear_geom(x, wraw.size(0));
   auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
   // telemetry buffer only when requested (steady-state training has
@@ -1583,6 +1855,51 @@ __global__ void im2col_flat_kernel(const T* __restrict__ x,
   }
 }
 
+// one-pass column pad: out[m, 0:K] = in[m, 0:K], out[m, K:K8] = 0.
+// (at::constant_pad_nd fills the WHOLE output then copies -- two full
+// passes, ~0.55 ms at [1.6M, 65->72]; this writes each chunk once.)
+template <typename T>
+__global__ void pad_cols_kernel(const T* __restrict__ in, T* __restrict__ out,
+                                int64_t nchunks_total, int nchunks, int K) {
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < nchunks_total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int seg = (int)(idx % nchunks);
+    int64_t m = idx / nchunks;
+    const T* src = in + m * K + seg * 8;
+    int base = seg * 8;
+    T vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      vals[j] = (base + j < K) ? src[j] : from_f32<T>(0.0f);
+    T* dst = out + (m * (int64_t)nchunks + seg) * 8;
+    if (sizeof(T) == 2) {
+      *(bf16x8*)dst = *(bf16x8*)vals;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[j] = vals[j];
+    }
+  }
+}
+
+torch::Tensor pad_cols8(const torch::Tensor& in, int64_t K8) {
+  TORCH_CHECK(in.dim() == 2 && in.is_contiguous());
+  int64_t M = in.size(0);
+  int K = (int)in.size(1);
+  auto out = torch::empty({M, K8}, in.options());
+  int nchunks = (int)(K8 >> 3);
+  int64_t total = M * nchunks;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+  NN_DISPATCH(in.scalar_type(), "pad_cols8", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((pad_cols_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)in.data_ptr(), (T*)out.data_ptr(), total,
+                       nchunks, K);
+  });
+  HIP_CHECK_LAST();
+  return out;
+}
+
 }  // namespace
 
 // Materialize the im2col matrix [M, cols_p] with the FLAT column layout:
@@ -1658,7 +1975,7 @@ torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
       // odd K (e.g. 65 output channels) leaves gy rows 2-byte-misaligned,
       // forcing scalar staging loads; one zero-pad pass restores 16-B
       // vector loads (the extra dw rows are sliced off below)
-      auto gy2p = at::constant_pad_nd(gy2, {0, K8 - K}, 0);
+      auto gy2p = pad_cols8(gy2.contiguous(), K8);
       dw_flat = linear_wgrad(gy2p, col).narrow(0, 0, K);
     } else {
       dw_flat = linear_wgrad(gy2, col);                     // [K, cols_p]
