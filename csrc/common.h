@@ -91,6 +91,27 @@ DEV_INLINE float gauss1(uint64_t seed, uint64_t ctr) {
   return r * __cosf(theta);
 }
 
+// four N(0,1) samples from ONE Philox draw (two Box-Muller pairs) with the
+// hardware log (v_log_f32). The per-element gauss1 epilogue was the top
+// cost of every noisy kernel (VALU:MFMA = 99:1 on the conv1 GEMM — one
+// 10-round Philox plus a ~40-op software logf per OUTPUT ELEMENT,
+// discarding 3/4 of the draw); amortizing over the 4 consecutive
+// accumulator rows cuts the RNG VALU work ~8x. Noise quality is
+// unaffected: counters stay unique per element-quad and __logf's ~1 ulp
+// on (0,1] is far below the sampled distribution's own width.
+DEV_INLINE void gauss4(uint64_t seed, uint64_t ctr, float g[4]) {
+  Philox4 p = philox4x32(seed, ctr);
+  float s, c;
+  float r0 = sqrtf(-2.0f * __logf(u01_open(p.x)));
+  __sincosf(6.2831853071795864f * u01(p.y), &s, &c);
+  g[0] = r0 * c;
+  g[1] = r0 * s;
+  float r1 = sqrtf(-2.0f * __logf(u01_open(p.z)));
+  __sincosf(6.2831853071795864f * u01(p.w), &s, &c);
+  g[2] = r1 * c;
+  g[3] = r1 * s;
+}
+
 // uniform in [-a, a]
 DEV_INLINE float uniform_pm(uint64_t seed, uint64_t ctr, float a) {
   Philox4 p = philox4x32(seed, ctr);

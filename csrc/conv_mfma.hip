@@ -347,18 +347,20 @@ void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
   for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
     for (int fn = 0; fn < 2; ++fn) {
+      int64_t mb = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4);
+      int n = n0 + wn * 32 + fn * 16 + (lane & 15);
+      float g4[4];
+      if (SIGMA_MODE > 0) gauss4(seed, (uint64_t)(mb * g.K + n), g4);
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        int64_t m = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
-        int n = n0 + wn * 32 + fn * 16 + (lane & 15);
+        int64_t m = mb + reg;
         if (m < g.M && n < g.K) {
           float y = WANT_Y ? acc[fm][fn][reg] : 0.0f;
           if (BIAS) y += bias[n];
           float v = y;
           if (SIGMA_MODE > 0) {
             float sig = fmaxf(sacc[fm][fn][reg], 0.0f);
-            float noise = gauss1(seed, (uint64_t)(m * g.K + n))
-                          * sqrtf(factor * sig);
+            float noise = g4[reg] * sqrtf(factor * sig);
             v = y + noise;
             if (TELEM) {
               t_sum_noise += fabsf(noise);
@@ -525,18 +527,20 @@ void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
 #pragma unroll
       for (int fi = 0; fi < 3; ++fi) {
         if (fi >= nf_w) break;
+        int64_t mb = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4);
+        int k = (wn + 2 * fi) * 16 + (lane & 15);
+        float g4[4];
+        if (SIGMA_MODE > 0) gauss4(seed, (uint64_t)(mb * K + k), g4);
 #pragma unroll
         for (int reg = 0; reg < 4; ++reg) {
-          int64_t m = m0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
-          int k = (wn + 2 * fi) * 16 + (lane & 15);
+          int64_t m = mb + reg;
           if (m < M && k < K) {
             float y = WANT_Y ? acc[fm][fi][reg] : 0.0f;
             if (BIAS) y += bias[k];
             float v = y;
             if (SIGMA_MODE > 0) {
               float sig = fmaxf(sacc[fm][fi][reg], 0.0f);
-              float noise = gauss1(seed, (uint64_t)(m * K + k))
-                            * sqrtf(factor * sig);
+              float noise = g4[reg] * sqrtf(factor * sig);
               v = y + noise;
               if (TELEM) {
                 t_sum_noise += fabsf(noise);
@@ -1511,11 +1515,15 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
       for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
         for (int fn = 0; fn < 2; ++fn) {
+          int mb_local = m0 + ms * BM + wm * 32 + fm * 16 + 4 * (lane >> 4);
+          int k = n0 + wn * 32 + fn * 16 + (lane & 15);
+          float g4[4];
+          if (SIGMA_MODE > 0)
+            gauss4(seed,
+                   (uint64_t)(((int64_t)n * p.MI + mb_local) * g.K + k), g4);
 #pragma unroll
           for (int reg = 0; reg < 4; ++reg) {
-            int m_local = m0 + ms * BM + wm * 32 + fm * 16
-                          + 4 * (lane >> 4) + reg;
-            int k = n0 + wn * 32 + fn * 16 + (lane & 15);
+            int m_local = mb_local + reg;
             if (m_local < p.MI && k < g.K) {
               int64_t m = (int64_t)n * p.MI + m_local;
               float y = WANT_Y ? acc[ms][fm][fn][reg] : 0.0f;
@@ -1523,8 +1531,7 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
               float v = y;
               if (SIGMA_MODE > 0) {
                 float sig = fmaxf(sacc[ms][fm][fn][reg], 0.0f);
-                float noise = gauss1(seed, (uint64_t)(m * g.K + k))
-                              * sqrtf(factor * sig);
+                float noise = g4[reg] * sqrtf(factor * sig);
                 v = y + noise;
                 if (TELEM) {
                   t_sum_noise += fabsf(noise);
