@@ -1422,6 +1422,51 @@ DEV_INLINE void stage_wpad_row(char* lds, const T* __restrict__ w,
   Mma<T>::store8(lds, row, seg * 8, vals);
 }
 
+
+// register-pipelined weight staging: load the next chunk's values into
+// registers during the MFMA phase (the synchronous stage->barrier->mma
+// structure exposed the full global-load latency every round; PMC showed
+// 65% of wave-cycles parked). One wraw load feeds both the sigma and the
+// telemetry |w| stores.
+template <typename T>
+DEV_INLINE void load_wrow_regs(T dst[8], const T* __restrict__ w,
+                               const ConvGeom g, const PatchGeom p, int n0,
+                               int r, int ck) {
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  int k = n0 + row;
+  int c0 = ck + seg * 8;
+  if (k < g.K && sizeof(T) == 2 && c0 + 8 <= p.Kr) {
+    *(bf16x8*)dst = *(const bf16x8*)(w + ((int64_t)k * g.R + r) * p.Kr + c0);
+    return;
+  }
+  const T* pw = w + ((int64_t)k * g.R + r) * p.Kr;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    dst[j] = (k < g.K && c0 + j < p.Kr) ? pw[c0 + j] : from_f32<T>(0.0f);
+}
+
+template <typename T, bool ABS_TRANSFORM, int SIGMA_MODE>
+DEV_INLINE void store_wrow_regs(char* lds, const T src[8]) {
+  int row = threadIdx.x >> 2;
+  int seg = threadIdx.x & 3;
+  if (!ABS_TRANSFORM && sizeof(T) == 2) {
+    *(bf16x8*)(lds + row * Mma<T>::STRIDE + seg * 16) = *(const bf16x8*)src;
+    return;
+  }
+  float vals[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float v = to_f32(src[j]);
+    if (ABS_TRANSFORM) {
+      v = fabsf(v);
+      if (SIGMA_MODE == 2) v = v * v + v;
+    }
+    vals[j] = v;
+  }
+  Mma<T>::store8(lds, row, seg * 8, vals);
+}
+
 template <typename T, bool WANT_Y, int SIGMA_MODE, bool TELEM, bool BIAS>
 __global__ __launch_bounds__(kBlock)
 void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
@@ -1452,6 +1497,12 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
 
   float t_sum_sigma = 0.0f, t_sum_noise = 0.0f, t_max_y = -INFINITY;
 
+  const int CHW = (p.Kr + BK - 1) / BK;
+  const int ROUNDS = g.R * CHW;
+  T breg[8], creg[8];
+  if (WANT_Y) load_wrow_regs(breg, wq, g, p, n0, 0, 0);
+  if (SIGMA_MODE > 0) load_wrow_regs(creg, wraw, g, p, n0, 0, 0);
+
   // two M-subtiles share each staged weight tile (halves barrier count)
   for (int m0 = 0; m0 < p.MI; m0 += 2 * BM) {
     f32x4 acc[2][2][2] = {};
@@ -1459,14 +1510,49 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
     f32x4 tacc[2][2][2] = {};
     const int nsub = (m0 + BM < p.MI) ? 2 : 1;
 
-    for (int r = 0; r < g.R; ++r) {
-      for (int ck = 0; ck < p.Kr; ck += BK) {
-        if (WANT_Y) stage_wpad_row<T, false, 0>(b_lds, wq, g, p, n0, r, ck);
+    // hoist the per-fragment pixel decode: ow/oh (integer divisions) and
+    // the patch base offset are invariant across the (r, ck) rounds
+    int a_ohs[2][2];      // oh*stride - pad, or INT_MIN when m >= MI
+    int a_base[2][2];     // ((ohs*Wp + ow*stride) * C_pad) element offset
+#pragma unroll
+    for (int ms = 0; ms < 2; ++ms)
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        int m_local = m0 + ms * BM + wm * 32 + fm * 16 + (lane & 15);
+        if (m_local < p.MI) {
+          int ow = m_local % g.OW;
+          int oh = m_local / g.OW;
+          int ohs = oh * g.stride - g.pad;
+          a_ohs[ms][fm] = ohs;
+          a_base[ms][fm] = (ohs * p.Wp + ow * g.stride) * p.C_pad;
+        } else {
+          a_ohs[ms][fm] = INT_MIN / 2;
+          a_base[ms][fm] = 0;
+        }
+      }
+    const int rowpitch = p.Wp * p.C_pad;
+
+    int r = 0, ck = 0;
+    for (int t = 0; t < ROUNDS; ++t) {
+      {
+        if (WANT_Y) store_wrow_regs<T, false, 0>(b_lds, breg);
         if (SIGMA_MODE > 0)
-          stage_wpad_row<T, true, SIGMA_MODE>(c_lds, wraw, g, p, n0, r, ck);
+          store_wrow_regs<T, true, SIGMA_MODE>(c_lds, creg);
         if (TELEM && SIGMA_MODE == 2)
-          stage_wpad_row<T, true, 1>(d_lds, wraw, g, p, n0, r, ck);
+          store_wrow_regs<T, true, 1>(d_lds, creg);
         __syncthreads();
+        // prefetch the next round's weights while the MFMAs run
+        int rn = r, ckn = ck + BK;
+        if (ckn >= p.Kr) { ckn = 0; ++rn; }
+        bool more = true;
+        if (t + 1 == ROUNDS) {
+          rn = 0; ckn = 0;
+          more = (m0 + 2 * BM) < p.MI;  // wraps for the next m-tile pass
+        }
+        if (more) {
+          if (WANT_Y) load_wrow_regs(breg, wq, g, p, n0, rn, ckn);
+          if (SIGMA_MODE > 0) load_wrow_regs(creg, wraw, g, p, n0, rn, ckn);
+        }
         typename Mma<T>::frag bfrag[2], sfrag[2], tfrag[2];
 #pragma unroll
         for (int fn = 0; fn < 2; ++fn) {
@@ -1481,18 +1567,12 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
           if (ms == 1 && nsub == 1) break;
 #pragma unroll
           for (int fm = 0; fm < 2; ++fm) {
-            // per-lane patch fragment: 8 contiguous (s,c) at filter row r
-            int m_local = m0 + ms * BM + wm * 32 + fm * 16 + (lane & 15);
-            // span start for this row's 32 contraction elements at tap row r
+            // per-lane patch fragment: 8 contiguous (s,c) at filter row r;
+            // span start from the hoisted base (no divisions in the loop)
+            int ih = a_ohs[ms][fm] + r;
             int64_t off = zero_off;
-            if (m_local < p.MI) {
-              int ow = m_local % g.OW;
-              int oh = m_local / g.OW;
-              int ih = oh * g.stride - g.pad + r;
-              if (ih >= 0 && ih < g.H)
-                off = (((int64_t)ih * p.Wp + ow * g.stride) * p.C_pad + ck)
-                      * EB;
-            }
+            if (ih >= 0 && ih < g.H)
+              off = (int64_t)(a_base[ms][fm] + r * rowpitch + ck) * EB;
             auto a = Mma<T>::load_span(patch + off, lane);
 #pragma unroll
             for (int fn = 0; fn < 2; ++fn) {
@@ -1505,6 +1585,8 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
         }
         __syncthreads();
       }
+      ck += BK;
+      if (ck >= p.Kr) { ck = 0; ++r; }
     }
 
     // epilogue for both m-subtiles
