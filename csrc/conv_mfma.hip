@@ -471,27 +471,45 @@ void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
   float t_sum_sigma = 0.0f, t_sum_noise = 0.0f, t_max_y = -INFINITY;
   int64_t mtiles = (M + BM - 1) / BM;
 
+  // A tiles are register-prefetched: the next tile's global loads issue
+  // during the current tile's MFMA+epilogue (PMC: 41% of wave-cycles
+  // were parked on the synchronous A-load at 2 blocks/CU)
+  T areg[4][8];
+  auto load_a = [&](int64_t mt_) {
+    int64_t m = mt_ * BM + row;
+#pragma unroll
+    for (int ch = 0; ch < 4; ++ch) {
+      if (ch >= CH) break;
+      int col0 = ch * 32 + seg * 8;
+      if (m < M && col0 + 8 <= Kc && sizeof(T) == 2) {
+        *(bf16x8*)areg[ch] = *(const bf16x8*)(x2 + m * Kc + col0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          areg[ch][j] = (m < M && col0 + j < Kc)
+                            ? x2[m * Kc + col0 + j] : from_f32<T>(0.0f);
+      }
+    }
+  };
+  if (blockIdx.x < mtiles) load_a(blockIdx.x);
+
   for (int64_t mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
     int64_t m0 = mt * BM;
-    // stage A tile: one vector load per (row, chunk, seg)
-    {
-      int64_t m = m0 + row;
-      for (int ch = 0; ch < CH; ++ch) {
-        int col0 = ch * 32 + seg * 8;
-        if (m < M && col0 + 8 <= Kc && sizeof(T) == 2) {
-          *(bf16x8*)(a_lds + (size_t)ch * BM * STR + row * STR + seg * 16) =
-              *(const bf16x8*)(x2 + m * Kc + col0);
-        } else {
-          float vals[8];
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vals[j] = (m < M && col0 + j < Kc)
-                          ? to_f32(x2[m * Kc + col0 + j]) : 0.0f;
-          Mma<T>::store8(a_lds + (size_t)ch * BM * STR, row, seg * 8, vals);
-        }
+    for (int ch = 0; ch < 4; ++ch) {
+      if (ch >= CH) break;
+      if (sizeof(T) == 2) {
+        *(bf16x8*)(a_lds + (size_t)ch * BM * STR + row * STR + seg * 16) =
+            *(bf16x8*)areg[ch];
+      } else {
+        float vals[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vals[j] = to_f32(areg[ch][j]);
+        Mma<T>::store8(a_lds + (size_t)ch * BM * STR, row, seg * 8, vals);
       }
     }
     __syncthreads();
+    if (mt + gridDim.x < mtiles) load_a(mt + gridDim.x);
 
     f32x4 acc[2][3] = {};
     f32x4 sacc[2][3] = {};
@@ -1503,19 +1521,24 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
   if (WANT_Y) load_wrow_regs(breg, wq, g, p, n0, 0, 0);
   if (SIGMA_MODE > 0) load_wrow_regs(creg, wraw, g, p, n0, 0, 0);
 
-  // two M-subtiles share each staged weight tile (halves barrier count)
-  for (int m0 = 0; m0 < p.MI; m0 += 2 * BM) {
-    f32x4 acc[2][2][2] = {};
-    f32x4 sacc[2][2][2] = {};
-    f32x4 tacc[2][2][2] = {};
-    const int nsub = (m0 + BM < p.MI) ? 2 : 1;
+  // M-subtiles share each staged weight tile (fewer barriers). A plain
+  // conv (dgrad-via-patch) carries only the y accumulators, so it can
+  // afford 4 subtiles (196-pixel images finish in ONE weight pass);
+  // sigma variants carry 2-3 accumulator sets and stay at 2.
+  constexpr int MSUB = (SIGMA_MODE == 0) ? 4 : 2;
+  for (int m0 = 0; m0 < p.MI; m0 += MSUB * BM) {
+    f32x4 acc[MSUB][2][2] = {};
+    f32x4 sacc[MSUB][2][2] = {};
+    f32x4 tacc[MSUB][2][2] = {};
+    int rem = (int)((p.MI - m0 + BM - 1) / BM);
+    const int nsub = rem < MSUB ? rem : MSUB;
 
     // hoist the per-fragment pixel decode: ow/oh (integer divisions) and
     // the patch base offset are invariant across the (r, ck) rounds
-    int a_ohs[2][2];      // oh*stride - pad, or INT_MIN when m >= MI
-    int a_base[2][2];     // ((ohs*Wp + ow*stride) * C_pad) element offset
+    int a_ohs[MSUB][2];   // oh*stride - pad, or INT_MIN when m >= MI
+    int a_base[MSUB][2];  // ((ohs*Wp + ow*stride) * C_pad) element offset
 #pragma unroll
-    for (int ms = 0; ms < 2; ++ms)
+    for (int ms = 0; ms < MSUB; ++ms)
 #pragma unroll
       for (int fm = 0; fm < 2; ++fm) {
         int m_local = m0 + ms * BM + wm * 32 + fm * 16 + (lane & 15);
@@ -1547,7 +1570,7 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
         bool more = true;
         if (t + 1 == ROUNDS) {
           rn = 0; ckn = 0;
-          more = (m0 + 2 * BM) < p.MI;  // wraps for the next m-tile pass
+          more = (m0 + MSUB * BM) < p.MI;  // wraps for the next m-tile pass
         }
         if (more) {
           if (WANT_Y) load_wrow_regs(breg, wq, g, p, n0, rn, ckn);
@@ -1563,8 +1586,8 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
             tfrag[fn] = Mma<T>::load(d_lds, brow, lane);
         }
 #pragma unroll
-        for (int ms = 0; ms < 2; ++ms) {
-          if (ms == 1 && nsub == 1) break;
+        for (int ms = 0; ms < MSUB; ++ms) {
+          if (ms >= nsub) break;
 #pragma unroll
           for (int fm = 0; fm < 2; ++fm) {
             // per-lane patch fragment: 8 contiguous (s,c) at filter row r;
@@ -1589,10 +1612,10 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
       if (ck >= p.Kr) { ck = 0; ++r; }
     }
 
-    // epilogue for both m-subtiles
+    // epilogue for the resident m-subtiles
 #pragma unroll
-    for (int ms = 0; ms < 2; ++ms) {
-      if (ms == 1 && nsub == 1) break;
+    for (int ms = 0; ms < MSUB; ++ms) {
+      if (ms >= nsub) break;
 #pragma unroll
       for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
