@@ -29,6 +29,10 @@ torch::Tensor softmax_xent_bwd(torch::Tensor softmax, torch::Tensor target,
 
 // bn_act.hip
 std::vector<torch::Tensor> bn_stats(torch::Tensor x);
+std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
+                                             torch::Tensor running_mean,
+                                             torch::Tensor running_var,
+                                             double momentum, double eps);
 torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
                          torch::Tensor invstd, torch::Tensor gamma,
                          torch::Tensor beta, bool relu, double act_max);
@@ -129,6 +133,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("bn_stats", &bn_stats);
+  m.def("bn_stats_finalize", &bn_stats_finalize);
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("bn_act_bwd", &bn_act_bwd);
   m.def("bn_act_bwd_reduce", &bn_act_bwd_reduce);

@@ -67,6 +67,29 @@ template <typename scalar_t> struct DevT { using type = scalar_t; };
 template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct DevT<at::Half> { using type = _Float16; };
 
+// one tiny kernel replacing the eager mean/var/invstd/running-update chain
+// (~8 launches per BN layer per step)
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var, int C,
+                                   float inv_n, float unbias, float momentum,
+                                   float eps, int has_running) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = sum[c] * inv_n;
+  float v = fmaxf(sumsq[c] * inv_n - m * m, 0.0f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(v + eps);
+  if (has_running) {
+    running_mean[c] = running_mean[c] * (1.0f - momentum) + momentum * m;
+    running_var[c] =
+        running_var[c] * (1.0f - momentum) + momentum * v * unbias;
+  }
+}
+
 }  // namespace
 
 // x: NHWC-contiguous 4-D (memory_format=channels_last, passed as NCHW logical)
@@ -324,4 +347,59 @@ torch::Tensor bn_act_bwd_apply(torch::Tensor g, torch::Tensor x,
   });
   HIP_CHECK_LAST();
   return gx;
+}
+
+
+// batch stats + finalize in one shot: raw sum/sumsq pass, then one tiny
+// kernel computes (mean, invstd) and updates the (f32) running buffers
+// in-place -- replacing the ~8 eager launches of the Python chain.
+// Single-rank path only; SyncBN keeps the raw (mean, E[x^2]) route so the
+// all-reduce can sit between stats and normalization.
+std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
+                                             torch::Tensor running_mean,
+                                             torch::Tensor running_var,
+                                             double momentum, double eps) {
+  int C;
+  int64_t rows;
+  const void* ptr = x.data_ptr();
+  if (x.dim() == 4) {
+    TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "bn_stats_finalize: expected channels_last");
+    C = (int)x.size(1);
+    rows = x.size(0) * x.size(2) * x.size(3);
+  } else {
+    TORCH_CHECK(x.is_contiguous());
+    C = (int)x.size(1);
+    rows = x.size(0);
+  }
+  bool has_running = running_mean.numel() > 0;
+  TORCH_CHECK(!has_running || (running_mean.scalar_type() == torch::kFloat32 &&
+                               running_var.scalar_type() == torch::kFloat32),
+              "bn_stats_finalize: running stats must be f32");
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto sum = torch::zeros({C}, opts);
+  auto sumsq = torch::zeros({C}, opts);
+  int gx = (C + 63) / 64;
+  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64),
+                                  1024);
+  auto stream = c10::hip::getCurrentHIPStream();
+  NN_DISPATCH(x.scalar_type(), "bn_stats_finalize", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
+                       stream, (const T*)ptr, sum.data_ptr<float>(),
+                       sumsq.data_ptr<float>(), rows, C);
+  });
+  float n = (float)rows;
+  float unbias = n / std::max(n - 1.0f, 1.0f);
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     stream, sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     has_running ? running_mean.data_ptr<float>() : nullptr,
+                     has_running ? running_var.data_ptr<float>() : nullptr, C,
+                     1.0f / n, unbias, (float)momentum, (float)eps,
+                     has_running ? 1 : 0);
+  HIP_CHECK_LAST();
+  return {mean, invstd};
 }

@@ -103,18 +103,22 @@ class Net(nn.Module):
         a = self.args
         w_raw = layer.weight
         wq, bias = layer.effective_weight()
+        want_telemetry = i < 20
         with torch.no_grad():
-            input_max = x.detach().max()
+            # input_max is a full-tensor reduce over the activation; for
+            # merged-DAC layers it only feeds the power telemetry
+            # (first-20-batch), so skip it in steady state
             if merged_dac:
                 w_max = w_raw.detach().abs().max()
                 factor = 0.1 * w_max / current
                 sigma_mode = 'abs'
-                power_denom = input_max * w_max
+                power_denom = (x.detach().max() * w_max if want_telemetry
+                               else w_max)
             else:
+                input_max = x.detach().max()
                 factor = 0.1 * input_max / current
                 sigma_mode = 'abs2'
                 power_denom = input_max
-        want_telemetry = i < 20
         telem = ops.NoiseTelemetry() if want_telemetry else None
         if is_conv:
             out = ops.fused_noisy_conv2d(

@@ -559,6 +559,27 @@ class BnAct(torch.autograd.Function):
         import torch.distributed as dist
         ws = dist.get_world_size() if (sync and dist.is_initialized()) else 1
         ctx.sync_ws = ws if training else 1
+        if (training and ws == 1 and use_native(x)
+                and (running_mean is None
+                     or running_mean.dtype == torch.float32)):
+            # fused single-rank path: stats + invstd + running update in
+            # two kernels (the eager chain was ~8 tiny launches per layer)
+            xc = _nhwc(x) if x.dim() == 4 else x.contiguous()
+            empty = torch.empty(0, device=x.device, dtype=torch.float32)
+            mean, invstd = ext().bn_stats_finalize(
+                xc,
+                running_mean if running_mean is not None else empty,
+                running_var if running_var is not None else empty,
+                float(momentum), float(eps))
+            if use_native(x):
+                y = ext().bn_act_fwd(xc, mean, invstd,
+                                     weight.float().contiguous(),
+                                     bias.float().contiguous(),
+                                     bool(relu), float(act_max))
+            ctx.save_for_backward(x, weight, mean, invstd, y)
+            ctx.training = training
+            ctx.relu, ctx.act_max = relu, act_max
+            return y
         if training:
             if use_native(x):
                 mean, var = ext().bn_stats(_nhwc(x) if x.dim() == 4 else x.contiguous())
