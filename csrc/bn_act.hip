@@ -24,20 +24,21 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
   int c = blockIdx.x * 64 + (threadIdx.x & 63);
   int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
   if (c >= C) return;
-  // 4 independent accumulator pairs keep 4 loads in flight per thread
-  // (the single-accumulator loop was latency-bound at ~1/6 of HBM peak)
-  const int64_t st = (int64_t)gridDim.y * (kBlock / 64);
+  // 4 CONSECUTIVE rows per thread per iteration: the 4 in-flight loads
+  // span one contiguous ~4*C*2B region (strided variants fetched 4
+  // scattered 128-B lines each and ran 96% memory-wait)
+  const int64_t st = (int64_t)gridDim.y * (kBlock / 64) * 4;
   float s[4] = {}, sq[4] = {};
-  int64_t r = rstart;
-  for (; r + 3 * st < rows; r += 4 * st) {
+  int64_t r = (int64_t)rstart * 4;
+  for (; r + 3 < rows; r += st) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      float v = to_f32(x[(r + j * st) * C + c]);
+      float v = to_f32(x[(r + j) * C + c]);
       s[j] += v;
       sq[j] += v * v;
     }
   }
-  for (; r < rows; r += st) {
+  for (; r < rows; ++r) {
     float v = to_f32(x[r * C + c]);
     s[0] += v;
     sq[0] += v * v;
@@ -173,8 +174,8 @@ __global__ void bn_act_bwd_reduce_kernel(
   int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
   if (c >= C) return;
   float m = mean[c], is = invstd[c];
-  // unrolled x4: see bn_stats_kernel
-  const int64_t st = (int64_t)gridDim.y * (kBlock / 64);
+  // 4 consecutive rows per iteration: see bn_stats_kernel
+  const int64_t st = (int64_t)gridDim.y * (kBlock / 64) * 4;
   float s_g[4] = {}, s_gx[4] = {};
   auto body = [&](int64_t r, int j) {
     int64_t i = r * C + c;
@@ -186,12 +187,12 @@ __global__ void bn_act_bwd_reduce_kernel(
     s_g[j] += gv;
     s_gx[j] += gv * (to_f32(x[i]) - m) * is;
   };
-  int64_t r = rstart;
-  for (; r + 3 * st < rows; r += 4 * st) {
+  int64_t r = (int64_t)rstart * 4;
+  for (; r + 3 < rows; r += st) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) body(r + j * st, j);
+    for (int j = 0; j < 4; ++j) body(r + j, j);
   }
-  for (; r < rows; r += st) body(r, 0);
+  for (; r < rows; ++r) body(r, 0);
   atomicAdd(&sum_g[c], (s_g[0] + s_g[1]) + (s_g[2] + s_g[3]));
   atomicAdd(&sum_gx[c], (s_gx[0] + s_gx[1]) + (s_gx[2] + s_gx[3]));
 }

@@ -1920,6 +1920,56 @@ __global__ void im2col_kernel(const T* __restrict__ x, T* __restrict__ out,
   }
 }
 
+// Compile-time (C, S) variant: the % / / decode folds to multiply-shift
+// sequences, so there is no LDS table (whose strided reads were 8-way
+// bank-conflicted) and no per-block setup. Instantiated for the common
+// small-C shapes; the generic table kernel below covers the rest.
+template <typename T, int C, int S>
+__global__ void im2col_flat_tmpl_kernel(const T* __restrict__ x,
+                                        T* __restrict__ out, ConvGeom g,
+                                        int cols_p, int64_t nchunks_total) {
+  int nchunks = cols_p >> 3;
+  int rsc = g.R * g.S * C;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < nchunks_total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int seg = (int)(idx % nchunks);
+    int64_t m = idx / nchunks;
+    int64_t t = m;
+    int ow = (int)(t % g.OW);
+    t /= g.OW;
+    int oh = (int)(t % g.OH);
+    t /= g.OH;
+    int n = (int)t;
+    int oh0 = oh * g.stride - g.pad;
+    int ow0 = ow * g.stride - g.pad;
+    const T* xn = x + ((int64_t)n * g.H + oh0) * g.W * g.C + ow0 * g.C;
+    T vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = seg * 8 + j;
+      float v = 0.0f;
+      if (k < rsc) {
+        int c = k % C;       // constexpr divisor -> mul/shift
+        int rs = k / C;
+        int ss = rs % S;
+        int r = rs / S;
+        int ih = oh0 + r;
+        int iw = ow0 + ss;
+        if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
+          v = to_f32(xn[(r * g.W + ss) * g.C + c]);
+      }
+      vals[j] = from_f32<T>(v);
+    }
+    T* dst = out + m * cols_p + seg * 8;
+    if (sizeof(T) == 2) {
+      *(bf16x8*)dst = *(bf16x8*)vals;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[j] = vals[j];
+    }
+  }
+}
+
 // C % 8 != 0 (small-C convs): flat row of R*S*C columns padded to a
 // multiple of 8; each thread fills one aligned 8-column chunk. The
 // (r, s, c) decode of each column (two integer divisions) is computed
@@ -2060,10 +2110,20 @@ torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
     } else {
       int64_t nchunks_total = g.M * (cols_p >> 3);
       int blocks = (int)std::min<int64_t>((nchunks_total + 255) / 256, 8192);
-      hipLaunchKernelGGL((im2col_flat_kernel<T>), dim3(blocks), dim3(256),
-                         2 * (size_t)cols_p * sizeof(int), stream,
-                         (const T*)x.data_ptr(), (T*)col.data_ptr(), g,
-                         cols_p, nchunks_total);
+      if (g.C == 3 && g.S == 5) {
+        hipLaunchKernelGGL((im2col_flat_tmpl_kernel<T, 3, 5>), dim3(blocks),
+                           dim3(256), 0, stream, (const T*)x.data_ptr(),
+                           (T*)col.data_ptr(), g, cols_p, nchunks_total);
+      } else if (g.C == 3 && g.S == 3) {
+        hipLaunchKernelGGL((im2col_flat_tmpl_kernel<T, 3, 3>), dim3(blocks),
+                           dim3(256), 0, stream, (const T*)x.data_ptr(),
+                           (T*)col.data_ptr(), g, cols_p, nchunks_total);
+      } else {
+        hipLaunchKernelGGL((im2col_flat_kernel<T>), dim3(blocks), dim3(256),
+                           2 * (size_t)cols_p * sizeof(int), stream,
+                           (const T*)x.data_ptr(), (T*)col.data_ptr(), g,
+                           cols_p, nchunks_total);
+      }
     }
   });
   HIP_CHECK_LAST();
