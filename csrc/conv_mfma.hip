@@ -872,6 +872,221 @@ void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
       }
 }
 
+// --------------------------------------------------------------------------
+// Fast GEMM wgrad for the flat/linear case: dw[K, C] += gy[M, K]^T @ x[M, C]
+// with K % 8 == 0, C % 8 == 0, 16-bit dtype. 128-deep contraction rounds
+// (4x fewer barriers than the generic kernel), 16-B vector global loads,
+// and a register pipeline that loads round i+1 while round i's MFMAs run.
+// LDS images are [64 rows][128 m] at a 272-B row stride (16-B skew).
+// --------------------------------------------------------------------------
+
+constexpr int WG_BK = 128;          // contraction (m) depth per round
+constexpr int WG_LSTR = WG_BK * 2 + 16;
+
+template <typename T>
+DEV_INLINE void wgrad_mk_load(T dst[4][8], const T* __restrict__ src,
+                              int64_t m0, int64_t M, int cols, int c0) {
+  // thread t covers 4 (m_i, seg) slots: idx = t + i*256, m_i = idx>>3,
+  // seg = idx&7 (64 cols = 8 segs of 8)
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int idx = threadIdx.x + i * kBlock;
+    int mi = idx >> 3;
+    int seg = idx & 7;
+    int64_t m = m0 + mi;
+    int c = c0 + seg * 8;
+    if (m < M && c + 8 <= cols) {
+      *(bf16x8*)dst[i] = *(const bf16x8*)(src + m * cols + c);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[i][j] = from_f32<T>(0.0f);
+    }
+  }
+}
+
+// The transposed scatter writes 16 rows at stride 8 per instruction; at a
+// 272-B row stride those land in only 2 LDS banks (8-way conflict). A
+// row-dependent 16-B chunk rotation spreads them; fragment reads apply
+// the same rotation (still one aligned b128 per fragment).
+DEV_INLINE int wg_swz(int row, int chunk) { return ((chunk + (row >> 3)) & 15); }
+
+template <typename T>
+DEV_INLINE void wgrad_mk_store(char* lds, const T src[4][8]) {
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int idx = threadIdx.x + i * kBlock;
+    int mi = idx >> 3;
+    int seg = idx & 7;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = seg * 8 + j;
+      *(T*)(lds + row * WG_LSTR + wg_swz(row, mi >> 3) * 16 + (mi & 7) * 2) =
+          src[i][j];
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock)
+void wgrad_mk_kernel(const T* __restrict__ gy, const T* __restrict__ x,
+                     float* __restrict__ dw /* [K, C] f32 */, int64_t M,
+                     int K, int C, int mchunks_per_block) {
+  int c0 = blockIdx.x * BN;
+  int k0 = blockIdx.y * BM;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* g_lds = smem;                      // [64 k rows][128 m]
+  char* x_lds = smem + BM * WG_LSTR;       // [64 c rows][128 m]
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+  f32x4 acc[2][2] = {};
+
+  int64_t m_start = (int64_t)blockIdx.z * mchunks_per_block * WG_BK;
+  int64_t m_end = m_start + (int64_t)mchunks_per_block * WG_BK;
+  if (m_end > M) m_end = M;
+
+  T greg[4][8], xreg[4][8];
+  if (m_start < m_end) {
+    wgrad_mk_load(greg, gy, m_start, M, K, k0);
+    wgrad_mk_load(xreg, x, m_start, M, C, c0);
+  }
+
+  for (int64_t m0 = m_start; m0 < m_end; m0 += WG_BK) {
+    wgrad_mk_store(g_lds, greg);
+    wgrad_mk_store(x_lds, xreg);
+    __syncthreads();
+    if (m0 + WG_BK < m_end) {
+      wgrad_mk_load(greg, gy, m0 + WG_BK, M, K, k0);
+      wgrad_mk_load(xreg, x, m0 + WG_BK, M, C, c0);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        int arow = wm * 32 + fm * 16 + (lane & 15);
+        auto a = *(typename Mma<T>::frag*)(
+            g_lds + arow * WG_LSTR +
+            wg_swz(arow, ks * 4 + (lane >> 4)) * 16);
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          int brow = wn * 32 + fn * 16 + (lane & 15);
+          auto b = *(typename Mma<T>::frag*)(
+              x_lds + brow * WG_LSTR +
+              wg_swz(brow, ks * 4 + (lane >> 4)) * 16);
+          Mma<T>::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // each (ctile, ktile, z) block owns a disjoint region of its z-slice:
+  // plain stores into per-slice partials (summed host-side). The atomic
+  // version serialized ~90 contenders per output word on wide shapes.
+  float* slab = dw + (int64_t)blockIdx.z * K * C;
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int k = k0 + wm * 32 + fm * 16 + 4 * (lane >> 4) + reg;
+        int c = c0 + wn * 32 + fn * 16 + (lane & 15);
+        if (k < K && c < C) slab[(int64_t)k * C + c] = acc[fm][fn][reg];
+      }
+}
+
+// 128x128-tile variant for wide outputs (e.g. conv2 wgrad: K=120,
+// C=3000): the 64-wide c-tiles of wgrad_mk_kernel re-read the small gy
+// operand once per tile (47x = 2.3 GB at batch 2048); doubling both
+// tile sides quarters the cross-reads. No register prefetch (the tile
+// needs 8 staging slots per thread per tensor; the extra 64 VGPRs would
+// cost a wave); 4 waves as 2x2 quadrants of 64x64, acc[4][4].
+template <typename T>
+DEV_INLINE void wgrad_mk4_stage(char* lds, const T* __restrict__ src,
+                                int64_t m0, int64_t M, int cols, int c0) {
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int idx = threadIdx.x + i * kBlock;  // 0..2047
+    int mi = idx >> 4;                   // 0..127 (m)
+    int seg = idx & 15;                  // 0..15  (8-col group)
+    int64_t m = m0 + mi;
+    int c = c0 + seg * 8;
+    T vals[8];
+    if (m < M && c + 8 <= cols) {
+      *(bf16x8*)vals = *(const bf16x8*)(src + m * cols + c);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = from_f32<T>(0.0f);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = seg * 8 + j;
+      *(T*)(lds + row * WG_LSTR + wg_swz(row, mi >> 3) * 16 + (mi & 7) * 2) =
+          vals[j];
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock)
+void wgrad_mk4_kernel(const T* __restrict__ gy, const T* __restrict__ x,
+                      float* __restrict__ dw /* [K, C] f32 */, int64_t M,
+                      int K, int C, int mchunks_per_block) {
+  int c0 = blockIdx.x * 128;
+  int k0 = blockIdx.y * 128;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* g_lds = smem;                        // [128 k rows][128 m]
+  char* x_lds = smem + 128 * WG_LSTR;        // [128 c rows][128 m]
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+  f32x4 acc[4][4] = {};
+
+  int64_t m_start = (int64_t)blockIdx.z * mchunks_per_block * WG_BK;
+  int64_t m_end = m_start + (int64_t)mchunks_per_block * WG_BK;
+  if (m_end > M) m_end = M;
+
+  for (int64_t m0 = m_start; m0 < m_end; m0 += WG_BK) {
+    wgrad_mk4_stage(g_lds, gy, m0, M, K, k0);
+    wgrad_mk4_stage(x_lds, x, m0, M, C, c0);
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int arow = wm * 64 + fm * 16 + (lane & 15);
+        auto a = *(typename Mma<T>::frag*)(
+            g_lds + arow * WG_LSTR +
+            wg_swz(arow, ks * 4 + (lane >> 4)) * 16);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int brow = wn * 64 + fn * 16 + (lane & 15);
+          auto b = *(typename Mma<T>::frag*)(
+              x_lds + brow * WG_LSTR +
+              wg_swz(brow, ks * 4 + (lane >> 4)) * 16);
+          Mma<T>::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  float* slab = dw + (int64_t)blockIdx.z * K * C;  // see wgrad_mk_kernel
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int k = k0 + wm * 64 + fm * 16 + 4 * (lane >> 4) + reg;
+        int c = c0 + wn * 64 + fn * 16 + (lane & 15);
+        if (k < K && c < C) slab[(int64_t)k * C + c] = acc[fm][fn][reg];
+      }
+}
+
 template <typename scalar_t> struct DevT { using type = scalar_t; };
 template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct DevT<at::Half> { using type = _Float16; };
@@ -1110,6 +1325,43 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
   TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
   auto g = linear_geom(x, gy.size(1));
   auto dw_f = torch::zeros({g.K, g.C}, x.options().dtype(torch::kFloat32));
+  if ((g.K & 7) == 0 && (g.C & 7) == 0 && gy.element_size() == 2) {
+    // aligned 16-bit: 128-deep rounds. Wide outputs use the 128x128-tile
+    // variant (fewer cross-tile operand re-reads); narrow ones the
+    // 64x64 with a register pipeline.
+    bool wide = (int64_t)g.C * g.K >= 128 * 1024;
+    int tn = wide ? 128 : BN;
+    int tm = wide ? 128 : BM;
+    int ctiles = (g.C + tn - 1) / tn;
+    int ktiles = (g.K + tm - 1) / tm;
+    int64_t mtotal = (g.M + WG_BK - 1) / WG_BK;
+    int64_t target_z =
+        std::max<int64_t>(1, 2048 / std::max(1, ctiles * ktiles));
+    int chunks = (int)((mtotal + target_z - 1) / target_z);
+    int mslices = (int)((mtotal + chunks - 1) / chunks);  // no empty slices
+    dim3 grid(ctiles, ktiles, mslices);
+    size_t lds = (size_t)(tm + tn) * WG_LSTR;
+    // per-slice partials, plain stores, deterministic sum
+    auto parts = torch::empty({(int64_t)mslices, (int64_t)g.K * g.C},
+                              x.options().dtype(torch::kFloat32));
+    NN_DISPATCH(gy.scalar_type(), "linear_wgrad_mk", [&] {
+      using T = typename DevT<scalar_t>::type;
+      if (wide) {
+        hipLaunchKernelGGL((wgrad_mk4_kernel<T>), grid, dim3(kBlock), lds,
+                           c10::hip::getCurrentHIPStream(),
+                           (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                           parts.data_ptr<float>(), g.M, g.K, g.C, chunks);
+      } else {
+        hipLaunchKernelGGL((wgrad_mk_kernel<T>), grid, dim3(kBlock), lds,
+                           c10::hip::getCurrentHIPStream(),
+                           (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                           parts.data_ptr<float>(), g.M, g.K, g.C, chunks);
+      }
+    });
+    HIP_CHECK_LAST();
+    return parts.sum(0).view({(int64_t)g.K, (int64_t)g.C})
+        .to(x.scalar_type());
+  }
   int64_t mtotal = (g.M + BK - 1) / BK;
   int ctiles = (g.C + BN - 1) / BN;
   int ktiles = (g.K + BM - 1) / BM;
@@ -2157,13 +2409,12 @@ torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
   int64_t rsc = R * S * C;
   TORCH_CHECK((int64_t)gy.size(0) * gy.size(2) * gy.size(3) == M);
 
-  // dw[k, cols_p] = gy^T @ col. Tall-skinny contractions into a tiny
-  // output (conv1-type: M ~1e6, K*cols ~1e4) favor the split-M atomic
-  // wgrad kernel (2.5x over hipBLASLt there); wide outputs favor the
-  // library GEMM (measured: conv2-type 0.87 ms blas vs 1.19 ms ours).
+  // dw[k, cols_p] = gy^T @ col. Long contractions (M >= 64k) favor the
+  // custom split-M kernel (swizzled transposed staging: conv1 0.29 ms vs
+  // 1.98 blas; conv2 0.77 vs 0.86); short ones (fc layers) hipBLASLt.
   auto gy2 = gy.permute({0, 2, 3, 1}).reshape({M, K});  // raw view, free
   torch::Tensor dw_flat;
-  if (K * cols_p <= 65536 && M >= 262144) {
+  if (M >= 65536) {
     int64_t K8 = (K + 7) & ~7;
     if (K8 != K && gy.scalar_type() != torch::kFloat32) {
       // odd K (e.g. 65 output channels) leaves gy rows 2-byte-misaligned,
