@@ -21,13 +21,26 @@ constexpr int kBlock = 256;
 template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
                                 float* __restrict__ sumsq, int64_t rows, int C) {
-  int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
+  // channel/row thread split: 64 channel lanes when C allows; for C < 64
+  // pack ceil-free row groups of C lanes so no half-wave idles (MNv2's
+  // C=16..32 stem layers ran at 25-50% lane utilization otherwise)
+  int c, rgrp, ngrp;
+  if (C >= 64) {
+    c = blockIdx.x * 64 + (threadIdx.x & 63);
+    rgrp = threadIdx.x >> 6;
+    ngrp = kBlock / 64;
+  } else {
+    c = threadIdx.x % C;
+    rgrp = threadIdx.x / C;
+    ngrp = kBlock / C;
+    if (rgrp >= ngrp) return;
+  }
   if (c >= C) return;
+  int rstart = blockIdx.y * ngrp + rgrp;
   // 4 CONSECUTIVE rows per thread per iteration: the 4 in-flight loads
   // span one contiguous ~4*C*2B region (strided variants fetched 4
   // scattered 128-B lines each and ran 96% memory-wait)
-  const int64_t st = (int64_t)gridDim.y * (kBlock / 64) * 4;
+  const int64_t st = (int64_t)gridDim.y * ngrp * 4;
   float s[4] = {}, sq[4] = {};
   int64_t r = (int64_t)rstart * 4;
   for (; r + 3 < rows; r += st) {
@@ -112,8 +125,11 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum = torch::zeros({C}, opts);
   auto sumsq = torch::zeros({C}, opts);
-  int gx = (C + 63) / 64;
-  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64), 1024);
+  int gx = C >= 64 ? (C + 63) / 64 : 1;
+  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  // cap so atomic depth per channel (gy * ngrp) stays ~4k
+  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+                                  std::max(1, 4096 / ngrp));
   NN_DISPATCH(x.scalar_type(),
                                   "bn_stats", [&] {
     using T = typename DevT<scalar_t>::type;
@@ -170,12 +186,22 @@ __global__ void bn_act_bwd_reduce_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     float* __restrict__ sum_g, float* __restrict__ sum_gx, int64_t rows, int C,
     int do_relu, float act_max) {
-  int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  int rstart = blockIdx.y * (kBlock / 64) + (threadIdx.x >> 6);
+  int c, rgrp, ngrp;  // see bn_stats_kernel
+  if (C >= 64) {
+    c = blockIdx.x * 64 + (threadIdx.x & 63);
+    rgrp = threadIdx.x >> 6;
+    ngrp = kBlock / 64;
+  } else {
+    c = threadIdx.x % C;
+    rgrp = threadIdx.x / C;
+    ngrp = kBlock / C;
+    if (rgrp >= ngrp) return;
+  }
   if (c >= C) return;
+  int rstart = blockIdx.y * ngrp + rgrp;
   float m = mean[c], is = invstd[c];
   // 4 consecutive rows per iteration: see bn_stats_kernel
-  const int64_t st = (int64_t)gridDim.y * (kBlock / 64) * 4;
+  const int64_t st = (int64_t)gridDim.y * ngrp * 4;
   float s_g[4] = {}, s_gx[4] = {};
   auto body = [&](int64_t r, int j) {
     int64_t i = r * C + c;
@@ -247,8 +273,10 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
   auto sum_gx = torch::zeros({C}, opts);
   auto gx = torch::empty_like(g);
   auto stream = c10::hip::getCurrentHIPStream();
-  int gx_blocks = (C + 63) / 64;
-  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64), 1024);
+  int gx_blocks = C >= 64 ? (C + 63) / 64 : 1;
+  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+                                  std::max(1, 4096 / ngrp));
   int64_t n = x.numel();
   int eblocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd", [&] {
@@ -300,8 +328,10 @@ std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum_g = torch::zeros({C}, opts);
   auto sum_gx = torch::zeros({C}, opts);
-  int gx_blocks = (C + 63) / 64;
-  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64), 1024);
+  int gx_blocks = C >= 64 ? (C + 63) / 64 : 1;
+  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+                                  std::max(1, 4096 / ngrp));
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd_reduce", [&] {
     using T = typename DevT<scalar_t>::type;
     hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
@@ -380,9 +410,10 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum = torch::zeros({C}, opts);
   auto sumsq = torch::zeros({C}, opts);
-  int gx = (C + 63) / 64;
-  int gy = (int)std::min<int64_t>((rows + (kBlock / 64) - 1) / (kBlock / 64),
-                                  1024);
+  int gx = C >= 64 ? (C + 63) / 64 : 1;
+  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+                                  std::max(1, 4096 / ngrp));
   auto stream = c10::hip::getCurrentHIPStream();
   NN_DISPATCH(x.scalar_type(), "bn_stats_finalize", [&] {
     using T = typename DevT<scalar_t>::type;

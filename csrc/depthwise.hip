@@ -81,38 +81,55 @@ __global__ void dwconv_dgrad_kernel(const T* __restrict__ gy,
   }
 }
 
-// Each thread owns ONE channel and a slice of the output pixels,
-// accumulating all R*S taps in registers; one atomicAdd per tap per
-// thread-slice (instead of one per OUTPUT ELEMENT, which serialized on
-// the same dw[c][r][s] word ~N*OH*OW deep).
+// Each thread owns ONE (channel, pixel-stream) pair, accumulating all
+// R*S taps in registers; one atomicAdd per tap per thread (instead of
+// one per OUTPUT ELEMENT, which serialized on the same dw[c][r][s] word
+// ~N*OH*OW deep). Channels ride the FAST lane index so gy/x reads stay
+// coalesced; for C < kBlock several pixel streams share a block so no
+// lanes idle (the one-channel-per-thread layout left 224 of 256 lanes
+// dead on MobileNet's 32-channel stem).
 template <typename T, int MAXTAPS>
 __global__ void dwconv_wgrad_kernel(const T* __restrict__ gy,
                                     const T* __restrict__ x,
                                     float* __restrict__ dw, int C,
                                     int H, int W, int OH, int OW, int R,
                                     int S, int stride, int pad,
-                                    int64_t npix) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+                                    int64_t npix, int cspan, int pgrp) {
+  // cspan = min(C, kBlock) rounded context: threads [0, cspan*pgrp)
+  int c_in_span = threadIdx.x % cspan;
+  int grp = threadIdx.x / cspan;
+  if (grp >= pgrp) return;
+  int c = blockIdx.x * cspan + c_in_span;
   if (c >= C) return;
   float acc[MAXTAPS];
   int taps = R * S;
 #pragma unroll
   for (int t = 0; t < MAXTAPS; ++t) acc[t] = 0.0f;
-  // pixel slice: grid.y slices over N*OH*OW
-  for (int64_t pix = blockIdx.y; pix < npix; pix += gridDim.y) {
-    int64_t t = pix;
-    int ow = (int)(t % OW);
-    t /= OW;
-    int oh = (int)(t % OH);
-    int64_t nb = t / OH;
-    float g = to_f32(gy[pix * C + c]);
-    for (int r = 0; r < R; ++r) {
-      int ih = oh * stride - pad + r;
-      if (ih < 0 || ih >= H) continue;
-      for (int s = 0; s < S; ++s) {
-        int iw = ow * stride - pad + s;
-        if (iw < 0 || iw >= W) continue;
-        acc[r * S + s] += g * to_f32(x[((nb * H + ih) * W + iw) * C + c]);
+  // pixel streams walk CONSECUTIVE pixels in chunks of 8 so the tap
+  // windows of successive outputs overlap in cache (grid-strided single
+  // pixels gave zero x reuse across iterations)
+  constexpr int CHUNK = 8;
+  int64_t nchunks = (npix + CHUNK - 1) / CHUNK;
+  for (int64_t chunk = (int64_t)blockIdx.y * pgrp + grp; chunk < nchunks;
+       chunk += (int64_t)gridDim.y * pgrp) {
+#pragma unroll
+    for (int u = 0; u < CHUNK; ++u) {
+      int64_t pix = chunk * CHUNK + u;
+      if (pix >= npix) break;
+      int64_t t = pix;
+      int ow = (int)(t % OW);
+      t /= OW;
+      int oh = (int)(t % OH);
+      int64_t nb = t / OH;
+      float g = to_f32(gy[pix * C + c]);
+      for (int r = 0; r < R; ++r) {
+        int ih = oh * stride - pad + r;
+        if (ih < 0 || ih >= H) continue;
+        for (int s = 0; s < S; ++s) {
+          int iw = ow * stride - pad + s;
+          if (iw < 0 || iw >= W) continue;
+          acc[r * S + s] += g * to_f32(x[((nb * H + ih) * W + iw) * C + c]);
+        }
       }
     }
   }
@@ -184,8 +201,12 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   int OH = (int)gy.size(2), OW = (int)gy.size(3);
   auto dw_f = torch::zeros({C, (int)R, (int)S}, x.options().dtype(torch::kFloat32));
   int64_t npix = (int64_t)N * OH * OW;
-  int cblocks = (C + kBlock - 1) / kBlock;
-  int mslices = (int)std::min<int64_t>(npix,
+  int cspan = std::min(C, kBlock);
+  int pgrp = kBlock / cspan;  // pixel streams sharing one block
+  int cblocks = (C + cspan - 1) / cspan;
+  int64_t nchunks = (npix + 7) / 8;
+  int mslices = (int)std::min<int64_t>(
+      (nchunks + pgrp - 1) / pgrp,
       std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
   TORCH_CHECK(R * S <= 64, "dwconv_wgrad: filter too large");
   NN_DISPATCH(gy.scalar_type(), "dwconv_wgrad", [&] {
@@ -196,19 +217,19 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
                          dim3(kBlock), 0, stream, (const T*)gy.data_ptr(),
                          (const T*)x.data_ptr(), dw_f.data_ptr<float>(), C, H,
                          W, OH, OW, (int)R, (int)S, (int)stride, (int)pad,
-                         npix);
+                         npix, cspan, pgrp);
     else if (R * S <= 25)
       hipLaunchKernelGGL((dwconv_wgrad_kernel<T, 25>), dim3(cblocks, mslices),
                          dim3(kBlock), 0, stream, (const T*)gy.data_ptr(),
                          (const T*)x.data_ptr(), dw_f.data_ptr<float>(), C, H,
                          W, OH, OW, (int)R, (int)S, (int)stride, (int)pad,
-                         npix);
+                         npix, cspan, pgrp);
     else
       hipLaunchKernelGGL((dwconv_wgrad_kernel<T, 64>), dim3(cblocks, mslices),
                          dim3(kBlock), 0, stream, (const T*)gy.data_ptr(),
                          (const T*)x.data_ptr(), dw_f.data_ptr<float>(), C, H,
                          W, OH, OW, (int)R, (int)S, (int)stride, (int)pad,
-                         npix);
+                         npix, cspan, pgrp);
   });
   HIP_CHECK_LAST();
   return dw_f.view({C, 1, (int)R, (int)S}).to(x.scalar_type());

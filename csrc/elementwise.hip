@@ -23,18 +23,34 @@ inline int grid_1d(int64_t n, int per_thread = 1) {
 // --------------------------------------------------------------------------
 // fake_quant: out = round(clamp((x-min)/scale + U(-s,s), 0, qmax))*scale + min
 // --------------------------------------------------------------------------
+// 4 elements per thread: ONE Philox draw covers all four stochastic-
+// rounding offsets (the per-element draw made quantization RNG-bound).
 template <typename T, bool STOCH>
 __global__ void fake_quant_kernel(const T* __restrict__ x, T* __restrict__ out,
                                   int64_t n, float min_value, float inv_scale,
                                   float scale, float qmax, float stoch,
                                   uint64_t seed) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    float q = (to_f32(x[i]) - min_value) * inv_scale;
-    if (STOCH) q += uniform_pm(seed, (uint64_t)i, stoch);
-    q = fminf(fmaxf(q, 0.0f), qmax);
-    q = nearbyintf(q);
-    out[i] = from_f32<T>(q * scale + min_value);
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       i0 < n; i0 += stride) {
+    float u[4] = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (STOCH) {
+      Philox4 ph = philox4x32(seed, (uint64_t)(i0 >> 2));
+      u[0] = (2.0f * u01(ph.x) - 1.0f) * stoch;
+      u[1] = (2.0f * u01(ph.y) - 1.0f) * stoch;
+      u[2] = (2.0f * u01(ph.z) - 1.0f) * stoch;
+      u[3] = (2.0f * u01(ph.w) - 1.0f) * stoch;
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int64_t i = i0 + j;
+      if (i >= n) continue;
+      float q = (to_f32(x[i]) - min_value) * inv_scale;
+      if (STOCH) q += u[j];
+      q = fminf(fmaxf(q, 0.0f), qmax);
+      q = nearbyintf(q);
+      out[i] = from_f32<T>(q * scale + min_value);
+    }
   }
 }
 
@@ -134,13 +150,13 @@ torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value
                                   "fake_quant_fwd", [&] {
     using T = typename DevT<scalar_t>::type;
     if (stochastic > 0) {
-      hipLaunchKernelGGL((fake_quant_kernel<T, true>), dim3(grid_1d(n)),
+      hipLaunchKernelGGL((fake_quant_kernel<T, true>), dim3(grid_1d((n + 3) / 4)),
                          dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
                          (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
                          (float)min_value, 1.0f / scale, scale, qmax,
                          (float)stochastic, (uint64_t)seed);
     } else {
-      hipLaunchKernelGGL((fake_quant_kernel<T, false>), dim3(grid_1d(n)),
+      hipLaunchKernelGGL((fake_quant_kernel<T, false>), dim3(grid_1d((n + 3) / 4)),
                          dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
                          (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
                          (float)min_value, 1.0f / scale, scale, qmax, 0.0f,
