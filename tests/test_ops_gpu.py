@@ -741,3 +741,48 @@ def test_im2col_materialize_flat_layout():
         assert torch.allclose(col[:, :rsc].float().cpu(), ref_u.cpu(),
                               atol=1e-1, rtol=1e-2), C
         assert (col[:, rsc:] == 0).all()
+
+
+@pytest.mark.gpu
+def test_linear_wgrad_wide_tile_variant():
+    """K*C >= 128k routes to the 128x128-tile wgrad kernel; verify vs
+    fp32 matmul, plus the narrow-tile (64x64) path on an aligned shape."""
+    torch.manual_seed(3)
+    for M, K, C in ((256, 128, 1024), (300, 16, 8200), (4096, 72, 80)):
+        g = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(M, C, device="cuda", dtype=torch.bfloat16)
+        dw = ops.ext().linear_wgrad(g, x).float()
+        ref_dw = g.float().t() @ x.float()
+        err = (dw - ref_dw).abs().max() / (ref_dw.abs().max() + 1e-6)
+        assert err < 5e-2, (M, K, C, err)
+
+
+@pytest.mark.gpu
+def test_wgrad_deterministic():
+    """Per-slice partial buffers give bit-identical wgrad across runs."""
+    torch.manual_seed(4)
+    g = torch.randn(200000, 72, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(200000, 80, device="cuda", dtype=torch.bfloat16)
+    a = ops.ext().linear_wgrad(g, x)
+    b = ops.ext().linear_wgrad(g, x)
+    assert torch.equal(a, b)
+
+
+@pytest.mark.gpu
+def test_bn_finalize_running_stats_match_torch():
+    """bn_stats_finalize updates running mean/var exactly like
+    torch.nn.BatchNorm2d (momentum semantics + unbiased var)."""
+    torch.manual_seed(5)
+    x = torch.randn(8, 24, 10, 10, device="cuda", dtype=torch.bfloat16)
+    xc = cl(x)
+    bn = torch.nn.BatchNorm2d(24, momentum=0.1, eps=1e-5).cuda().float()
+    bn.train()
+    bn(x.float())
+    rm = torch.zeros(24, device="cuda")
+    rv = torch.ones(24, device="cuda")
+    mean, invstd = ops.ext().bn_stats_finalize(xc, rm, rv, 0.1, 1e-5)
+    assert torch.allclose(rm, bn.running_mean, atol=5e-3), \
+        (rm - bn.running_mean).abs().max()
+    assert torch.allclose(rv, bn.running_var, atol=5e-3, rtol=1e-2)
+    ref_mean = x.float().mean(dim=(0, 2, 3))
+    assert torch.allclose(mean, ref_mean, atol=5e-3)
