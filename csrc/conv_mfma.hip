@@ -1608,6 +1608,7 @@ struct PatchGeom {
   int C_pad;   // channels padded to a multiple of 8
   int Wp;      // W + 2*pad (patch holds the horizontal borders)
   int Kr;      // per-row contraction length = S * C_pad
+  int Krs;     // padded weight row stride = round32(Kr) (zero tail)
   int MI;      // outputs per image = OH * OW
 };
 
@@ -1698,6 +1699,13 @@ DEV_INLINE void stage_wpad_row(char* lds, const T* __restrict__ w,
 // structure exposed the full global-load latency every round; PMC showed
 // 65% of wave-cycles parked). One wraw load feeds both the sigma and the
 // telemetry |w| stores.
+//
+// The weight image is FULLY padded host-side (rows to the k-tile
+// multiple, columns to round32(Kr), all zeros) so this load is one
+// UNCONDITIONAL 16-B vector: the earlier bounds-checked fallback
+// compiled to 16 exec-masked global_load_ushort each trailed by
+// s_waitcnt vmcnt(0) -- a serial latency storm on every round for any
+// block touching a tile edge.
 template <typename T>
 DEV_INLINE void load_wrow_regs(T dst[8], const T* __restrict__ w,
                                const ConvGeom g, const PatchGeom p, int n0,
@@ -1705,15 +1713,13 @@ DEV_INLINE void load_wrow_regs(T dst[8], const T* __restrict__ w,
   int row = threadIdx.x >> 2;
   int seg = threadIdx.x & 3;
   int k = n0 + row;
-  int c0 = ck + seg * 8;
-  if (k < g.K && sizeof(T) == 2 && c0 + 8 <= p.Kr) {
-    *(bf16x8*)dst = *(const bf16x8*)(w + ((int64_t)k * g.R + r) * p.Kr + c0);
-    return;
-  }
-  const T* pw = w + ((int64_t)k * g.R + r) * p.Kr;
+  const T* src = w + ((int64_t)k * g.R + r) * p.Krs + ck + seg * 8;
+  if (sizeof(T) == 2) {
+    *(bf16x8*)dst = *(const bf16x8*)src;
+  } else {
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    dst[j] = (k < g.K && c0 + j < p.Kr) ? pw[c0 + j] : from_f32<T>(0.0f);
+    for (int j = 0; j < 8; ++j) dst[j] = src[j];
+  }
 }
 
 template <typename T, bool ABS_TRANSFORM, int SIGMA_MODE>
@@ -1938,6 +1944,20 @@ inline torch::Tensor pad_weight_raw(const torch::Tensor& w, int c_pad) {
   return padded.contiguous();  // [K, R, S, C_pad]
 }
 
+// fully padded weight image for the patch kernel: [k_pad, R, krs] with
+// zero rows beyond K and zero column tails beyond S*C_pad, so the
+// staging loads need no bounds checks (see load_wrow_regs)
+inline torch::Tensor pad_weight_full(const torch::Tensor& w, int c_pad,
+                                     int k_pad, int krs) {
+  int64_t K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  auto raw = w.permute({0, 2, 3, 1});  // [K, R, S, C] view
+  auto out = torch::zeros({(int64_t)k_pad, R, (int64_t)krs}, w.options());
+  auto padded = at::constant_pad_nd(raw, {0, c_pad - C}, 0)
+                    .reshape({K, R, S * (int64_t)c_pad});
+  out.narrow(0, 0, K).narrow(2, 0, S * (int64_t)c_pad).copy_(padded);
+  return out;
+}
+
 }  // namespace
 
 
@@ -1955,6 +1975,7 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
   p.C_pad = (g.C + 7) & ~7;
   p.Wp = g.W + 2 * g.pad;
   p.Kr = g.S * p.C_pad;
+  p.Krs = (p.Kr + 31) & ~31;
   p.MI = g.OH * g.OW;
   auto out = torch::empty({g.N, g.K, g.OH, g.OW},
                           x.options().memory_format(at::MemoryFormat::ChannelsLast));
@@ -1971,9 +1992,14 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
         }()
       : torch::empty({0}, x.options().dtype(torch::kFloat32));
   auto factor_f = factor.to(torch::kFloat32).reshape({1}).contiguous();
-  auto wq_pad = want_y ? pad_weight_raw(wq, p.C_pad) : torch::Tensor();
-  auto wraw_pad = (sigma_mode > 0) ? pad_weight_raw(wraw, p.C_pad)
-                                   : torch::Tensor();
+  // fully padded weight image (rows to the k-tile multiple, columns to
+  // round32): staging loads become unconditional 16-B vectors
+  int k_pad = (g.K + BN - 1) / BN * BN;
+  auto wq_pad = want_y ? pad_weight_full(wq, p.C_pad, k_pad, p.Krs)
+                       : torch::Tensor();
+  auto wraw_pad = (sigma_mode > 0)
+                      ? pad_weight_full(wraw, p.C_pad, k_pad, p.Krs)
+                      : torch::Tensor();
   size_t lds = 0;  // per-dtype, set inside the dispatch
   dim3 grid((g.K + BN - 1) / BN, g.N);
   NN_DISPATCH(x.scalar_type(), "conv_fwd_patch", [&] {
