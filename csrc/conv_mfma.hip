@@ -436,7 +436,8 @@ void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
   const int nfrag = (K + 15) >> 4;                 // <= 6
   const int nf_w = (nfrag - wn + 1) >> 1;          // fragments of this wave
 
-  // stage the full weight set once (rows k < K, zero-padded to 96)
+  // stage the full weight set once; the host pads rows to 96 and columns
+  // to round32(Kc) with zeros, so the loads are unconditional
   for (int rb = 0; rb < KROWS; rb += BM) {
     int k = rb + row;
     if (k >= KROWS) continue;  // rows 96..127 of the second pass
@@ -446,17 +447,15 @@ void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
       const T* pw = wq + (int64_t)k * Kc + col0;
       const T* pr = wraw + (int64_t)k * Kc + col0;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float v = (WANT_Y && k < K && col0 + j < Kc) ? to_f32(pw[j]) : 0.0f;
-        vals[j] = v;
-      }
+      for (int j = 0; j < 8; ++j)
+        vals[j] = WANT_Y ? to_f32(pw[j]) : 0.0f;
       if (WANT_Y)
         Mma<T>::store8(b_lds + (size_t)ch * KROWS * STR, k, seg * 8, vals);
       if (SIGMA_MODE > 0) {
         float sv[8], tv[8];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          float v = (k < K && col0 + j < Kc) ? fabsf(to_f32(pr[j])) : 0.0f;
+          float v = fabsf(to_f32(pr[j]));
           tv[j] = v;
           sv[j] = (SIGMA_MODE == 2) ? v * v + v : v;
         }
@@ -481,7 +480,7 @@ void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
     for (int ch = 0; ch < 4; ++ch) {
       if (ch >= CH) break;
       int col0 = ch * 32 + seg * 8;
-      if (m < M && col0 + 8 <= Kc && sizeof(T) == 2) {
+      if (m < M && sizeof(T) == 2) {  // Kc is round32: no column tail
         *(bf16x8*)areg[ch] = *(const bf16x8*)(x2 + m * Kc + col0);
       } else {
 #pragma unroll
@@ -1388,7 +1387,8 @@ namespace {
 
 // host side of the small-K fused GEMM (see conv_fwd_smallk_kernel)
 bool smallk_eligible(const torch::Tensor& x, int64_t K) {
-  return x.size(1) <= 128 && (x.size(1) & 7) == 0 && K <= 96 &&
+  // Kc % 32 == 0: the A/W staging loads carry no column bounds checks
+  return x.size(1) <= 128 && (x.size(1) & 31) == 0 && K <= 96 &&
          x.element_size() == 2;
 }
 
@@ -1400,6 +1400,13 @@ std::vector<torch::Tensor> linear_fwd_fused_smallk(
   int K = (int)wraw.size(0);
   int Kc = (int)x.size(1);
   int CH = (Kc + 31) >> 5;
+  // the kernel stages a fixed 96-row weight image with no row bounds
+  // checks: pad the (tiny) weight tensors with zero rows
+  if (K < 96) {
+    wraw = at::constant_pad_nd(wraw, {0, 0, 0, 96 - K}, 0);
+    if (want_y) wq = at::constant_pad_nd(wq, {0, 0, 0, 96 - K}, 0);
+    else wq = wraw;
+  }
   auto out = torch::empty({M, (int64_t)K}, x.options());
   bool has_bias = bias.numel() > 0;
   torch::Tensor bias_f;
@@ -2071,9 +2078,11 @@ std::vector<torch::Tensor> conv_fwd_fused_col_impl(
                      (int)x.size(1), (int)wraw.size(0), (int)wraw.size(2),
                      (int)wraw.size(3), (int)stride, (int)pad);
   int rsc = g.R * g.S * g.C;
-  int cols_p = (rsc + 7) & ~7;
+  int cols_p = ((g.C & 7) == 0) ? rsc : ((rsc + 31) & ~31);
   auto col = im2col_materialize(x, g.K, stride, pad, g.R, g.S);  // [M,cols_p]
-  // flat weight rows: raw [K,R,S,C] view reshaped to [K, R*S*C], zero-padded
+  // flat weight rows: raw [K,R,S,C] view reshaped to [K, R*S*C],
+  // column-padded to match col (row padding to the 96-row LDS image
+  // happens inside the small-K host so K stays the logical count)
   auto flat_w = [&](const torch::Tensor& w) {
     auto raw = w.permute({0, 2, 3, 1}).reshape({(int64_t)g.K, (int64_t)rsc});
     return at::constant_pad_nd(raw, {0, cols_p - rsc}, 0).contiguous();
@@ -2374,7 +2383,12 @@ torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
                      (int)x.size(1), (int)K, (int)R, (int)S, (int)stride,
                      (int)pad);
   int taps = g.R * g.S;
-  int cols_p = (taps * g.C + 7) & ~7;
+  // small-C (flat-kernel) path: round32 so every 32-wide contraction
+  // chunk of the consumers (small-K fused GEMM, wgrad stagers) is fully
+  // in-bounds -- bounds-checked tails compile to serialized exec-masked
+  // scalar loads. The aligned span path writes exactly taps*C columns
+  // (no zeroed tail), so it keeps the tight width.
+  int cols_p = ((g.C & 7) == 0) ? taps * g.C : ((taps * g.C + 31) & ~31);
   auto col = torch::empty({g.M, cols_p}, x.options());
   NN_DISPATCH(x.scalar_type(), "im2col", [&] {
     using T = typename DevT<scalar_t>::type;

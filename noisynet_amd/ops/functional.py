@@ -137,7 +137,9 @@ def _conv_wgrad_raw(g, x, stride, padding, w_shape, col=None):
         if R * S > 1:
             # im2col-GEMM wgrad when the buffer is affordable (<2 GB):
             # one coalesced materialization pass beats per-K-chunk gather
-            cols_p = (R * S * x.shape[1] + 7) // 8 * 8
+            C_in = x.shape[1]
+            cols_p = (R * S * C_in if C_in % 8 == 0
+                      else (R * S * C_in + 31) // 32 * 32)
             m = g.shape[0] * g.shape[2] * g.shape[3]
             col_bytes = m * cols_p * x.element_size()
             if col_bytes <= 2 << 30:
