@@ -1666,41 +1666,6 @@ DEV_INLINE void stage_patch(char* patch, const T* __restrict__ x,
   }
 }
 
-// weight tile rows from channel-padded weights [K, R, S*C_pad]
-template <typename T, bool ABS_TRANSFORM, int SIGMA_MODE>
-DEV_INLINE void stage_wpad_row(char* lds, const T* __restrict__ w,
-                               const ConvGeom g, const PatchGeom p, int n0,
-                               int r, int ck) {
-  int row = threadIdx.x >> 2;
-  int seg = threadIdx.x & 3;
-  int k = n0 + row;
-  float vals[8];
-  if (k < g.K) {
-    const T* pw = w + ((int64_t)k * g.R + r) * p.Kr;
-    int c0 = ck + seg * 8;
-    if (!ABS_TRANSFORM && sizeof(T) == 2 && c0 + 8 <= p.Kr) {
-      *(bf16x8*)(lds + row * Mma<T>::STRIDE + seg * 16) =
-          *(const bf16x8*)(pw + c0);
-      return;
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int c = c0 + j;
-      float v = (c < p.Kr) ? to_f32(pw[c]) : 0.0f;
-      if (ABS_TRANSFORM) {
-        v = fabsf(v);
-        if (SIGMA_MODE == 2) v = v * v + v;
-      }
-      vals[j] = v;
-    }
-  } else {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) vals[j] = 0.0f;
-  }
-  Mma<T>::store8(lds, row, seg * 8, vals);
-}
-
-
 // register-pipelined weight staging: load the next chunk's values into
 // registers during the MFMA phase (the synchronous stage->barrier->mma
 // structure exposed the full global-load latency every round; PMC showed
