@@ -21,43 +21,59 @@ constexpr int kBlock = 256;
 template <typename T>
 __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
                                 float* __restrict__ sumsq, int64_t rows, int C) {
-  // channel/row thread split: 64 channel lanes when C allows; for C < 64
-  // pack ceil-free row groups of C lanes so no half-wave idles (MNv2's
-  // C=16..32 stem layers ran at 25-50% lane utilization otherwise)
-  int c, rgrp, ngrp;
-  if (C >= 64) {
-    c = blockIdx.x * 64 + (threadIdx.x & 63);
-    rgrp = threadIdx.x >> 6;
-    ngrp = kBlock / 64;
-  } else {
-    c = threadIdx.x % C;
-    rgrp = threadIdx.x / C;
-    ngrp = kBlock / C;
-    if (rgrp >= ngrp) return;
+  // channel/row thread split: each block owns a <=64-channel span and
+  // packs kBlock/span row groups, so a partial span (C=65's lone tail
+  // channel, MNv2's 16-32-channel stems) still uses every lane instead
+  // of leaving one active lane per 64 as the critical path.
+  __shared__ float ls[64], lsq[64];
+  int cbase = blockIdx.x * 64;
+  int cw = C - cbase;
+  if (cw > 64) cw = 64;
+  int c_l = (cw == 64) ? (threadIdx.x & 63) : (threadIdx.x % cw);
+  int rgrp = (cw == 64) ? (threadIdx.x >> 6) : (threadIdx.x / cw);
+  int ngrp = kBlock / cw;
+  if (threadIdx.x < 64) {
+    ls[threadIdx.x] = 0.0f;
+    lsq[threadIdx.x] = 0.0f;
   }
-  if (c >= C) return;
-  int rstart = blockIdx.y * ngrp + rgrp;
-  // 4 CONSECUTIVE rows per thread per iteration: the 4 in-flight loads
-  // span one contiguous ~4*C*2B region (strided variants fetched 4
-  // scattered 128-B lines each and ran 96% memory-wait)
-  const int64_t st = (int64_t)gridDim.y * ngrp * 4;
-  float s[4] = {}, sq[4] = {};
-  int64_t r = (int64_t)rstart * 4;
-  for (; r + 3 < rows; r += st) {
+  __syncthreads();
+  float acc_s = 0.0f, acc_q = 0.0f;
+  if (rgrp < ngrp) {
+    int c = cbase + c_l;
+    int rstart = blockIdx.y * ngrp + rgrp;
+    // 4 CONSECUTIVE rows per thread per iteration: the 4 in-flight loads
+    // span one contiguous ~4*C*2B region (strided variants fetched 4
+    // scattered 128-B lines each and ran 96% memory-wait)
+    const int64_t st = (int64_t)gridDim.y * ngrp * 4;
+    float s[4] = {}, sq[4] = {};
+    int64_t r = (int64_t)rstart * 4;
+    for (; r + 3 < rows; r += st) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      float v = to_f32(x[(r + j) * C + c]);
-      s[j] += v;
-      sq[j] += v * v;
+      for (int j = 0; j < 4; ++j) {
+        float v = to_f32(x[(r + j) * C + c]);
+        s[j] += v;
+        sq[j] += v * v;
+      }
     }
+    for (; r < rows; ++r) {
+      float v = to_f32(x[r * C + c]);
+      s[0] += v;
+      sq[0] += v * v;
+    }
+    acc_s = (s[0] + s[1]) + (s[2] + s[3]);
+    acc_q = (sq[0] + sq[1]) + (sq[2] + sq[3]);
   }
-  for (; r < rows; ++r) {
-    float v = to_f32(x[r * C + c]);
-    s[0] += v;
-    sq[0] += v * v;
+  // LDS pre-reduce across the block's row groups: one global atomic per
+  // (block, channel) instead of one per thread
+  if (rgrp < ngrp) {
+    atomicAdd(&ls[c_l], acc_s);
+    atomicAdd(&lsq[c_l], acc_q);
   }
-  atomicAdd(&sum[c], (s[0] + s[1]) + (s[2] + s[3]));
-  atomicAdd(&sumsq[c], (sq[0] + sq[1]) + (sq[2] + sq[3]));
+  __syncthreads();
+  if (threadIdx.x < cw) {
+    atomicAdd(&sum[cbase + threadIdx.x], ls[threadIdx.x]);
+    atomicAdd(&sumsq[cbase + threadIdx.x], lsq[threadIdx.x]);
+  }
 }
 
 template <typename T>
@@ -125,8 +141,8 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum = torch::zeros({C}, opts);
   auto sumsq = torch::zeros({C}, opts);
-  int gx = C >= 64 ? (C + 63) / 64 : 1;
-  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gx = (C + 63) / 64;
+  int ngrp = kBlock / std::min(C, 64);  // first block's packing
   // cap so atomic depth per channel (gy * ngrp) stays ~4k
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
@@ -186,41 +202,54 @@ __global__ void bn_act_bwd_reduce_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     float* __restrict__ sum_g, float* __restrict__ sum_gx, int64_t rows, int C,
     int do_relu, float act_max) {
-  int c, rgrp, ngrp;  // see bn_stats_kernel
-  if (C >= 64) {
-    c = blockIdx.x * 64 + (threadIdx.x & 63);
-    rgrp = threadIdx.x >> 6;
-    ngrp = kBlock / 64;
-  } else {
-    c = threadIdx.x % C;
-    rgrp = threadIdx.x / C;
-    ngrp = kBlock / C;
-    if (rgrp >= ngrp) return;
+  __shared__ float ls[64], lsq[64];  // see bn_stats_kernel
+  int cbase = blockIdx.x * 64;
+  int cw = C - cbase;
+  if (cw > 64) cw = 64;
+  int c_l = (cw == 64) ? (threadIdx.x & 63) : (threadIdx.x % cw);
+  int rgrp = (cw == 64) ? (threadIdx.x >> 6) : (threadIdx.x / cw);
+  int ngrp = kBlock / cw;
+  if (threadIdx.x < 64) {
+    ls[threadIdx.x] = 0.0f;
+    lsq[threadIdx.x] = 0.0f;
   }
-  if (c >= C) return;
-  int rstart = blockIdx.y * ngrp + rgrp;
-  float m = mean[c], is = invstd[c];
-  // 4 consecutive rows per iteration: see bn_stats_kernel
-  const int64_t st = (int64_t)gridDim.y * ngrp * 4;
-  float s_g[4] = {}, s_gx[4] = {};
-  auto body = [&](int64_t r, int j) {
-    int64_t i = r * C + c;
-    float yv = to_f32(y[i]);
-    float mask = 1.0f;
-    if (do_relu && yv <= 0.0f) mask = 0.0f;
-    if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
-    float gv = to_f32(g[i]) * mask;
-    s_g[j] += gv;
-    s_gx[j] += gv * (to_f32(x[i]) - m) * is;
-  };
-  int64_t r = (int64_t)rstart * 4;
-  for (; r + 3 < rows; r += st) {
+  __syncthreads();
+  float acc_g = 0.0f, acc_gx = 0.0f;
+  if (rgrp < ngrp) {
+    int c = cbase + c_l;
+    int rstart = blockIdx.y * ngrp + rgrp;
+    float m = mean[c], is = invstd[c];
+    // 4 consecutive rows per iteration: see bn_stats_kernel
+    const int64_t st = (int64_t)gridDim.y * ngrp * 4;
+    float s_g[4] = {}, s_gx[4] = {};
+    auto body = [&](int64_t r, int j) {
+      int64_t i = r * C + c;
+      float yv = to_f32(y[i]);
+      float mask = 1.0f;
+      if (do_relu && yv <= 0.0f) mask = 0.0f;
+      if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
+      float gv = to_f32(g[i]) * mask;
+      s_g[j] += gv;
+      s_gx[j] += gv * (to_f32(x[i]) - m) * is;
+    };
+    int64_t r = (int64_t)rstart * 4;
+    for (; r + 3 < rows; r += st) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) body(r + j, j);
+      for (int j = 0; j < 4; ++j) body(r + j, j);
+    }
+    for (; r < rows; ++r) body(r, 0);
+    acc_g = (s_g[0] + s_g[1]) + (s_g[2] + s_g[3]);
+    acc_gx = (s_gx[0] + s_gx[1]) + (s_gx[2] + s_gx[3]);
   }
-  for (; r < rows; ++r) body(r, 0);
-  atomicAdd(&sum_g[c], (s_g[0] + s_g[1]) + (s_g[2] + s_g[3]));
-  atomicAdd(&sum_gx[c], (s_gx[0] + s_gx[1]) + (s_gx[2] + s_gx[3]));
+  if (rgrp < ngrp) {
+    atomicAdd(&ls[c_l], acc_g);
+    atomicAdd(&lsq[c_l], acc_gx);
+  }
+  __syncthreads();
+  if (threadIdx.x < cw) {
+    atomicAdd(&sum_g[cbase + threadIdx.x], ls[threadIdx.x]);
+    atomicAdd(&sum_gx[cbase + threadIdx.x], lsq[threadIdx.x]);
+  }
 }
 
 template <typename T, bool TRAIN>
@@ -273,8 +302,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
   auto sum_gx = torch::zeros({C}, opts);
   auto gx = torch::empty_like(g);
   auto stream = c10::hip::getCurrentHIPStream();
-  int gx_blocks = C >= 64 ? (C + 63) / 64 : 1;
-  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gx_blocks = (C + 63) / 64;
+  int ngrp = kBlock / std::min(C, 64);  // first block's packing
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
   int64_t n = x.numel();
@@ -328,8 +357,8 @@ std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum_g = torch::zeros({C}, opts);
   auto sum_gx = torch::zeros({C}, opts);
-  int gx_blocks = C >= 64 ? (C + 63) / 64 : 1;
-  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gx_blocks = (C + 63) / 64;
+  int ngrp = kBlock / std::min(C, 64);  // first block's packing
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd_reduce", [&] {
@@ -410,8 +439,8 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum = torch::zeros({C}, opts);
   auto sumsq = torch::zeros({C}, opts);
-  int gx = C >= 64 ? (C + 63) / 64 : 1;
-  int ngrp = C >= 64 ? kBlock / 64 : kBlock / C;
+  int gx = (C + 63) / 64;
+  int ngrp = kBlock / std::min(C, 64);  // first block's packing
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
   auto stream = c10::hip::getCurrentHIPStream();
