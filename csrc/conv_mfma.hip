@@ -2231,28 +2231,13 @@ __global__ void im2col_flat_tmpl_kernel(const T* __restrict__ x,
 template <typename T>
 __global__ void im2col_flat_kernel(const T* __restrict__ x,
                                    T* __restrict__ out, ConvGeom g,
-                                   int cols_p, int64_t nchunks_total) {
-  extern __shared__ int tbl[];  // [cols_p] delta, then [cols_p] (r<<8)|s
-  int* rs_tbl = tbl + cols_p;
+                                   int cols_p, int64_t nchunks_total,
+                                   uint64_t cmul, int cshift, uint64_t smul,
+                                   int sshift) {
+  // (r, s, c) decode via host-computed magic-number division (the LDS
+  // table variant was 8-way bank-conflicted; real division was worse)
   int nchunks = cols_p >> 3;
   int rsc = g.R * g.S * g.C;
-  for (int k = threadIdx.x; k < cols_p; k += blockDim.x) {
-    if (k < rsc) {
-      int c = k % g.C;
-      int rs = k / g.C;
-      int s = rs % g.S;
-      int r = rs / g.S;
-      // delta into the image for (r, s, c), relative to the output
-      // pixel's (oh*stride, ow*stride) corner; r/s kept for bounds
-      tbl[k] = (r * g.W + s) * g.C + c;
-      rs_tbl[k] = (r << 8) | s;
-    } else {
-      tbl[k] = -1;
-      rs_tbl[k] = 0;
-    }
-  }
-  __syncthreads();
-
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        idx < nchunks_total; idx += (int64_t)gridDim.x * blockDim.x) {
     int seg = (int)(idx % nchunks);
@@ -2269,14 +2254,20 @@ __global__ void im2col_flat_kernel(const T* __restrict__ x,
     T vals[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int e = tbl[seg * 8 + j];
+      int k = seg * 8 + j;
       float v = 0.0f;
-      if (e >= 0) {
-        int rs = rs_tbl[seg * 8 + j];
-        int ih = oh0 + (rs >> 8);
-        int iw = ow0 + (rs & 0xff);
+      if (k < rsc) {
+        // 64-bit multiplier: for divisors just above a power of two the
+        // magic constant needs 33 bits; k < cols_p << 2^20 so the product
+        // stays well under 2^64
+        int rs = (int)(((uint64_t)k * cmul) >> (32 + cshift));
+        int c = k - rs * g.C;
+        int r = (int)(((uint64_t)rs * smul) >> (32 + sshift));
+        int ss = rs - r * g.S;
+        int ih = oh0 + r;
+        int iw = ow0 + ss;
         if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
-          v = to_f32(xn[e]);
+          v = to_f32(xn[(r * g.W + ss) * g.C + c]);
       }
       vals[j] = from_f32<T>(v);
     }
@@ -2376,10 +2367,19 @@ torch::Tensor im2col_materialize(torch::Tensor x, int64_t K, int64_t stride,
                            dim3(256), 0, stream, (const T*)x.data_ptr(),
                            (T*)col.data_ptr(), g, cols_p, nchunks_total);
       } else {
+        auto magic = [](uint32_t d, uint64_t& mul, int& sh) {
+          sh = 0;
+          while ((1u << sh) < d) ++sh;
+          mul = (uint64_t)(((__uint128_t)1 << (32 + sh)) / d) + 1;
+        };
+        uint64_t cmul, smul;
+        int cshift, sshift;
+        magic((uint32_t)g.C, cmul, cshift);
+        magic((uint32_t)g.S, smul, sshift);
         hipLaunchKernelGGL((im2col_flat_kernel<T>), dim3(blocks), dim3(256),
-                           2 * (size_t)cols_p * sizeof(int), stream,
-                           (const T*)x.data_ptr(), (T*)col.data_ptr(), g,
-                           cols_p, nchunks_total);
+                           0, stream, (const T*)x.data_ptr(),
+                           (T*)col.data_ptr(), g, cols_p, nchunks_total,
+                           cmul, cshift, smul, sshift);
       }
     }
   });
