@@ -164,3 +164,33 @@ def capture_and_emit(model, args, arrays, epoch, i, s, acc):
         np.save(os.path.join(out_dir, 'layers.npy'),
                 np.array(arrays, dtype=object), allow_pickle=True)
         print('arrays saved to', os.path.join(out_dir, 'layers.npy'))
+
+
+def compare_result_dirs(dirs, out='comparison.png', pctl=99.98):
+    """Load saved layer captures (layers*.npy) from several result dirs
+    and render them side by side (reference plot_histograms.py:589-605
+    compares four training runs this way)."""
+    loaded, names = [], []
+    for d in dirs:
+        cands = sorted(
+            f for f in os.listdir(d) if f.startswith('layers') and
+            f.endswith('.npy')) if os.path.isdir(d) else []
+        if not cands:
+            print('no layers*.npy in', d)
+            continue
+        arr = np.load(os.path.join(d, cands[-1]), allow_pickle=True)
+        loaded.append(arr)
+        names.append(os.path.basename(os.path.normpath(d)))
+    if not loaded:
+        return None
+    flat = [x for arr in loaded for x in list(arr)]
+    labels = [f'{n}:{j}' for n, arr in zip(names, loaded)
+              for j in range(len(list(arr)))]
+    plot_grid(flat, labels, out, pctl=pctl)
+    print('saved comparison to', out)
+    return out
+
+
+if __name__ == '__main__':
+    import sys
+    compare_result_dirs(sys.argv[1:] or ['.'])
