@@ -787,3 +787,35 @@ def test_bn_finalize_running_stats_match_torch():
     assert torch.allclose(rv, bn.running_var, atol=5e-3, rtol=1e-2)
     ref_mean = x.float().mean(dim=(0, 2, 3))
     assert torch.allclose(mean, ref_mean, atol=5e-3)
+
+
+@pytest.mark.gpu
+def test_maxpool2x2_odd_dims():
+    """Odd H/W leave a tail row/column outside every window; backward must
+    zero them (the all-cells-write rewrite handles them explicitly)."""
+    torch.manual_seed(7)
+    for H, W in ((9, 9), (8, 9), (9, 8), (10, 10)):
+        x = torch.randn(3, 13, H, W, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        xc = cl(x.detach()).requires_grad_(True)
+        y = ops.maxpool2x2(xc)
+        g = torch.randn_like(y)
+        y.backward(g)
+        xr = x.detach().float().cpu().requires_grad_(True)
+        yr = F.max_pool2d(xr, 2, 2)
+        yr.backward(g.float().cpu())
+        assert torch.allclose(xc.grad.float().cpu(), xr.grad, atol=1e-2), (H, W)
+
+
+@pytest.mark.gpu
+def test_bn_stats_odd_channels():
+    """Channel counts around the 64-lane span boundary (the C=65 tail
+    channel was a 1-lane critical path before per-block packing)."""
+    torch.manual_seed(8)
+    for C in (1, 24, 63, 64, 65, 100, 128, 130):
+        x = torch.randn(16, C, 7, 7, device="cuda", dtype=torch.bfloat16)
+        mean, var = ops.ext().bn_stats(cl(x))
+        ref_mean = x.float().mean(dim=(0, 2, 3))
+        ref_var = x.float().var(dim=(0, 2, 3), unbiased=False)
+        assert torch.allclose(mean.cpu(), ref_mean.cpu(), atol=5e-3), C
+        assert torch.allclose(var.cpu(), ref_var.cpu(), atol=5e-3, rtol=1e-2), C
