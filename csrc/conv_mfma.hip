@@ -163,6 +163,7 @@ struct ConvGeom {
   int flat;            // 1: contraction runs over flattened (r,s,c) in one
                        // K-loop (small-C layers: conv1's C=3 would otherwise
                        // waste 90% of each 32-deep MFMA K-step per tap)
+  int Kw;     // rows readable in the weight tensor (host may row-pad)
 };
 
 // --------------------------------------------------------------------------
@@ -246,19 +247,34 @@ DEV_INLINE void stage_w_tap(char* lds, const T* __restrict__ w,
   int k = n0 + row;
   float vals[8];
   int span = g.flat ? g.R * g.S * g.C : g.C;
-  if (k < g.K) {
+  if (k < g.Kw) {  // Kw >= K when the host row-pads (e.g. fc weights)
     const T* pw = g.flat ? (w + (int64_t)k * span)
                          : (w + (((int64_t)k * g.R + r) * g.S + s) * g.C);
     int c0 = ck + seg * 8;
+    if (sizeof(T) == 2 && c0 + 8 <= span) {
+      // one 16-B vector load; the ABS transform runs on registers
+      T raw[8];
+      *(bf16x8*)raw = *(const bf16x8*)(pw + c0);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int c = c0 + j;
-      float v = (c < span) ? to_f32(pw[c]) : 0.0f;
-      if (ABS_TRANSFORM) {
-        v = fabsf(v);
-        if (SIGMA_MODE == 2) v = v * v + v;
+      for (int j = 0; j < 8; ++j) {
+        float v = to_f32(raw[j]);
+        if (ABS_TRANSFORM) {
+          v = fabsf(v);
+          if (SIGMA_MODE == 2) v = v * v + v;
+        }
+        vals[j] = v;
       }
-      vals[j] = v;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int c = c0 + j;
+        float v = (c < span) ? to_f32(pw[c]) : 0.0f;
+        if (ABS_TRANSFORM) {
+          v = fabsf(v);
+          if (SIGMA_MODE == 2) v = v * v + v;
+        }
+        vals[j] = v;
+      }
     }
   } else {
 #pragma unroll
@@ -1094,6 +1110,7 @@ ConvGeom make_geom(int N, int H, int W, int C, int K, int R, int S, int stride,
                    int pad) {
   ConvGeom g;
   g.N = N; g.H = H; g.W = W; g.C = C; g.K = K; g.R = R; g.S = S;
+  g.Kw = K;  // weight rows staged-safe (== K unless the host row-pads)
   g.stride = stride; g.pad = pad;
   g.OH = (H + 2 * pad - R) / stride + 1;
   g.OW = (W + 2 * pad - S) / stride + 1;
@@ -1485,7 +1502,18 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
     return linear_fwd_fused_smallk(x, wq, wraw, bias, sigma_mode, factor,
                                    seed, telem, /*want_y=*/true);
   auto g = linear_geom(x, wraw.size(0));
-  auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
+  // row-pad the (small) weight matrices to the 64-row k-tile multiple so
+  // the staging guard k < Kw never diverges inside a tile (the masked
+  // scalar fallback serializes a vmcnt(0) per element)
+  int k_pad = (g.K + BM - 1) / BM * BM;
+  if (k_pad != g.K && wraw.element_size() == 2) {
+    wq = at::constant_pad_nd(wq, {0, 0, 0, k_pad - g.K}, 0);
+    wraw = at::constant_pad_nd(wraw, {0, 0, 0, k_pad - g.K}, 0);
+    g.Kw = k_pad;
+  }
+  // NOTE: out/geometry use g.K (the logical row count), never the
+  // possibly row-padded wraw.size(0)
+  auto out = torch::empty({x.size(0), (int64_t)g.K}, x.options());
   bool has_bias = bias.numel() > 0;
   torch::Tensor bias_f;
   if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
@@ -1547,7 +1575,7 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
                                    factor, seed, telem, /*want_y=*/false);
   }
   auto g = linear_geom(x, wraw.size(0));
-  auto out = torch::empty({x.size(0), wraw.size(0)}, x.options());
+  auto out = torch::empty({x.size(0), (int64_t)g.K}, x.options());
   // telemetry buffer only when requested (steady-state training has
   // telem=false; the zeros+fill launches were ~12 tiny kernels/step)
   auto tele = telem
