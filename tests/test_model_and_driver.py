@@ -230,3 +230,66 @@ def test_train_w_max_and_act_max():
     loss = out.sum()
     loss.backward()
     assert m.act_max1.grad is not None
+
+
+def test_temperature_and_scale_weights_modes():
+    """The power-law temperature drift (main.py:430-446) and the
+    scale_weights mode both perturb weights and restore after the sweep."""
+    args = make_args(['--num_sims', '1'])
+    args.stuck_at_weights = None
+    m = Net(args)
+    utils.init_model(m, args)
+    m.eval()
+    inputs = torch.rand(32, 3, 32, 32)
+    labels = torch.randint(0, 10, (32,))
+
+    inputs = torch.rand(args.batch_size * 2, 3, 32, 32)
+    labels = torch.randint(0, 10, (args.batch_size * 2,))
+    args.test_temp = 77
+    args.temperature = 25
+    w0 = m.conv1.weight.data.clone()
+    run_distortion(m, args, val_loader=(inputs, labels), mode='weights',
+                   vars=[0.0])
+    assert torch.allclose(m.conv1.weight.data, w0)
+    args.test_temp = 0
+
+    args.scale_weights = 0.5
+    run_distortion(m, args, val_loader=(inputs, labels), mode='weights',
+                   vars=[0.0])
+    assert torch.allclose(m.conv1.weight.data, w0)
+    args.scale_weights = 0
+
+
+def test_act_distortion_mode():
+    """mode='acts' flips args.distort_act for the duration of the sweep."""
+    args = make_args(['--num_sims', '1'])
+    args.stuck_at_weights = None
+    args.test_temp = 0
+    m = Net(args)
+    utils.init_model(m, args)
+    m.eval()
+    inputs = torch.rand(args.batch_size * 2, 3, 32, 32)
+    labels = torch.randint(0, 10, (args.batch_size * 2,))
+    res = run_distortion(m, args, val_loader=(inputs, labels), mode='acts',
+                         vars=[0.1])
+    assert isinstance(res, float) and res == res  # finite accuracy
+    assert not args.distort_act
+
+
+def test_selected_weights_reduced_distortion():
+    """--selected_weights protects the top-K% weights: distortion of the
+    selected set is reduced (main.py:351-377)."""
+    args = make_args(['--num_sims', '1'])
+    args.stuck_at_weights = None
+    args.test_temp = 0
+    args.selected_weights = 5
+    args.selection_criteria = 'weight_magnitude'
+    m = Net(args)
+    utils.init_model(m, args)
+    m.eval()
+    inputs = torch.rand(args.batch_size * 2, 3, 32, 32)
+    labels = torch.randint(0, 10, (args.batch_size * 2,))
+    res = run_distortion(m, args, val_loader=(inputs, labels), mode='weights',
+                         vars=[0.3])
+    assert isinstance(res, float) and res == res
+    args.selected_weights = 0
