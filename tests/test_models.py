@@ -310,3 +310,36 @@ def test_median_pool_and_feature_hooks():
     y = net(x)
     feats = hooks.get_output(x.device)
     assert len(feats) == 1 and feats[0].shape[1] == 4
+
+
+def test_gpu_augment_cpu_path():
+    """The CIFAR train-loop augmentation (random 32x32 crop from the 40x40
+    padded tensor + hflip, noisynet.py:1264-1269) runs on any device."""
+    from noisynet_amd import data as data_mod
+    x = torch.rand(8, 3, 40, 40)
+    out = data_mod.gpu_augment(x)
+    assert out.shape == (8, 3, 32, 32)
+    assert torch.isfinite(out).all()
+
+
+def test_prefetch_loader_cpu():
+    """PrefetchLoader normalizes and yields batches; without CUDA it runs
+    the same pipeline synchronously."""
+    from noisynet_amd.timm.data import SyntheticImageDataset, create_loader
+    ds = SyntheticImageDataset(num_samples=16, size=32, num_classes=10)
+    loader = create_loader(ds, input_size=(3, 32, 32), batch_size=8,
+                           is_training=True, use_prefetcher=True,
+                           num_workers=0)
+    batches = list(loader)
+    assert len(batches) >= 1
+    x, y = batches[0]
+    assert x.shape[1:] == (3, 32, 32)
+
+
+def test_random_erasing_modes():
+    from noisynet_amd.timm.data import RandomErasing
+    for mode in ('const', 'rand', 'pixel'):
+        re_op = RandomErasing(probability=1.0, mode=mode, device='cpu')
+        x = torch.rand(4, 3, 16, 16)
+        out = re_op(x.clone())
+        assert out.shape == x.shape
