@@ -1,16 +1,22 @@
 #!/usr/bin/env python3
 """Flagship benchmark: NoisyNet 4-layer CIFAR-10 ConvNet, 4-bit quantized
-activations+weights, analog noise I_max=1nA, act_max=5, bf16, synthetic
-4-bit-CIFAR-shaped data (BASELINE.json metric: images/sec whole node).
+activations (q_a=4, the README flagship config leaves weights unquantized),
+analog noise I_max=1nA, act_max=5, bf16, synthetic 4-bit-CIFAR-shaped data
+(BASELINE.json metric: images/sec whole node + top-1).
 
-Single GPU:       python bench.py --steps 30 --warmup 10
+Single GPU:       python bench.py --steps 100 --warmup 20
 Multi-GPU (driver): python -m torch.distributed.run --nnodes=1
     --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+Accuracy:         python bench.py --top1 30   (trains the flagship on the
+    learnable synthetic CIFAR and reports held-out top-1 in the JSON line)
 
 Weak scaling: per-GPU batch is fixed (--batch), global batch = N * batch.
-Each timed step is a FULL training step: forward (fused quant+conv+sigma+
-noise kernels), cross-entropy, backward (dgrad/wgrad), bucketed RCCL
-all-reduce overlapped with backward, fused SGD update with weight clamp.
+Each timed step is a FULL training step on a FRESH batch sliced from a
+GPU-resident dataset, with the reference's GPU crop/flip augmentation
+(noisynet.py:1261-1269) inside the timed region: forward (fused
+quant+conv+sigma+noise kernels), fused softmax-xent, backward (dgrad/
+wgrad), bucketed RCCL all-reduce overlapped with backward, fused SGD
+update with weight clamp.
 """
 
 import argparse
@@ -23,7 +29,9 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+from noisynet_amd import data as data_mod  # noqa: E402
 from noisynet_amd import distributed as dist_mod  # noqa: E402
+from noisynet_amd import ops  # noqa: E402
 from noisynet_amd import optim as native_optim  # noqa: E402
 from noisynet_amd import utils  # noqa: E402
 from noisynet_amd.config import broadcast_per_layer, build_noisynet_parser  # noqa: E402
@@ -34,8 +42,8 @@ from noisynet_amd.quant import finish_calibration, start_calibration  # noqa: E4
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--model", type=str, default="noisynet",
                    choices=["noisynet", "resnet18", "mobilenet_v2",
                             "efficientnet_b0"],
@@ -46,6 +54,17 @@ def parse_args():
                    choices=["bf16", "fp32"])
     p.add_argument("--no-noise", action="store_true",
                    help="disable analog/weight noise (noise-free baseline)")
+    p.add_argument("--no-augment", action="store_true",
+                   help="skip the in-loop GPU crop/flip (flagship only)")
+    p.add_argument("--top1", type=int, default=0, metavar="EPOCHS",
+                   help="after the throughput run, train the flagship on the "
+                        "learnable synthetic CIFAR for EPOCHS epochs and "
+                        "report held-out top-1 (the accuracy half of the "
+                        "BASELINE metric)")
+    p.add_argument("--graph", action="store_true", default=None,
+                   help="capture the training step in a hipGraph (default: "
+                        "auto per model)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
     p.add_argument("--seed", type=int, default=42)
     args = p.parse_args()
     if args.batch == 0:
@@ -60,7 +79,9 @@ def flagship_args(bench):
     argv = ["--q_a", "4", "--act_max", "5", "--w_max1", "0.3",
             "--LR", "0.005", "--L2_1", "0.0005", "--L2_2", "0.0002",
             "--batch_size", str(bench.batch), "--optim", "SGD",
-            "--calculate_running", "--no-augment"]
+            "--calculate_running"]
+    if bench.no_augment:
+        argv += ["--no-augment"]
     if not bench.no_noise:
         argv = ["--current", "1"] + argv
     args = build_noisynet_parser().parse_args(argv)
@@ -100,7 +121,6 @@ def build_secondary(bench, device, dtype):
         from noisynet_amd.timm.models import create_model
         model = create_model("efficientnet_b0", num_classes=num_classes)
         model.args = args
-
     model = model.to(device)
     if dtype is torch.bfloat16:
         model = model.bfloat16()
@@ -110,6 +130,77 @@ def build_secondary(bench, device, dtype):
     if device.type == "cuda":
         model = model.to(memory_format=torch.channels_last)
     return model, args, image_size, num_classes
+
+
+def run_top1(bench, device, dtype, rank, distributed):
+    """Train the flagship to convergence on the learnable synthetic CIFAR
+    and return held-out top-1 (the accuracy half of the BASELINE metric,
+    reference README.md:6-13 semantics on synthetic data)."""
+    import numpy as np
+
+    args = flagship_args(bench)
+    args.batch_size = 256
+    args.augment = True
+    torch.manual_seed(bench.seed)
+    model = Net(args)
+    utils.init_model(model, args)
+    model = model.to(device)
+    if device.type == "cuda":
+        model = model.to(memory_format=torch.channels_last)
+
+    tr, trl, te, tel = data_mod.synthesize_cifar4bit(50000, 10000)
+    X = torch.from_numpy(tr).to(device)
+    X = torch.nn.functional.pad(X, (4, 4, 4, 4))
+    y = torch.from_numpy(trl).to(device)
+    Xt = torch.from_numpy(te).to(device)
+    yt = torch.from_numpy(tel).to(device)
+    if device.type == "cuda":
+        X = X.contiguous(memory_format=torch.channels_last)
+        Xt = Xt.contiguous(memory_format=torch.channels_last)
+
+    param_groups = [
+        {"params": model.conv1.parameters(), "weight_decay": args.L2_1,
+         "lr": args.LR, "clamp": (-args.w_max1, args.w_max1)},
+        {"params": model.conv2.parameters(), "weight_decay": args.L2_2,
+         "lr": args.LR},
+        {"params": model.linear1.parameters(), "weight_decay": args.L2_3,
+         "lr": args.LR},
+        {"params": model.linear2.parameters(), "weight_decay": args.L2_4,
+         "lr": args.LR},
+        {"params": [p for m in (model.bn1, model.bn2, model.bn3, model.bn4)
+                    for p in m.parameters()], "weight_decay": 0.0,
+         "lr": args.LR},
+    ]
+    optimizer = native_optim.SGD(param_groups, lr=args.LR,
+                                 momentum=args.momentum, nesterov=True)
+    bs = args.batch_size
+    n = X.shape[0]
+    start_calibration(model)
+    curve = []
+    for epoch in range(bench.top1):
+        model.train()
+        perm = torch.randperm(n, device=device)
+        for i in range(n // bs):
+            idx = perm[i * bs:(i + 1) * bs]
+            xb = data_mod.gpu_augment(X[idx])
+            if epoch == 0 and i == 5:
+                finish_calibration(model, device)
+            out = model(xb, epoch, i)
+            loss = ops.cross_entropy(out, y[idx])
+            optimizer.zero_grad(set_to_none=False)
+            loss.backward()
+            optimizer.step()
+        model.eval()
+        correct = 0
+        with torch.no_grad():
+            for i in range(0, Xt.shape[0], 1000):
+                out = model(Xt[i:i + 1000], epoch, 100)
+                correct += int((out.argmax(1) == yt[i:i + 1000]).sum())
+        acc = 100.0 * correct / Xt.shape[0]
+        curve.append(round(acc, 2))
+        if rank == 0:
+            print("# top1 epoch %d: %.2f%%" % (epoch, acc), file=sys.stderr)
+    return curve
 
 
 def main():
@@ -149,13 +240,20 @@ def main():
 
     dp = dist_mod.DataParallel(model) if distributed else None
 
-    # synthetic 4-bit-grid data, resident on device
+    # GPU-resident synthetic 4-bit-grid dataset, several batches deep so
+    # every timed step sees a fresh slice (reference noisynet.py:1249-1276)
+    flagship = bench.model == "noisynet"
+    augment = flagship and not bench.no_augment
+    n_slices = 4
     g = torch.Generator().manual_seed(bench.seed + rank)
-    data = (torch.randint(0, 16, (bench.batch, 3, image_size, image_size),
+    store = image_size + 8 if augment else image_size
+    data = (torch.randint(0, 16,
+                          (bench.batch * n_slices, 3, store, store),
                           generator=g).to(device=device, dtype=dtype) / 15.0)
     if device.type == "cuda":
         data = data.contiguous(memory_format=torch.channels_last)
-    labels = torch.randint(0, num_classes, (bench.batch,), generator=g).to(device)
+    labels = torch.randint(0, num_classes, (bench.batch * n_slices,),
+                           generator=g).to(device)
 
     if bench.model == "noisynet":
         param_groups = [
@@ -176,13 +274,17 @@ def main():
     else:
         optimizer = native_optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                                      weight_decay=1e-4, nesterov=False)
-    criterion = torch.nn.CrossEntropyLoss()
 
     def step(i):
         model.train()
-        out = model(data, 0, i) if bench.model != "efficientnet_b0" \
-            else model(data)
-        loss = criterion(out.float(), labels)
+        s = (i % n_slices) * bench.batch
+        xb = data[s:s + bench.batch]
+        if augment:
+            xb = data_mod.gpu_augment(xb)
+        yb = labels[s:s + bench.batch]
+        out = model(xb, 0, i) if bench.model != "efficientnet_b0" \
+            else model(xb)
+        loss = ops.cross_entropy(out, yb)
         optimizer.zero_grad(set_to_none=False)
         loss.backward()
         if dp is not None:
@@ -194,7 +296,10 @@ def main():
     start_calibration(model)
     with torch.no_grad():
         for i in range(5):
-            model(data, 0, i)
+            xb = data[:bench.batch]
+            if augment:
+                xb = data_mod.gpu_augment(xb)
+            model(xb, 0, i) if bench.model != "efficientnet_b0" else model(xb)
     finish_calibration(model, device)
 
     for i in range(bench.warmup):
@@ -222,8 +327,12 @@ def main():
     ms_per_step = elapsed / bench.steps * 1000.0
     images_per_sec = n_gpus * bench.batch * bench.steps / elapsed
 
+    top1_curve = None
+    if bench.top1 > 0 and bench.model == "noisynet":
+        top1_curve = run_top1(bench, device, dtype, rank, distributed)
+
     if rank == 0:
-        print(json.dumps({
+        out = {
             "metric": "images/sec (whole node) + top-1, NoisyNet 4-bit "
                       "CIFAR-10 I_max=1nA, 1/2/4/8 GPU",
             "value": round(images_per_sec, 2),
@@ -243,6 +352,7 @@ def main():
                 "global_batch": n_gpus * bench.batch,
                 "seq_len": None,
                 "image_size": image_size,
+                "augment": bool(augment),
                 "q_a": 4,
                 "q_w": 4 if bench.model == "resnet18" else 0,
                 "current_nA": (0 if (bench.no_noise or bench.model != "noisynet")
@@ -250,7 +360,12 @@ def main():
                 "act_max": 5 if bench.model == "noisynet" else 0,
                 "parallelism": "dp%d" % n_gpus,
             },
-        }))
+        }
+        if top1_curve is not None:
+            out["top1"] = top1_curve[-1]
+            out["top1_curve"] = top1_curve
+            out["top1_epochs"] = bench.top1
+        print(json.dumps(out))
 
     sys.stdout.flush()
     if distributed:

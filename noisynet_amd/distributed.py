@@ -63,12 +63,14 @@ def reduce_tensor(tensor, world_size=None):
 
 
 class GradBucket:
-    __slots__ = ("params", "numel", "flat", "work", "ready")
+    __slots__ = ("params", "numel", "bytes", "dtype", "flat", "work", "ready")
 
-    def __init__(self):
+    def __init__(self, dtype):
         self.params = []
         self.numel = 0
-        self.flat = None
+        self.bytes = 0
+        self.dtype = dtype
+        self.flat = None      # persistent flat buffer, allocated on first use
         self.work = None
         self.ready = 0
 
@@ -77,15 +79,28 @@ class DataParallel:
     """Bucketed, overlapped gradient all-reduce.
 
     Buckets are built in REVERSE parameter order (gradients become ready
-    back-to-front during backward), each up to ``bucket_cap_mb``. When the
-    last gradient of a bucket lands (post-accumulate-grad hook), the bucket
-    is flattened into a persistent buffer and an async all-reduce is
-    launched immediately -- RCCL runs it on its own stream, overlapping the
-    remaining backward compute. ``finish()`` waits for all works and
-    scatters the averaged gradients back.
+    back-to-front during backward) and are homogeneous in dtype, each up to
+    ``bucket_cap_mb`` of ACTUAL grad bytes (bf16 params get bf16-sized
+    buckets). When the last gradient of a bucket lands
+    (post-accumulate-grad hook), the grads are packed into the bucket's
+    persistent flat buffer and an async all-reduce is launched immediately
+    -- RCCL runs it on its own stream, overlapping the remaining backward
+    compute. ``finish()`` waits for all works and scatters the averaged
+    gradients back.
+
+    A parameter whose grad is absent at reduce time (e.g. manually-updated
+    clip params like ``w_max1`` under ``--train_w_max``, drivers/cifar.py)
+    contributes ZEROS to its bucket segment, so the collective shape is
+    identical on every rank regardless of which params got autograd grads;
+    the averaged values are scattered back only into grads that exist.
+
+    Gradient accumulation: wrap non-final backwards in ``no_sync()`` so the
+    hooks skip launching; the final (unwrapped) backward reduces the fully
+    accumulated grads. One ``finish()`` per synchronized backward.
 
     With ``delay_allreduce=True`` hooks are skipped and ``finish()`` does
-    one bucketed all-reduce sweep after backward (apex parity mode).
+    one bucketed all-reduce sweep after backward (apex parity mode,
+    reference main.py:798).
     """
 
     def __init__(self, model, bucket_cap_mb=25, delay_allreduce=False,
@@ -96,21 +111,25 @@ class DataParallel:
         self.enabled = dist.is_initialized() and dist.get_world_size() > 1
         self.world_size = dist.get_world_size() if self.enabled else 1
         self._hooks = []
+        self._sync = True
 
         params = [p for p in model.parameters() if p.requires_grad]
         cap = int(bucket_cap_mb * 1024 * 1024)
         self.buckets = []
         self.param_bucket = {}
-        bucket = GradBucket()
+        bucket = None
         for p in reversed(params):
             bytes_p = p.numel() * p.element_size()
-            if bucket.params and (bucket.numel * 4 + bytes_p) > cap:
-                self.buckets.append(bucket)
-                bucket = GradBucket()
+            if bucket is None or bucket.dtype != p.dtype or \
+                    (bucket.params and bucket.bytes + bytes_p > cap):
+                if bucket is not None and bucket.params:
+                    self.buckets.append(bucket)
+                bucket = GradBucket(p.dtype)
             bucket.params.append(p)
             bucket.numel += p.numel()
+            bucket.bytes += bytes_p
             self.param_bucket[p] = bucket
-        if bucket.params:
+        if bucket is not None and bucket.params:
             self.buckets.append(bucket)
 
         if self.enabled:
@@ -130,14 +149,49 @@ class DataParallel:
                            torch.float64, torch.int64, torch.int32):
                 dist.broadcast(t.data, src=0, group=self.group)
 
+    def no_sync(self):
+        """Context manager: skip gradient sync (accumulation steps)."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def ctx():
+            self._sync = False
+            try:
+                yield
+            finally:
+                self._sync = True
+        return ctx()
+
     def _launch(self, bucket):
-        flat = torch.cat([p.grad.reshape(-1) for p in bucket.params])
-        flat /= self.world_size
-        bucket.flat = flat
+        # pack grads into the persistent flat buffer (no per-step torch.cat
+        # allocation); absent grads contribute zeros so every rank reduces
+        # the same shape.
+        dev = bucket.params[0].device
+        for p in bucket.params:
+            if p.grad is not None:
+                dev = p.grad.device
+                break
+        if bucket.flat is None or bucket.flat.device != dev:
+            bucket.flat = torch.empty(bucket.numel, dtype=bucket.dtype,
+                                      device=dev)
+        flat = bucket.flat
+        off = 0
+        inv = 1.0 / self.world_size
+        for p in bucket.params:
+            n = p.numel()
+            seg = flat[off:off + n]
+            if p.grad is not None:
+                seg.copy_(p.grad.reshape(-1))
+            else:
+                seg.zero_()
+            off += n
+        flat.mul_(inv)
         bucket.work = dist.all_reduce(flat, op=dist.ReduceOp.SUM,
                                       group=self.group, async_op=True)
 
     def _hook(self, p):
+        if not self._sync:
+            return
         bucket = self.param_bucket[p]
         bucket.ready += 1
         if bucket.ready == len(bucket.params):
@@ -148,19 +202,18 @@ class DataParallel:
         if not self.enabled:
             return
         for bucket in self.buckets:
-            if self.delay or bucket.work is None:
-                # grads may be absent for params unused this step
-                if all(p.grad is not None for p in bucket.params):
-                    self._launch(bucket)
-            if bucket.work is not None:
-                bucket.work.wait()
-                off = 0
-                for p in bucket.params:
-                    n = p.numel()
+            if bucket.work is None:
+                # delay mode, or a bucket whose last hook never fired
+                # (params with manually-written grads / unused params)
+                self._launch(bucket)
+            bucket.work.wait()
+            off = 0
+            for p in bucket.params:
+                n = p.numel()
+                if p.grad is not None:
                     p.grad.copy_(bucket.flat[off:off + n].view_as(p.grad))
-                    off += n
+                off += n
             bucket.work = None
-            bucket.flat = None
             bucket.ready = 0
 
     def remove(self):

@@ -26,7 +26,8 @@ except ImportError:  # pragma: no cover
 from .. import distributed as dist_mod
 from ..timm.data import (Dataset, FastCollateMixup, create_loader,
                          resolve_data_config)
-from ..timm.loss import LabelSmoothingCrossEntropy, SoftTargetCrossEntropy
+from ..timm.loss import (FusedCrossEntropy, LabelSmoothingCrossEntropy,
+                         SoftTargetCrossEntropy)
 from ..timm.models import create_model, resume_checkpoint
 from ..timm.optim import create_optimizer
 from ..timm.scheduler import create_scheduler
@@ -215,12 +216,12 @@ def main(argv=None):
 
     if args.mixup > 0.:
         train_loss_fn = SoftTargetCrossEntropy()
-        validate_loss_fn = nn.CrossEntropyLoss()
+        validate_loss_fn = FusedCrossEntropy()
     elif args.smoothing:
         train_loss_fn = LabelSmoothingCrossEntropy(smoothing=args.smoothing)
-        validate_loss_fn = nn.CrossEntropyLoss()
+        validate_loss_fn = FusedCrossEntropy()
     else:
-        train_loss_fn = validate_loss_fn = nn.CrossEntropyLoss()
+        train_loss_fn = validate_loss_fn = FusedCrossEntropy()
 
     eval_metric = args.eval_metric
     best_metric = None

@@ -22,6 +22,7 @@ import torch.nn as nn
 
 from .. import data as data_mod
 from .. import distributed as dist_mod
+from .. import ops
 from .. import optim as native_optim
 from .. import utils
 from ..config import build_main_parser
@@ -125,7 +126,12 @@ def load_from_checkpoint(args, model, optimizer=None, device='cpu'):
 
 def train(model, dp, args, train_loader, val_loader, optimizer, device,
           start_epoch=0):
-    criterion = nn.CrossEntropyLoss()
+    # fused softmax-xent HIP kernel unless the L3 gradient penalty needs a
+    # twice-differentiable loss (main.py:894-904)
+    if args.L3 > 0:
+        criterion = nn.CrossEntropyLoss()
+    else:
+        criterion = lambda out, tgt: ops.cross_entropy(out, tgt)  # noqa: E731
     best_acc = 0.0
     num_iter = len(train_loader)
     for epoch in range(start_epoch, args.epochs):

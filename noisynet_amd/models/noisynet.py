@@ -218,7 +218,8 @@ class Net(nn.Module):
                            self.bn4.running_mean, self.bn4.running_var,
                            self.training or not args.track_running_stats,
                            self.bn4.momentum, self.bn4.eps,
-                           relu=False, act_max=0.0)
+                           relu=False, act_max=0.0,
+                           sync=getattr(args, 'sync_bn', False))
         self.linear2_out = x
         return x
 

@@ -35,8 +35,13 @@ class UniformQuantize(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_output):
         (input,) = ctx.saved_tensors
-        grad = ops.reference.fake_quant_backward(
-            grad_output, input, ctx.min_value, ctx.max_value)
+        if input.is_cuda:
+            grad = ops.ext().ste_mask(grad_output, input,
+                                      float(ctx.min_value),
+                                      float(ctx.max_value))
+        else:
+            grad = ops.reference.fake_quant_backward(
+                grad_output, input, ctx.min_value, ctx.max_value)
         return grad, None, None, None, None, None, None
 
 

@@ -1,1 +1,2 @@
-from .cross_entropy import LabelSmoothingCrossEntropy, SoftTargetCrossEntropy  # noqa: F401
+from .cross_entropy import (FusedCrossEntropy, LabelSmoothingCrossEntropy,  # noqa: F401
+                            SoftTargetCrossEntropy)

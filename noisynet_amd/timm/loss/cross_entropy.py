@@ -25,3 +25,12 @@ class SoftTargetCrossEntropy(nn.Module):
     def forward(self, x, target):
         loss = torch.sum(-target * F.log_softmax(x, dim=-1), dim=-1)
         return loss.mean()
+
+
+class FusedCrossEntropy(nn.Module):
+    """Plain cross-entropy routed through the fused softmax-xent HIP kernel
+    (csrc/softmax_xent.hip) on GPU; eager log_softmax+nll on CPU."""
+
+    def forward(self, x, target):
+        from ... import ops
+        return ops.cross_entropy(x, target)

@@ -1,29 +1,44 @@
-"""Forward-hook feature collector (reference timm/models/feature_hooks.py:31)."""
+"""Feature-map capture via module forward hooks.
 
-from collections import OrderedDict, defaultdict
-from functools import partial
+Capability parity with the reference's hook collector used by
+``EfficientNetFeatures`` (timm/models/feature_hooks.py:31): given a list of
+hook specs ``[{'name': module_name, 'type': 'forward'|'forward_pre'}, ...]``
+and the model's ``named_modules()``, it records each hooked module's output
+tensor per device; ``get_output(device)`` drains the captures in hook
+order.
+"""
 
 
 class FeatureHooks:
     def __init__(self, hooks, named_modules):
-        modules = {k: v for k, v in named_modules}
-        for h in hooks:
-            hook_name = h['name']
-            m = modules[hook_name]
-            hook_fn = partial(self._collect_output_hook, hook_name)
-            if h.get('type', 'forward') == 'forward_pre':
-                m.register_forward_pre_hook(hook_fn)
+        by_name = dict(named_modules)
+        self._order = [spec['name'] for spec in hooks]
+        self._captures = {}  # device -> {name: tensor}
+        for spec in hooks:
+            module = by_name[spec['name']]
+            if spec.get('type', 'forward') == 'forward_pre':
+                module.register_forward_pre_hook(
+                    self._make_pre_hook(spec['name']))
             else:
-                m.register_forward_hook(hook_fn)
-        self._feature_outputs = defaultdict(OrderedDict)
+                module.register_forward_hook(
+                    self._make_fwd_hook(spec['name']))
 
-    def _collect_output_hook(self, name, *args):
-        x = args[-1]  # tensor we want is last argument, output for fwd hooks
-        if isinstance(x, tuple):
-            x = x[0]
-        self._feature_outputs[x.device][name] = x
+    def _store(self, name, tensor):
+        if isinstance(tensor, tuple):
+            tensor = tensor[0]
+        self._captures.setdefault(tensor.device, {})[name] = tensor
+
+    def _make_fwd_hook(self, name):
+        def hook(module, inputs, output):
+            self._store(name, output)
+        return hook
+
+    def _make_pre_hook(self, name):
+        def hook(module, inputs):
+            self._store(name, inputs[0] if isinstance(inputs, tuple)
+                        else inputs)
+        return hook
 
     def get_output(self, device):
-        output = tuple(self._feature_outputs[device].values())
-        self._feature_outputs[device] = OrderedDict()
-        return output
+        grabbed = self._captures.pop(device, {})
+        return tuple(grabbed[n] for n in self._order if n in grabbed)
