@@ -93,27 +93,157 @@ def _bench_worker(rank, world_size, port, q):
     q.put((rank, buf.getvalue()))
 
 
-def test_bench_multiprocess_cpu():
-    """bench.py runs under the torchrun env contract with gloo, ws=2."""
+@pytest.mark.parametrize("world_size", [2, 8])
+def test_bench_multiprocess_cpu(world_size):
+    """bench.py runs under the torchrun env contract with gloo -- ws=2 and
+    the full 8-rank shape the driver's SCALE run will launch (dry-run of
+    rendezvous, bucket all-reduce and teardown at dp8)."""
     import json
-    world_size = 2
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_bench_worker, args=(r, world_size, 29521, q))
+    procs = [ctx.Process(target=_bench_worker,
+                         args=(r, world_size, 29521 + world_size, q))
              for r in range(world_size)]
     for p in procs:
         p.start()
-    outs = {r: o for r, o in (q.get(timeout=300) for _ in range(world_size))}
+    outs = {r: o for r, o in (q.get(timeout=600) for _ in range(world_size))}
     for p in procs:
         p.join(timeout=300)
     # rank 0 printed exactly one JSON line
     lines = [l for l in outs[0].strip().splitlines() if l.startswith("{")]
     assert len(lines) == 1
     rec = json.loads(lines[0])
-    assert rec["n_gpus"] == 2
+    assert rec["n_gpus"] == world_size
     assert rec["scaling"] == "weak"
     assert rec["value"] > 0
-    assert outs[1].strip() == "" or "{" not in outs[1]
+    for r in range(1, world_size):
+        assert "{" not in outs[r]
+
+
+def _missing_grad_worker(rank, world_size, port, q, delay):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    from noisynet_amd import distributed as dist_mod
+
+    dist_mod.init_distributed(backend="gloo")
+    torch.manual_seed(55)
+
+    class M(torch.nn.Module):
+        """Model with manually-updated clip params (w_max1-style,
+        drivers/cifar.py:283) that never receive autograd grads, sharing
+        buckets with real params; plus an unused layer."""
+
+        def __init__(self):
+            super().__init__()
+            self.fc1 = torch.nn.Linear(16, 32)
+            self.w_max1 = torch.nn.Parameter(torch.tensor(0.25))
+            self.fc2 = torch.nn.Linear(32, 4)
+            self.unused = torch.nn.Linear(8, 8)
+            self.w_min1 = torch.nn.Parameter(torch.tensor(-0.25))
+
+        def forward(self, x):
+            return self.fc2(torch.relu(self.fc1(x)))
+
+    model = M()
+    dp = dist_mod.DataParallel(model, bucket_cap_mb=1, delay_allreduce=delay)
+    torch.manual_seed(300 + rank)
+    x = torch.randn(8, 16)
+    y = torch.randn(8, 4)
+    torch.nn.functional.mse_loss(model(x), y).backward()
+    dp.finish()  # must not hang or skip fc* grads
+
+    grad = model.fc1.weight.grad.clone()
+    ref = []
+    for r in range(world_size):
+        torch.manual_seed(55)
+        m2 = M()
+        torch.manual_seed(300 + r)
+        xr, yr = torch.randn(8, 16), torch.randn(8, 4)
+        torch.nn.functional.mse_loss(m2(xr), yr).backward()
+        ref.append(m2.fc1.weight.grad)
+    expected = torch.stack(ref).mean(0)
+    ok = torch.allclose(grad, expected, atol=1e-6)
+    ok_none = model.w_max1.grad is None and model.unused.weight.grad is None
+    q.put((rank, bool(ok and ok_none)))
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world_size,delay", [(2, False), (4, False), (4, True)])
+def test_dataparallel_missing_grads_and_ws4(world_size, delay):
+    """Buckets containing never-gradded params (manual w_max updates) and
+    unused layers still reduce the real grads identically on every rank --
+    at world_size 4 as well as 2 (8-GPU readiness drill)."""
+    port = 29541 + world_size + (100 if delay else 0)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_missing_grad_worker,
+                         args=(r, world_size, port, q, delay))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world_size)]
+    for p in procs:
+        p.join(timeout=120)
+    assert all(ok for _, ok in results), results
+
+
+def _accum_worker(rank, world_size, port, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    from noisynet_amd import distributed as dist_mod
+
+    dist_mod.init_distributed(backend="gloo")
+    torch.manual_seed(66)
+    model = torch.nn.Linear(16, 4)
+    dp = dist_mod.DataParallel(model, bucket_cap_mb=1)
+    torch.manual_seed(400 + rank)
+    x1, y1 = torch.randn(8, 16), torch.randn(8, 4)
+    x2, y2 = torch.randn(8, 16), torch.randn(8, 4)
+    with dp.no_sync():
+        torch.nn.functional.mse_loss(model(x1), y1).backward()
+    torch.nn.functional.mse_loss(model(x2), y2).backward()
+    dp.finish()
+    grad = model.weight.grad.clone()
+
+    ref = []
+    for r in range(world_size):
+        torch.manual_seed(66)
+        m2 = torch.nn.Linear(16, 4)
+        torch.manual_seed(400 + r)
+        a1, b1 = torch.randn(8, 16), torch.randn(8, 4)
+        a2, b2 = torch.randn(8, 16), torch.randn(8, 4)
+        torch.nn.functional.mse_loss(m2(a1), b1).backward()
+        torch.nn.functional.mse_loss(m2(a2), b2).backward()
+        ref.append(m2.weight.grad)
+    expected = torch.stack(ref).mean(0)
+    q.put((rank, bool(torch.allclose(grad, expected, atol=1e-6))))
+    dist.destroy_process_group()
+
+
+def test_dataparallel_grad_accumulation_no_sync():
+    """no_sync() accumulation: only the final backward is reduced, over the
+    fully accumulated gradients."""
+    world_size = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_accum_worker, args=(r, world_size, 29551, q))
+             for r in range(world_size)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(world_size)]
+    for p in procs:
+        p.join(timeout=120)
+    assert all(ok for _, ok in results), results
 
 
 def _syncbn_worker(rank, world_size, port, q):
