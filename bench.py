@@ -275,7 +275,16 @@ def main():
         optimizer = native_optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                                      weight_decay=1e-4, nesterov=False)
 
-    def step(i):
+    # hipGraph capture: one graph launch per step instead of ~900 host
+    # launches (decisive for the host-bound secondary models). Single
+    # process only -- the multi-GPU path keeps eager launches so the
+    # bucketed RCCL all-reduce can overlap backward.
+    use_graph = bench.graph
+    if use_graph is None:
+        use_graph = device.type == "cuda" and not distributed
+    use_graph = use_graph and device.type == "cuda" and not distributed
+
+    def eager_step(i):
         model.train()
         s = (i % n_slices) * bench.batch
         xb = data[s:s + bench.batch]
@@ -301,6 +310,41 @@ def main():
                 xb = data_mod.gpu_augment(xb)
             model(xb, 0, i) if bench.model != "efficientnet_b0" else model(xb)
     finish_calibration(model, device)
+
+    if use_graph:
+        from noisynet_amd.graphs import GraphedTrainStep
+
+        static_x = torch.empty(bench.batch, 3, image_size, image_size,
+                               device=device, dtype=dtype).contiguous(
+                                   memory_format=torch.channels_last)
+        static_y = torch.empty(bench.batch, dtype=torch.int64, device=device)
+
+        def fill(i):
+            s = (i % n_slices) * bench.batch
+            xb = data[s:s + bench.batch]
+            if augment:
+                xb = data_mod.gpu_augment(xb)
+            static_x.copy_(xb)
+            static_y.copy_(labels[s:s + bench.batch])
+
+        def graph_body():
+            out = model(static_x, 0, 1000) \
+                if bench.model != "efficientnet_b0" else model(static_x)
+            loss = ops.cross_entropy(out, static_y)
+            optimizer.zero_grad(set_to_none=False)
+            loss.backward()
+            optimizer.step()
+            return loss
+
+        model.train()
+        fill(0)
+        gstep = GraphedTrainStep(graph_body)
+
+        def step(i):
+            fill(i)
+            return gstep.replay()
+    else:
+        step = eager_step
 
     for i in range(bench.warmup):
         step(i + 100)  # i>=20: telemetry off in steady state
@@ -353,6 +397,7 @@ def main():
                 "seq_len": None,
                 "image_size": image_size,
                 "augment": bool(augment),
+                "hip_graph": bool(use_graph),
                 "q_a": 4,
                 "q_w": 4 if bench.model == "resnet18" else 0,
                 "current_nA": (0 if (bench.no_noise or bench.model != "noisynet")

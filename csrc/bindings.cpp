@@ -12,6 +12,8 @@ torch::Tensor relu_clip_fwd(torch::Tensor x, bool relu, double act_max);
 torch::Tensor relu_clip_bwd(torch::Tensor g, torch::Tensor y, bool relu,
                             double act_max);
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed);
+void set_seed_buffer(torch::Tensor t);
+void clear_seed_buffer();
 
 // optimizer.hip
 void sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
@@ -128,6 +130,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_clip_fwd", &relu_clip_fwd);
   m.def("relu_clip_bwd", &relu_clip_bwd);
   m.def("dropout_fwd", &dropout_fwd);
+  m.def("set_seed_buffer", &set_seed_buffer);
+  m.def("clear_seed_buffer", &clear_seed_buffer);
   m.def("sgd_step", &sgd_step);
   m.def("adamw_step", &adamw_step);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);

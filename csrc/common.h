@@ -119,6 +119,29 @@ DEV_INLINE float uniform_pm(uint64_t seed, uint64_t ctr, float a) {
 }
 
 // ---------------------------------------------------------------------------
+// hipGraph-replay seed indirection
+//
+// Under graph capture every kernel argument is frozen into the graph, so a
+// by-value seed would replay the SAME noise each step. Python installs a
+// 1-element int64 device buffer (the step counter, incremented by a node
+// inside the captured region); RNG kernels receive its pointer and mix the
+// live counter with the per-launch salt captured at record time. base ==
+// nullptr (eager mode) keeps the original by-value behaviour bit-for-bit.
+// ---------------------------------------------------------------------------
+DEV_INLINE uint64_t graph_seed(const int64_t* base, uint64_t salt) {
+  if (base == nullptr) return salt;
+  uint64_t s = (uint64_t)(*base) * 0x9E3779B97F4A7C15ull + salt;
+  s ^= s >> 31;
+  s *= 0xBF58476D1CE4E5B9ull;
+  s ^= s >> 29;
+  return s;
+}
+
+// device pointer to the live step counter (nullptr = eager mode);
+// defined in elementwise.hip, set via set_seed_buffer()/clear_seed_buffer()
+extern int64_t* g_seed_base;
+
+// ---------------------------------------------------------------------------
 // wave/block reductions
 // ---------------------------------------------------------------------------
 DEV_INLINE float wave_sum(float v) {

@@ -298,7 +298,9 @@ void conv_fwd_kernel(const T* __restrict__ x, const T* __restrict__ wq,
                      const T* __restrict__ wraw, const float* __restrict__ bias,
                      T* __restrict__ out, ConvGeom g,
                      const float* __restrict__ factor_p,
-                     uint64_t seed, float* __restrict__ telem /* [3] */) {
+                     uint64_t seed, float* __restrict__ telem /* [3] */,
+                     const int64_t* __restrict__ seed_base = nullptr) {
+  seed = graph_seed(seed_base, seed);
   const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
   // grid: x = n-tiles, y = m-tiles
   int n0 = blockIdx.x * BN;
@@ -431,7 +433,9 @@ void conv_fwd_smallk_kernel(const T* __restrict__ x2, const T* __restrict__ wq,
                             const float* __restrict__ bias,
                             T* __restrict__ out, int64_t M, int K, int Kc,
                             const float* __restrict__ factor_p, uint64_t seed,
-                            float* __restrict__ telem) {
+                            float* __restrict__ telem,
+                            const int64_t* __restrict__ seed_base = nullptr) {
+  seed = graph_seed(seed_base, seed);
   const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
   constexpr int STR = Mma<T>::STRIDE;
   const int CH = (Kc + 31) >> 5;     // 32-wide contraction chunks (<= 4),
@@ -1204,7 +1208,7 @@ std::vector<torch::Tensor> conv_fwd_fused_impl(torch::Tensor x,
       hipLaunchKernelGGL((conv_fwd_kernel<T, decltype(wy)::value,
                           decltype(sm)::value, decltype(tl)::value,
                           decltype(bi)::value>), grid, dim3(kBlock), lds,
-                         stream, xp, wqp, wrp, bp, op, g, f, sd, tp);
+                         stream, xp, wqp, wrp, bp, op, g, f, sd, tp, g_seed_base);
     };
     using TT = std::true_type; using FF = std::false_type;
     using S1 = std::integral_constant<int, 1>;
@@ -1461,7 +1465,7 @@ std::vector<torch::Tensor> linear_fwd_fused_smallk(
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             (int)lds);
       hipLaunchKernelGGL(kfn, dim3(blocks), dim3(kBlock), lds, stream, xp,
-                         wqp, wrp, bp, op, M, K, Kc, f, sd, tp);
+                         wqp, wrp, bp, op, M, K, Kc, f, sd, tp, g_seed_base);
     };
     using TT = std::true_type; using FF = std::false_type;
     using S0 = std::integral_constant<int, 0>;
@@ -1546,7 +1550,7 @@ std::vector<torch::Tensor> linear_fwd_fused(torch::Tensor x, torch::Tensor wq,
       hipLaunchKernelGGL((conv_fwd_kernel<T, true, decltype(sm)::value,
                           decltype(tl)::value, decltype(bi)::value>), grid,
                          dim3(kBlock), lds, stream, xp, wqp, wrp, bp, op, g,
-                         f, sd, tp);
+                         f, sd, tp, g_seed_base);
     };
     using TT = std::true_type; using FF = std::false_type;
     using S1 = std::integral_constant<int, 1>;
@@ -1603,20 +1607,20 @@ std::vector<torch::Tensor> sigma_noise_linear_impl(torch::Tensor x,
       if (telem)
         hipLaunchKernelGGL((conv_fwd_kernel<T, false, 1, true, false>), grid,
                            dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
-                           op, g, f, sd, tp);
+                           op, g, f, sd, tp, g_seed_base);
       else
         hipLaunchKernelGGL((conv_fwd_kernel<T, false, 1, false, false>), grid,
                            dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
-                           op, g, f, sd, tp);
+                           op, g, f, sd, tp, g_seed_base);
     } else {
       if (telem)
         hipLaunchKernelGGL((conv_fwd_kernel<T, false, 2, true, false>), grid,
                            dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
-                           op, g, f, sd, tp);
+                           op, g, f, sd, tp, g_seed_base);
       else
         hipLaunchKernelGGL((conv_fwd_kernel<T, false, 2, false, false>), grid,
                            dim3(kBlock), lds, stream, xp, wrp, wrp, nullptr,
-                           op, g, f, sd, tp);
+                           op, g, f, sd, tp, g_seed_base);
     }
   });
   HIP_CHECK_LAST();
@@ -1750,7 +1754,9 @@ void conv_fwd_patch_kernel(const T* __restrict__ x, const T* __restrict__ wq,
                            const float* __restrict__ bias,
                            T* __restrict__ out, ConvGeom g, PatchGeom p,
                            const float* __restrict__ factor_p, uint64_t seed,
-                           float* __restrict__ telem) {
+                           float* __restrict__ telem,
+                           const int64_t* __restrict__ seed_base = nullptr) {
+  seed = graph_seed(seed_base, seed);
   const float factor = (SIGMA_MODE > 0) ? factor_p[0] : 0.0f;
   constexpr int EB = (int)sizeof(T) == 4 ? 4 : 2;
   constexpr int STR = Mma<T>::STRIDE;
@@ -2020,7 +2026,7 @@ std::vector<torch::Tensor> conv_fwd_fused_patch_impl(
       hipLaunchKernelGGL((conv_fwd_patch_kernel<T, decltype(wy)::value,
                           decltype(sm)::value, decltype(tl)::value,
                           decltype(bi)::value>), grid, dim3(kBlock), lds,
-                         stream, xp, wqp, wrp, bp, op, g, p, f, sd, tp);
+                         stream, xp, wqp, wrp, bp, op, g, p, f, sd, tp, g_seed_base);
     };
     using TT = std::true_type; using FF = std::false_type;
     using S0 = std::integral_constant<int, 0>;

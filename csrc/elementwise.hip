@@ -29,7 +29,9 @@ template <typename T, bool STOCH>
 __global__ void fake_quant_kernel(const T* __restrict__ x, T* __restrict__ out,
                                   int64_t n, float min_value, float inv_scale,
                                   float scale, float qmax, float stoch,
-                                  uint64_t seed) {
+                                  uint64_t seed,
+                                  const int64_t* __restrict__ seed_base) {
+  seed = graph_seed(seed_base, seed);
   int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
   for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
        i0 < n; i0 += stride) {
@@ -73,7 +75,9 @@ __global__ void ste_mask_kernel(const T* __restrict__ g, const T* __restrict__ x
 // --------------------------------------------------------------------------
 template <typename T>
 __global__ void mult_uniform_kernel(const T* __restrict__ x, T* __restrict__ out,
-                                    int64_t n, float a, uint64_t seed) {
+                                    int64_t n, float a, uint64_t seed,
+                                    const int64_t* __restrict__ seed_base) {
+  seed = graph_seed(seed_base, seed);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     float v = to_f32(x[i]);
@@ -102,7 +106,9 @@ __global__ void relu_clip_kernel(const T* __restrict__ x, T* __restrict__ out,
 template <typename T>
 __global__ void dropout_kernel(const T* __restrict__ x, T* __restrict__ out,
                                T* __restrict__ mask, int64_t n, float p,
-                               float inv_keep, uint64_t seed) {
+                               float inv_keep, uint64_t seed,
+                               const int64_t* __restrict__ seed_base) {
+  seed = graph_seed(seed_base, seed);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     Philox4 r = philox4x32(seed, (uint64_t)i);
@@ -137,6 +143,20 @@ template <> struct DevT<at::Half> { using type = _Float16; };
 // host wrappers
 // ===========================================================================
 
+// hipGraph-replay seed indirection (see common.h graph_seed): Python keeps
+// the 1-element int64 step-counter tensor alive for the lifetime of the
+// captured graph and increments it inside the captured region.
+int64_t* g_seed_base = nullptr;
+
+void set_seed_buffer(torch::Tensor t) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kInt64 &&
+                  t.numel() == 1 && t.is_contiguous(),
+              "set_seed_buffer: need a 1-element contiguous int64 GPU tensor");
+  g_seed_base = t.data_ptr<int64_t>();
+}
+
+void clear_seed_buffer() { g_seed_base = nullptr; }
+
 torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value,
                              double max_value, double stochastic, int64_t seed) {
   TORCH_CHECK(x.is_cuda(), "fake_quant_fwd: expected GPU tensor");
@@ -154,13 +174,13 @@ torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value
                          dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
                          (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
                          (float)min_value, 1.0f / scale, scale, qmax,
-                         (float)stochastic, (uint64_t)seed);
+                         (float)stochastic, (uint64_t)seed, g_seed_base);
     } else {
       hipLaunchKernelGGL((fake_quant_kernel<T, false>), dim3(grid_1d((n + 3) / 4)),
                          dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
                          (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
                          (float)min_value, 1.0f / scale, scale, qmax, 0.0f,
-                         (uint64_t)seed);
+                         (uint64_t)seed, g_seed_base);
     }
   });
   HIP_CHECK_LAST();
@@ -195,7 +215,7 @@ torch::Tensor mult_uniform_noise(torch::Tensor x, double a, int64_t seed) {
     hipLaunchKernelGGL((mult_uniform_kernel<T>), dim3(grid_1d(n)), dim3(kBlock),
                        0, c10::hip::getCurrentHIPStream(),
                        (const T*)xc.data_ptr(), (T*)out.data_ptr(), n, (float)a,
-                       (uint64_t)seed);
+                       (uint64_t)seed, g_seed_base);
   });
   HIP_CHECK_LAST();
   return out;
@@ -228,7 +248,7 @@ std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed) 
     hipLaunchKernelGGL((dropout_kernel<T>), dim3(grid_1d(n)), dim3(kBlock), 0,
                        c10::hip::getCurrentHIPStream(), (const T*)xc.data_ptr(),
                        (T*)out.data_ptr(), (T*)mask.data_ptr(), n, (float)p,
-                       inv_keep, (uint64_t)seed);
+                       inv_keep, (uint64_t)seed, g_seed_base);
   });
   HIP_CHECK_LAST();
   return {out, mask};

@@ -76,8 +76,17 @@ class QuantMeasure(nn.Module):
         self.calculate_running = calculate_running
         self.running_list = []
         self.pctl = pctl
+        # calibrated range cached as Python floats: the hot path must not
+        # read device scalars (host sync; also illegal under hipGraph
+        # capture). Invalidated when buffers change (finish_calibration,
+        # checkpoint load).
+        self._range_cache = None
         if pctl < 1:
             raise ValueError('pctl is {} please check'.format(pctl))
+
+    def _load_from_state_dict(self, *args, **kwargs):
+        self._range_cache = None
+        return super()._load_from_state_dict(*args, **kwargs)
 
     def forward(self, input):
         with torch.no_grad():
@@ -104,7 +113,10 @@ class QuantMeasure(nn.Module):
                         pctl_val = ops.kth_percentile(input, self.pctl)
                     max_value = input.max().item()
                     self.running_list.append(pctl_val)
+            elif self._range_cache is not None:
+                min_value, max_value = self._range_cache
             else:
+                cacheable = True
                 if self.min_value < 0 and float(self.running_min.min()) < 0:
                     min_value = float(self.running_min.item())
                     max_value = float(self.running_max.item())
@@ -113,7 +125,10 @@ class QuantMeasure(nn.Module):
                 elif float(self.running_max) > 0:
                     max_value = float(self.running_max)
                 else:
-                    max_value = input.max().item()
+                    max_value = input.max().item()  # data-dependent
+                    cacheable = False
+                if cacheable:
+                    self._range_cache = (min_value, max_value)
 
             stoch = self.stochastic if self.training else 0
 
@@ -129,6 +144,7 @@ def finish_calibration(model, device=None):
     with torch.no_grad():
         for m in model.modules():
             if isinstance(m, QuantMeasure):
+                m._range_cache = None
                 if m.calculate_running and m.running_list:
                     m.calculate_running = False
                     vals = torch.stack([torch.as_tensor(v, dtype=torch.float32)

@@ -27,11 +27,12 @@ def test_rccl_init_world_size_1():
     dist.init_process_group(backend="nccl", init_method="env://",
                             rank=0, world_size=1)
     try:
-        t = torch.arange(1024, dtype=torch.bfloat16, device="cuda")
+        # values exactly representable in bf16 (ws=1 all_reduce == identity)
+        ref = torch.arange(1024, dtype=torch.float32).to(torch.bfloat16)
+        t = ref.clone().to("cuda")
         dist.all_reduce(t)
         torch.cuda.synchronize()
-        assert torch.equal(t.float().cpu(),
-                           torch.arange(1024, dtype=torch.float32))
+        assert torch.equal(t.cpu(), ref)
         dist.barrier()
     finally:
         dist.destroy_process_group()

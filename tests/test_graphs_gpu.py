@@ -1,0 +1,120 @@
+"""hipGraph-captured training step: numerics vs eager, RNG across replays.
+
+The captured step must run the SAME kernels in the SAME order as eager
+launches, so a deterministic config (no noise, no stochastic rounding, no
+dropout) must produce identical parameters; and with RNG on, each replay
+must draw FRESH noise via the device step counter (csrc/common.h
+graph_seed) rather than replaying the capture-time noise.
+"""
+
+import pytest
+import torch
+
+from noisynet_amd import ops, utils
+from noisynet_amd import optim as native_optim
+from noisynet_amd.config import broadcast_per_layer, build_noisynet_parser
+from noisynet_amd.graphs import GraphedTrainStep
+from noisynet_amd.models.noisynet import Net
+from noisynet_amd.quant import finish_calibration, start_calibration
+
+pytestmark = pytest.mark.gpu
+
+
+def _build(argv, seed=3):
+    args = build_noisynet_parser().parse_args(argv)
+    broadcast_per_layer(args)
+    torch.manual_seed(seed)
+    model = Net(args)
+    utils.init_model(model, args)
+    model = model.cuda().to(memory_format=torch.channels_last)
+    opt = native_optim.SGD(model.parameters(), lr=0.01, momentum=0.9,
+                           nesterov=True)
+    return model, opt
+
+
+def _calibrate(model, x):
+    start_calibration(model)
+    with torch.no_grad():
+        for i in range(6):
+            model(x, 0, i)
+    finish_calibration(model, torch.device('cuda'))
+
+
+def test_graphed_step_matches_eager_deterministic():
+    argv = ['--q_a', '4', '--act_max', '5', '--batch_size', '64',
+            '--stochastic', '0', '--calculate_running', '--no-augment']
+    torch.manual_seed(0)
+    x = torch.rand(64, 3, 32, 32).cuda().contiguous(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (64,)).cuda()
+    steps = 6
+
+    # eager
+    model_e, opt_e = _build(argv)
+    _calibrate(model_e, x)
+    model_e.train()
+    for i in range(steps):
+        loss = ops.cross_entropy(model_e(x, 0, 1000 + i), y)
+        opt_e.zero_grad(set_to_none=False)
+        loss.backward()
+        opt_e.step()
+
+    # graphed (same seed -> same init)
+    model_g, opt_g = _build(argv)
+    _calibrate(model_g, x)
+    model_g.train()
+    static_x = x.clone()
+    static_y = y.clone()
+
+    def body():
+        loss = ops.cross_entropy(model_g(static_x, 0, 1000), static_y)
+        opt_g.zero_grad(set_to_none=False)
+        loss.backward()
+        opt_g.step()
+        return loss
+
+    # warmup executes the step once (materializes grad/momentum buffers so
+    # zero_grad is captured as a real zeroing); capture itself executes
+    # nothing; so warmup(1) + (steps-1) replays == steps eager updates
+    gstep = GraphedTrainStep(body, warmup=1)
+    for _ in range(steps - 1):
+        gstep.replay()
+    torch.cuda.synchronize()
+
+    for (ne, pe), (ng, pg) in zip(model_e.named_parameters(),
+                                  model_g.named_parameters()):
+        assert ne == ng
+        assert torch.allclose(pe.float(), pg.float(), atol=1e-6), \
+            (ne, (pe - pg).abs().max().item())
+    gstep.close()
+
+
+def test_graphed_replays_draw_fresh_noise():
+    argv = ['--current', '1', '--q_a', '4', '--act_max', '5', '--batch_size',
+            '64', '--calculate_running', '--no-augment']
+    model, opt = _build(argv, seed=5)
+    torch.manual_seed(1)
+    x = torch.rand(64, 3, 32, 32).cuda().contiguous(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (64,)).cuda()
+    _calibrate(model, x)
+    model.train()
+
+    losses = []
+
+    def body():
+        loss = ops.cross_entropy(model(x, 0, 1000), y)
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        # no optimizer step: identical weights each replay, so any loss
+        # difference comes from the noise draw alone
+        return loss
+
+    gstep = GraphedTrainStep(body, warmup=1)
+    for _ in range(4):
+        losses.append(float(gstep.replay()))
+    gstep.close()
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+    # same weights + same data -> differences are pure RNG; frozen seeds
+    # would make every replay identical
+    assert len(set(losses)) > 1, losses
