@@ -28,6 +28,8 @@ std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
                                             torch::Tensor target);
 torch::Tensor softmax_xent_bwd(torch::Tensor softmax, torch::Tensor target,
                                double gscale);
+torch::Tensor softmax_xent_bwd_t(torch::Tensor softmax, torch::Tensor target,
+                                 torch::Tensor gscale);
 
 // bn_act.hip
 std::vector<torch::Tensor> bn_stats(torch::Tensor x);
@@ -136,6 +138,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("softmax_xent_bwd_t", &softmax_xent_bwd_t);
   m.def("bn_stats", &bn_stats);
   m.def("bn_stats_finalize", &bn_stats_finalize);
   m.def("bn_act_fwd", &bn_act_fwd);

@@ -55,7 +55,11 @@ template <typename T>
 __global__ void softmax_xent_bwd_kernel(const T* __restrict__ softmax,
                                         const int64_t* __restrict__ target,
                                         T* __restrict__ grad, int64_t nrows,
-                                        int ncols, float scale) {
+                                        int ncols, float scale,
+                                        const float* __restrict__ scale_p
+                                        = nullptr) {
+  // device-resident upstream grad scale (graph capture / no host sync)
+  if (scale_p != nullptr) scale *= scale_p[0];
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        i < nrows * (int64_t)ncols; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t row = i / ncols;
@@ -108,6 +112,32 @@ torch::Tensor softmax_xent_bwd(torch::Tensor softmax, torch::Tensor target,
                        c10::hip::getCurrentHIPStream(),
                        (const T*)softmax.data_ptr(), tgt.data_ptr<int64_t>(),
                        (T*)grad.data_ptr(), nrows, ncols, scale);
+  });
+  HIP_CHECK_LAST();
+  return grad;
+}
+
+// gscale as a device-resident f32 scalar: no host read of the upstream
+// grad (required under hipGraph capture; also drops a per-step sync)
+torch::Tensor softmax_xent_bwd_t(torch::Tensor softmax, torch::Tensor target,
+                                 torch::Tensor gscale) {
+  TORCH_CHECK(gscale.is_cuda() && gscale.scalar_type() == torch::kFloat32 &&
+                  gscale.numel() == 1,
+              "softmax_xent_bwd_t: gscale must be a 1-element f32 GPU tensor");
+  int64_t nrows = softmax.size(0);
+  int ncols = (int)softmax.size(1);
+  auto grad = torch::empty_like(softmax);
+  auto tgt = target.contiguous();
+  int64_t n = nrows * ncols;
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 8192);
+  float inv_rows = (float)(1.0 / (double)nrows);
+  NN_DISPATCH(softmax.scalar_type(), "softmax_xent_bwd_t", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((softmax_xent_bwd_kernel<T>), dim3(blocks), dim3(256), 0,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)softmax.data_ptr(), tgt.data_ptr<int64_t>(),
+                       (T*)grad.data_ptr(), nrows, ncols, inv_rows,
+                       gscale.data_ptr<float>());
   });
   HIP_CHECK_LAST();
   return grad;

@@ -990,7 +990,12 @@ class SoftmaxXent(torch.autograd.Function):
         softmax, target = ctx.saved_tensors
         n, c = softmax.shape
         if use_native(softmax):
-            gx = ext().softmax_xent_bwd(softmax, target, float(g))
+            if isinstance(g, torch.Tensor) and g.is_cuda:
+                # device-resident scale: no host sync (graph-capturable)
+                gx = ext().softmax_xent_bwd_t(
+                    softmax, target, g.detach().to(torch.float32).reshape(1))
+            else:
+                gx = ext().softmax_xent_bwd(softmax, target, float(g))
         else:
             onehot = F.one_hot(target, c).to(softmax.dtype)
             gx = (softmax - onehot) * (g / n)
