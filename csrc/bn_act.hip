@@ -17,26 +17,27 @@ namespace {
 constexpr int kBlock = 256;
 
 // Grid: (C + 63)/64 blocks in x, rows-chunks in y. Each block handles 64
-// channels x kBlock/64-row slab, accumulating into global atomics (f32).
+// channels x kBlock/64-row slab. DETERMINISTIC: each block writes its
+// per-channel partial to partial_sum[blockIdx.y * C + c] (no atomics; the
+// intra-block combine iterates row-groups in a fixed order), and a second
+// fixed-order pass reduces over blockIdx.y -- same-seed runs are
+// bit-identical, which run-to-run atomic ordering was not.
 template <typename T>
-__global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
-                                float* __restrict__ sumsq, int64_t rows, int C) {
+__global__ void bn_stats_kernel(const T* __restrict__ x,
+                                float* __restrict__ partial_sum,
+                                float* __restrict__ partial_sumsq,
+                                int64_t rows, int C) {
   // channel/row thread split: each block owns a <=64-channel span and
   // packs kBlock/span row groups, so a partial span (C=65's lone tail
   // channel, MNv2's 16-32-channel stems) still uses every lane instead
   // of leaving one active lane per 64 as the critical path.
-  __shared__ float ls[64], lsq[64];
+  __shared__ float ls[kBlock], lsq[kBlock];
   int cbase = blockIdx.x * 64;
   int cw = C - cbase;
   if (cw > 64) cw = 64;
   int c_l = (cw == 64) ? (threadIdx.x & 63) : (threadIdx.x % cw);
   int rgrp = (cw == 64) ? (threadIdx.x >> 6) : (threadIdx.x / cw);
   int ngrp = kBlock / cw;
-  if (threadIdx.x < 64) {
-    ls[threadIdx.x] = 0.0f;
-    lsq[threadIdx.x] = 0.0f;
-  }
-  __syncthreads();
   float acc_s = 0.0f, acc_q = 0.0f;
   if (rgrp < ngrp) {
     int c = cbase + c_l;
@@ -63,17 +64,38 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
     acc_s = (s[0] + s[1]) + (s[2] + s[3]);
     acc_q = (sq[0] + sq[1]) + (sq[2] + sq[3]);
   }
-  // LDS pre-reduce across the block's row groups: one global atomic per
-  // (block, channel) instead of one per thread
-  if (rgrp < ngrp) {
-    atomicAdd(&ls[c_l], acc_s);
-    atomicAdd(&lsq[c_l], acc_q);
-  }
+  // slotted LDS (thread tid owns slot tid = rgrp*cw + c_l), combined in
+  // FIXED row-group order by the channel threads
+  ls[threadIdx.x] = acc_s;
+  lsq[threadIdx.x] = acc_q;
   __syncthreads();
   if (threadIdx.x < cw) {
-    atomicAdd(&sum[cbase + threadIdx.x], ls[threadIdx.x]);
-    atomicAdd(&sumsq[cbase + threadIdx.x], lsq[threadIdx.x]);
+    float s = 0.0f, q = 0.0f;
+    for (int gp = 0; gp < ngrp; ++gp) {
+      s += ls[gp * cw + threadIdx.x];
+      q += lsq[gp * cw + threadIdx.x];
+    }
+    int64_t o = (int64_t)blockIdx.y * C + cbase + threadIdx.x;
+    partial_sum[o] = s;
+    partial_sumsq[o] = q;
   }
+}
+
+// fixed-order reduction of two [gy, C] partial arrays to [C]
+__global__ void bn_partials_reduce_kernel(const float* __restrict__ pa,
+                                          const float* __restrict__ pb,
+                                          float* __restrict__ outa,
+                                          float* __restrict__ outb, int gy,
+                                          int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.0f, b = 0.0f;
+  for (int r = 0; r < gy; ++r) {
+    a += pa[(int64_t)r * C + c];
+    b += pb[(int64_t)r * C + c];
+  }
+  outa[c] = a;
+  outb[c] = b;
 }
 
 template <typename T>
@@ -99,18 +121,25 @@ template <> struct DevT<at::Half> { using type = _Float16; };
 
 // one tiny kernel replacing the eager mean/var/invstd/running-update chain
 // (~8 launches per BN layer per step)
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
+__global__ void bn_finalize_kernel(const float* __restrict__ partial_sum,
+                                   const float* __restrict__ partial_sumsq,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
-                                   float* __restrict__ running_var, int C,
-                                   float inv_n, float unbias, float momentum,
-                                   float eps, int has_running) {
+                                   float* __restrict__ running_var, int gy,
+                                   int C, float inv_n, float unbias,
+                                   float momentum, float eps,
+                                   int has_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = sum[c] * inv_n;
-  float v = fmaxf(sumsq[c] * inv_n - m * m, 0.0f);
+  // fixed-order partials reduce folded in (determinism, see bn_stats_kernel)
+  float s = 0.0f, q = 0.0f;
+  for (int r = 0; r < gy; ++r) {
+    s += partial_sum[(int64_t)r * C + c];
+    q += partial_sumsq[(int64_t)r * C + c];
+  }
+  float m = s * inv_n;
+  float v = fmaxf(q * inv_n - m * m, 0.0f);
   mean[c] = m;
   invstd[c] = rsqrtf(v + eps);
   if (has_running) {
@@ -139,20 +168,27 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
     rows = x.size(0);
   }
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum = torch::zeros({C}, opts);
-  auto sumsq = torch::zeros({C}, opts);
   int gx = (C + 63) / 64;
   int ngrp = kBlock / std::min(C, 64);  // first block's packing
-  // cap so atomic depth per channel (gy * ngrp) stays ~4k
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
+  auto partial_sum = torch::empty({gy, C}, opts);
+  auto partial_sumsq = torch::empty({gy, C}, opts);
+  auto sum = torch::empty({C}, opts);
+  auto sumsq = torch::empty({C}, opts);
+  auto stream = c10::hip::getCurrentHIPStream();
   NN_DISPATCH(x.scalar_type(),
                                   "bn_stats", [&] {
     using T = typename DevT<scalar_t>::type;
     hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
-                       c10::hip::getCurrentHIPStream(), (const T*)ptr,
-                       sum.data_ptr<float>(), sumsq.data_ptr<float>(), rows, C);
+                       stream, (const T*)ptr, partial_sum.data_ptr<float>(),
+                       partial_sumsq.data_ptr<float>(), rows, C);
   });
+  hipLaunchKernelGGL(bn_partials_reduce_kernel,
+                     dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0, stream,
+                     partial_sum.data_ptr<float>(),
+                     partial_sumsq.data_ptr<float>(), sum.data_ptr<float>(),
+                     sumsq.data_ptr<float>(), gy, C);
   HIP_CHECK_LAST();
   auto mean = sum / (double)rows;
   auto var = sumsq / (double)rows - mean * mean;
@@ -200,20 +236,15 @@ template <typename T>
 __global__ void bn_act_bwd_reduce_kernel(
     const T* __restrict__ g, const T* __restrict__ x, const T* __restrict__ y,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ sum_g, float* __restrict__ sum_gx, int64_t rows, int C,
-    int do_relu, float act_max) {
-  __shared__ float ls[64], lsq[64];  // see bn_stats_kernel
+    float* __restrict__ partial_g, float* __restrict__ partial_gx,
+    int64_t rows, int C, int do_relu, float act_max) {
+  __shared__ float ls[kBlock], lsq[kBlock];  // see bn_stats_kernel
   int cbase = blockIdx.x * 64;
   int cw = C - cbase;
   if (cw > 64) cw = 64;
   int c_l = (cw == 64) ? (threadIdx.x & 63) : (threadIdx.x % cw);
   int rgrp = (cw == 64) ? (threadIdx.x >> 6) : (threadIdx.x / cw);
   int ngrp = kBlock / cw;
-  if (threadIdx.x < 64) {
-    ls[threadIdx.x] = 0.0f;
-    lsq[threadIdx.x] = 0.0f;
-  }
-  __syncthreads();
   float acc_g = 0.0f, acc_gx = 0.0f;
   if (rgrp < ngrp) {
     int c = cbase + c_l;
@@ -241,14 +272,19 @@ __global__ void bn_act_bwd_reduce_kernel(
     acc_g = (s_g[0] + s_g[1]) + (s_g[2] + s_g[3]);
     acc_gx = (s_gx[0] + s_gx[1]) + (s_gx[2] + s_gx[3]);
   }
-  if (rgrp < ngrp) {
-    atomicAdd(&ls[c_l], acc_g);
-    atomicAdd(&lsq[c_l], acc_gx);
-  }
+  // slotted LDS + fixed-order combine (see bn_stats_kernel)
+  ls[threadIdx.x] = acc_g;
+  lsq[threadIdx.x] = acc_gx;
   __syncthreads();
   if (threadIdx.x < cw) {
-    atomicAdd(&sum_g[cbase + threadIdx.x], ls[threadIdx.x]);
-    atomicAdd(&sum_gx[cbase + threadIdx.x], lsq[threadIdx.x]);
+    float sg = 0.0f, sgx = 0.0f;
+    for (int gp = 0; gp < ngrp; ++gp) {
+      sg += ls[gp * cw + threadIdx.x];
+      sgx += lsq[gp * cw + threadIdx.x];
+    }
+    int64_t o = (int64_t)blockIdx.y * C + cbase + threadIdx.x;
+    partial_g[o] = sg;
+    partial_gx[o] = sgx;
   }
 }
 
@@ -298,14 +334,16 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
     rows = x.size(0);
   }
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum_g = torch::zeros({C}, opts);
-  auto sum_gx = torch::zeros({C}, opts);
+  auto sum_g = torch::empty({C}, opts);
+  auto sum_gx = torch::empty({C}, opts);
   auto gx = torch::empty_like(g);
   auto stream = c10::hip::getCurrentHIPStream();
   int gx_blocks = (C + 63) / 64;
   int ngrp = kBlock / std::min(C, 64);  // first block's packing
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
+  auto partial_g = torch::empty({gy, C}, opts);
+  auto partial_gx = torch::empty({gy, C}, opts);
   int64_t n = x.numel();
   int eblocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd", [&] {
@@ -314,8 +352,14 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
                        dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
                        (const T*)x.data_ptr(), (const T*)y.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(),
+                       partial_g.data_ptr<float>(),
+                       partial_gx.data_ptr<float>(),
                        rows, C, relu ? 1 : 0, (float)act_max);
+    hipLaunchKernelGGL(bn_partials_reduce_kernel,
+                       dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0,
+                       stream, partial_g.data_ptr<float>(),
+                       partial_gx.data_ptr<float>(), sum_g.data_ptr<float>(),
+                       sum_gx.data_ptr<float>(), gy, C);
     if (training) {
       hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
                          dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
@@ -355,22 +399,29 @@ std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
     rows = x.size(0);
   }
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum_g = torch::zeros({C}, opts);
-  auto sum_gx = torch::zeros({C}, opts);
+  auto sum_g = torch::empty({C}, opts);
+  auto sum_gx = torch::empty({C}, opts);
   int gx_blocks = (C + 63) / 64;
   int ngrp = kBlock / std::min(C, 64);  // first block's packing
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
+  auto partial_g = torch::empty({gy, C}, opts);
+  auto partial_gx = torch::empty({gy, C}, opts);
+  auto stream = c10::hip::getCurrentHIPStream();
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd_reduce", [&] {
     using T = typename DevT<scalar_t>::type;
     hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
-                       dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                       dim3(kBlock), 0, stream,
                        (const T*)g.data_ptr(), (const T*)x.data_ptr(),
                        (const T*)y.data_ptr(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), sum_g.data_ptr<float>(),
-                       sum_gx.data_ptr<float>(), rows, C, relu ? 1 : 0,
+                       invstd.data_ptr<float>(), partial_g.data_ptr<float>(),
+                       partial_gx.data_ptr<float>(), rows, C, relu ? 1 : 0,
                        (float)act_max);
   });
+  hipLaunchKernelGGL(bn_partials_reduce_kernel,
+                     dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0, stream,
+                     partial_g.data_ptr<float>(), partial_gx.data_ptr<float>(),
+                     sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(), gy, C);
   HIP_CHECK_LAST();
   return {sum_g, sum_gx};
 }
@@ -437,29 +488,30 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
                                running_var.scalar_type() == torch::kFloat32),
               "bn_stats_finalize: running stats must be f32");
   auto opts = x.options().dtype(torch::kFloat32);
-  auto sum = torch::zeros({C}, opts);
-  auto sumsq = torch::zeros({C}, opts);
   int gx = (C + 63) / 64;
   int ngrp = kBlock / std::min(C, 64);  // first block's packing
   int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
                                   std::max(1, 4096 / ngrp));
+  auto partial_sum = torch::empty({gy, C}, opts);
+  auto partial_sumsq = torch::empty({gy, C}, opts);
   auto stream = c10::hip::getCurrentHIPStream();
   NN_DISPATCH(x.scalar_type(), "bn_stats_finalize", [&] {
     using T = typename DevT<scalar_t>::type;
     hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
-                       stream, (const T*)ptr, sum.data_ptr<float>(),
-                       sumsq.data_ptr<float>(), rows, C);
+                       stream, (const T*)ptr, partial_sum.data_ptr<float>(),
+                       partial_sumsq.data_ptr<float>(), rows, C);
   });
   float n = (float)rows;
   float unbias = n / std::max(n - 1.0f, 1.0f);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
   hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
-                     stream, sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                     stream, partial_sum.data_ptr<float>(),
+                     partial_sumsq.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      has_running ? running_mean.data_ptr<float>() : nullptr,
-                     has_running ? running_var.data_ptr<float>() : nullptr, C,
-                     1.0f / n, unbias, (float)momentum, (float)eps,
+                     has_running ? running_var.data_ptr<float>() : nullptr, gy,
+                     C, 1.0f / n, unbias, (float)momentum, (float)eps,
                      has_running ? 1 : 0);
   HIP_CHECK_LAST();
   return {mean, invstd};

@@ -834,8 +834,8 @@ DEV_INLINE void stage_gx_transposed(char* g_lds, char* x_lds,
 template <typename T>
 __global__ __launch_bounds__(kBlock)
 void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
-                       float* __restrict__ dw /* [K,R,S,C] f32 */, ConvGeom g,
-                       int mchunks_per_block) {
+                       float* __restrict__ dw /* [mslices][K,R,S,C] f32 */,
+                       ConvGeom g, int mchunks_per_block, int64_t wspan) {
   // grid: x = c-tiles (or flat rsc-tiles), y = k-tiles, z = taps * m-slices
   int taps = g.flat ? 1 : g.R * g.S;
   int tap = blockIdx.z % taps;
@@ -886,7 +886,10 @@ void conv_wgrad_kernel(const T* __restrict__ gy, const T* __restrict__ x,
           int64_t off = g.flat
               ? (int64_t)k * span + c
               : (((int64_t)k * g.R + r) * g.S + s) * g.C + c;
-          atomicAdd(&dw[off], acc[fm][fn][reg]);
+          // one writer per (mslice, k, tap, c): plain store into this
+          // m-slice's partial image; the host sums partials in fixed
+          // order (deterministic, no atomic race ordering)
+          dw[(int64_t)mslice * wspan + off] = acc[fm][fn][reg];
         }
       }
 }
@@ -1269,8 +1272,6 @@ torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   auto g = make_geom((int)x.size(0), (int)x.size(2), (int)x.size(3),
                      (int)x.size(1), (int)gy.size(1), (int)R, (int)S,
                      (int)stride, (int)pad);
-  auto dw_f = torch::zeros({g.K, g.R, g.S, g.C},
-                           x.options().dtype(torch::kFloat32));
   // slice the contraction so ~2048 blocks are in flight
   int taps = g.flat ? 1 : g.R * g.S;
   int span = g.flat ? g.R * g.S * g.C : g.C;
@@ -1281,6 +1282,11 @@ torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   int mslices = (int)std::min<int64_t>(
       mtotal, std::max<int64_t>(1, target_z / taps));
   int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
+  // per-mslice partial weight images, summed by torch in fixed order
+  // (deterministic; the atomicAdd variant gave run-to-run noise)
+  auto dw_f = torch::zeros({mslices, g.K, g.R, g.S, g.C},
+                           x.options().dtype(torch::kFloat32));
+  int64_t wspan = (int64_t)g.K * g.R * g.S * g.C;
   dim3 grid(ctiles, ktiles, taps * mslices);
   size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(gy.scalar_type(),
@@ -1290,11 +1296,11 @@ torch::Tensor conv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
     hipLaunchKernelGGL((conv_wgrad_kernel<T>), grid, dim3(kBlock), lds,
                        c10::hip::getCurrentHIPStream(),
                        (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
-                       dw_f.data_ptr<float>(), g, chunks_per_block);
+                       dw_f.data_ptr<float>(), g, chunks_per_block, wspan);
   });
   HIP_CHECK_LAST();
   // dw as NCHW-logical [K, C, R, S] channels_last == raw [K,R,S,C]
-  auto dw = dw_f.to(x.scalar_type());
+  auto dw = dw_f.sum(0).to(x.scalar_type());
   return dw.permute({0, 3, 1, 2}).contiguous(at::MemoryFormat::ChannelsLast);
 }
 
@@ -1344,7 +1350,6 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
   TORCH_CHECK(gy.dim() == 2 && gy.is_contiguous());
   TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
   auto g = linear_geom(x, gy.size(1));
-  auto dw_f = torch::zeros({g.K, g.C}, x.options().dtype(torch::kFloat32));
   if ((g.K & 7) == 0 && (g.C & 7) == 0 && gy.element_size() == 2) {
     // aligned 16-bit: 128-deep rounds. Wide outputs use the 128x128-tile
     // variant (fewer cross-tile operand re-reads); narrow ones the
@@ -1390,6 +1395,10 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
   if (mslices < 1) mslices = 1;
   int chunks_per_block = (int)((mtotal + mslices - 1) / mslices);
   dim3 grid(ctiles, ktiles, mslices);
+  // per-mslice partials, deterministic fixed-order sum (see conv_wgrad)
+  auto parts_f = torch::zeros({mslices, g.K, g.C},
+                              x.options().dtype(torch::kFloat32));
+  int64_t wspan = (int64_t)g.K * g.C;
   size_t lds = 0;  // per-dtype, set inside the dispatch
   NN_DISPATCH(gy.scalar_type(),
                                   "linear_wgrad", [&] {
@@ -1398,10 +1407,10 @@ torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x) {
     hipLaunchKernelGGL((conv_wgrad_kernel<T>), grid, dim3(kBlock), lds,
                        c10::hip::getCurrentHIPStream(),
                        (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
-                       dw_f.data_ptr<float>(), g, chunks_per_block);
+                       parts_f.data_ptr<float>(), g, chunks_per_block, wspan);
   });
   HIP_CHECK_LAST();
-  return dw_f.to(x.scalar_type());
+  return parts_f.sum(0).to(x.scalar_type());
 }
 
 namespace {

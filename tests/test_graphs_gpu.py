@@ -118,3 +118,30 @@ def test_graphed_replays_draw_fresh_noise():
     # same weights + same data -> differences are pure RNG; frozen seeds
     # would make every replay identical
     assert len(set(losses)) > 1, losses
+
+
+def test_full_step_run_to_run_determinism():
+    """Two eager runs with identical seeds are BIT-identical: BN statistics
+    and wgrad use fixed-order partial reductions, not atomics (run-to-run
+    atomic ordering made trajectories diverge ~1e-2 within 6 steps)."""
+    argv = ['--q_a', '4', '--act_max', '5', '--batch_size', '64',
+            '--stochastic', '0', '--calculate_running', '--no-augment']
+    torch.manual_seed(0)
+    x = torch.rand(64, 3, 32, 32).cuda().contiguous(
+        memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (64,)).cuda()
+
+    def run():
+        model, opt = _build(argv)
+        _calibrate(model, x)
+        model.train()
+        for i in range(6):
+            loss = ops.cross_entropy(model(x, 0, 1000 + i), y)
+            opt.zero_grad(set_to_none=False)
+            loss.backward()
+            opt.step()
+        return {n: p.detach().clone() for n, p in model.named_parameters()}
+
+    p1, p2 = run(), run()
+    for n in p1:
+        assert torch.equal(p1[n], p2[n]), n
