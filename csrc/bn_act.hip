@@ -16,6 +16,9 @@ namespace {
 
 constexpr int kBlock = 256;
 
+template <typename T>
+struct alignas(16) BnPack8 { T v[8]; };
+
 // Grid: (C + 63)/64 blocks in x, rows-chunks in y. Each block handles 64
 // channels x kBlock/64-row slab. DETERMINISTIC: each block writes its
 // per-channel partial to partial_sum[blockIdx.y * C + c] (no atomics; the
@@ -81,6 +84,75 @@ __global__ void bn_stats_kernel(const T* __restrict__ x,
   }
 }
 
+// 8-channel-vector stats form (16-bit dtype, C % 8 == 0): 16-B row loads.
+// Same deterministic partials contract as bn_stats_kernel.
+template <typename T>
+__global__ void bn_stats_vec_kernel(const T* __restrict__ x,
+                                    float* __restrict__ partial_sum,
+                                    float* __restrict__ partial_sumsq,
+                                    int64_t rows, int C) {
+  using V = BnPack8<T>;
+  const int Cv = C >> 3;
+  __shared__ float ls[kBlock * 8];
+  __shared__ float lsq[kBlock * 8];
+  int cvspan = Cv < 64 ? Cv : 64;
+  int cbase_v = blockIdx.x * 64;
+  int cw = Cv - cbase_v;
+  if (cw > cvspan) cw = cvspan;
+  int c_l = threadIdx.x % cw;
+  int rgrp = threadIdx.x / cw;
+  int ngrp = kBlock / cw;
+  float acc_s[8] = {}, acc_q[8] = {};
+  if (rgrp < ngrp) {
+    int cv = cbase_v + c_l;
+    int rstart = blockIdx.y * ngrp + rgrp;
+    const int64_t st = (int64_t)gridDim.y * ngrp * 2;
+    int64_t r = (int64_t)rstart * 2;
+    for (; r + 1 < rows; r += st) {
+      V a = ((const V*)x)[r * Cv + cv];
+      V b = ((const V*)x)[(r + 1) * Cv + cv];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        float va = to_f32(a.v[u]), vb = to_f32(b.v[u]);
+        acc_s[u] += va + vb;
+        acc_q[u] += va * va + vb * vb;
+      }
+    }
+    for (; r < rows; ++r) {
+      V a = ((const V*)x)[r * Cv + cv];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        float va = to_f32(a.v[u]);
+        acc_s[u] += va;
+        acc_q[u] += va * va;
+      }
+    }
+  }
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    ls[threadIdx.x * 8 + u] = acc_s[u];
+    lsq[threadIdx.x * 8 + u] = acc_q[u];
+  }
+  __syncthreads();
+  if (threadIdx.x < cw) {
+    float s[8] = {}, q[8] = {};
+    for (int gp = 0; gp < ngrp; ++gp) {
+      int slot = (gp * cw + threadIdx.x) * 8;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        s[u] += ls[slot + u];
+        q[u] += lsq[slot + u];
+      }
+    }
+    int64_t o = (int64_t)blockIdx.y * C + (cbase_v + threadIdx.x) * 8;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      partial_sum[o + u] = s[u];
+      partial_sumsq[o + u] = q[u];
+    }
+  }
+}
+
 // fixed-order reduction of two [gy, C] partial arrays to [C]
 __global__ void bn_partials_reduce_kernel(const float* __restrict__ pa,
                                           const float* __restrict__ pb,
@@ -112,6 +184,35 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     if (do_relu) v = fmaxf(v, 0.0f);
     if (act_max > 0.0f) v = fminf(v, act_max);
     y[i] = from_f32<T>(v);
+  }
+}
+
+// 8-wide form (16-bit dtype, C % 8 == 0): 16-B loads/stores, channel
+// params fetched per-vector (the scalar form issued one 2-B load per lane)
+template <typename T>
+__global__ void bn_act_fwd_vec_kernel(const T* __restrict__ x,
+                                      T* __restrict__ y,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd,
+                                      const float* __restrict__ gamma,
+                                      const float* __restrict__ beta,
+                                      int64_t n_vec, int Cv, int do_relu,
+                                      float act_max) {
+  using V = BnPack8<T>;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_vec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % Cv) * 8;
+    V xv = ((const V*)x)[i];
+    V o;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      int c = c0 + u;
+      float v = (to_f32(xv.v[u]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      if (do_relu) v = fmaxf(v, 0.0f);
+      if (act_max > 0.0f) v = fminf(v, act_max);
+      o.v[u] = from_f32<T>(v);
+    }
+    ((V*)y)[i] = o;
   }
 }
 
@@ -168,9 +269,12 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
     rows = x.size(0);
   }
   auto opts = x.options().dtype(torch::kFloat32);
-  int gx = (C + 63) / 64;
-  int ngrp = kBlock / std::min(C, 64);  // first block's packing
-  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+  bool vec16 = x.element_size() == 2 && (C & 7) == 0;
+  int span = vec16 ? std::min(C / 8, 64) : std::min(C, 64);
+  int gx = vec16 ? (C / 8 + 63) / 64 : (C + 63) / 64;
+  int ngrp = kBlock / span;  // first block's packing
+  int rper = vec16 ? 2 : 4;
+  int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
                                   std::max(1, 4096 / ngrp));
   auto partial_sum = torch::empty({gy, C}, opts);
   auto partial_sumsq = torch::empty({gy, C}, opts);
@@ -180,9 +284,15 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
   NN_DISPATCH(x.scalar_type(),
                                   "bn_stats", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
-                       stream, (const T*)ptr, partial_sum.data_ptr<float>(),
-                       partial_sumsq.data_ptr<float>(), rows, C);
+    if (vec16 && sizeof(T) == 2)
+      hipLaunchKernelGGL((bn_stats_vec_kernel<T>), dim3(gx, gy), dim3(kBlock),
+                         0, stream, (const T*)ptr,
+                         partial_sum.data_ptr<float>(),
+                         partial_sumsq.data_ptr<float>(), rows, C);
+    else
+      hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
+                         stream, (const T*)ptr, partial_sum.data_ptr<float>(),
+                         partial_sumsq.data_ptr<float>(), rows, C);
   });
   hipLaunchKernelGGL(bn_partials_reduce_kernel,
                      dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0, stream,
@@ -208,16 +318,28 @@ torch::Tensor bn_act_fwd(torch::Tensor x, torch::Tensor mean,
   }
   auto y = torch::empty_like(x);
   int64_t n = x.numel();
-  int blocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
   NN_DISPATCH(x.scalar_type(),
                                   "bn_act_fwd", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((bn_act_fwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
-                       c10::hip::getCurrentHIPStream(), (const T*)x.data_ptr(),
-                       (T*)y.data_ptr(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                       beta.data_ptr<float>(), n, C, relu ? 1 : 0,
-                       (float)act_max);
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (sizeof(T) == 2 && (C & 7) == 0) {
+      int64_t n_vec = n / 8;
+      int blocks = (int)std::min<int64_t>((n_vec + kBlock - 1) / kBlock, 8192);
+      hipLaunchKernelGGL((bn_act_fwd_vec_kernel<T>), dim3(blocks),
+                         dim3(kBlock), 0, stream, (const T*)x.data_ptr(),
+                         (T*)y.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                         beta.data_ptr<float>(), n_vec, C / 8, relu ? 1 : 0,
+                         (float)act_max);
+    } else {
+      int blocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
+      hipLaunchKernelGGL((bn_act_fwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
+                         stream, (const T*)x.data_ptr(),
+                         (T*)y.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                         beta.data_ptr<float>(), n, C, relu ? 1 : 0,
+                         (float)act_max);
+    }
   });
   HIP_CHECK_LAST();
   return y;
@@ -288,6 +410,78 @@ __global__ void bn_act_bwd_reduce_kernel(
   }
 }
 
+// 8-channel-vector backward reduce (16-bit dtype, C % 8 == 0)
+template <typename T>
+__global__ void bn_act_bwd_reduce_vec_kernel(
+    const T* __restrict__ g, const T* __restrict__ x, const T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ partial_g, float* __restrict__ partial_gx,
+    int64_t rows, int C, int do_relu, float act_max) {
+  using V = BnPack8<T>;
+  const int Cv = C >> 3;
+  __shared__ float ls[kBlock * 8];
+  __shared__ float lsq[kBlock * 8];
+  int cvspan = Cv < 64 ? Cv : 64;
+  int cbase_v = blockIdx.x * 64;
+  int cw = Cv - cbase_v;
+  if (cw > cvspan) cw = cvspan;
+  int c_l = threadIdx.x % cw;
+  int rgrp = threadIdx.x / cw;
+  int ngrp = kBlock / cw;
+  float acc_g[8] = {}, acc_gx[8] = {};
+  if (rgrp < ngrp) {
+    int cv = cbase_v + c_l;
+    int c0 = cv * 8;
+    float m[8], is[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      m[u] = mean[c0 + u];
+      is[u] = invstd[c0 + u];
+    }
+    int rstart = blockIdx.y * ngrp + rgrp;
+    const int64_t st = (int64_t)gridDim.y * ngrp;
+    for (int64_t r = rstart; r < rows; r += st) {
+      int64_t o = r * Cv + cv;
+      V gv8 = ((const V*)g)[o];
+      V xv8 = ((const V*)x)[o];
+      V yv8 = ((const V*)y)[o];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        float yv = to_f32(yv8.v[u]);
+        float mask = 1.0f;
+        if (do_relu && yv <= 0.0f) mask = 0.0f;
+        if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
+        float gv = to_f32(gv8.v[u]) * mask;
+        acc_g[u] += gv;
+        acc_gx[u] += gv * (to_f32(xv8.v[u]) - m[u]) * is[u];
+      }
+    }
+  }
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    ls[threadIdx.x * 8 + u] = acc_g[u];
+    lsq[threadIdx.x * 8 + u] = acc_gx[u];
+  }
+  __syncthreads();
+  if (threadIdx.x < cw) {
+    float s[8] = {}, q[8] = {};
+    for (int gp = 0; gp < ngrp; ++gp) {
+      int slot = (gp * cw + threadIdx.x) * 8;
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        s[u] += ls[slot + u];
+        q[u] += lsq[slot + u];
+      }
+    }
+    int64_t o = (int64_t)blockIdx.y * C + (cbase_v + threadIdx.x) * 8;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      partial_g[o + u] = s[u];
+      partial_gx[o + u] = q[u];
+    }
+  }
+}
+
 template <typename T, bool TRAIN>
 __global__ void bn_act_bwd_apply_kernel(
     const T* __restrict__ g, const T* __restrict__ x, const T* __restrict__ y,
@@ -311,6 +505,43 @@ __global__ void bn_act_bwd_apply_kernel(
       gv = gv * gi;
     }
     gx[i] = from_f32<T>(gv);
+  }
+}
+
+template <typename T, bool TRAIN>
+__global__ void bn_act_bwd_apply_vec_kernel(
+    const T* __restrict__ g, const T* __restrict__ x, const T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ sum_g,
+    const float* __restrict__ sum_gx, T* __restrict__ gx, int64_t n_vec,
+    int Cv, float inv_count, int do_relu, float act_max) {
+  using V = BnPack8<T>;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_vec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c0 = (int)(i % Cv) * 8;
+    V gv8 = ((const V*)g)[i];
+    V yv8 = ((const V*)y)[i];
+    V xv8;
+    if (TRAIN) xv8 = ((const V*)x)[i];
+    V o;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      int c = c0 + u;
+      float yv = to_f32(yv8.v[u]);
+      float mask = 1.0f;
+      if (do_relu && yv <= 0.0f) mask = 0.0f;
+      if (act_max > 0.0f && yv >= act_max) mask = 0.0f;
+      float gv = to_f32(gv8.v[u]) * mask;
+      float gi = gamma[c] * invstd[c];
+      if (TRAIN) {
+        float xhat = (to_f32(xv8.v[u]) - mean[c]) * invstd[c];
+        gv = gi * (gv - sum_g[c] * inv_count - xhat * sum_gx[c] * inv_count);
+      } else {
+        gv = gv * gi;
+      }
+      o.v[u] = from_f32<T>(gv);
+    }
+    ((V*)gx)[i] = o;
   }
 }
 
@@ -338,9 +569,12 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
   auto sum_gx = torch::empty({C}, opts);
   auto gx = torch::empty_like(g);
   auto stream = c10::hip::getCurrentHIPStream();
-  int gx_blocks = (C + 63) / 64;
-  int ngrp = kBlock / std::min(C, 64);  // first block's packing
-  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+  bool vec16 = x.element_size() == 2 && (C & 7) == 0;
+  int span = vec16 ? std::min(C / 8, 64) : std::min(C, 64);
+  int gx_blocks = vec16 ? (C / 8 + 63) / 64 : (C + 63) / 64;
+  int ngrp = kBlock / span;  // first block's packing
+  int rper = vec16 ? 1 : 4;
+  int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
                                   std::max(1, 4096 / ngrp));
   auto partial_g = torch::empty({gy, C}, opts);
   auto partial_gx = torch::empty({gy, C}, opts);
@@ -348,34 +582,66 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
   int eblocks = (int)std::min<int64_t>((n + kBlock - 1) / kBlock, 8192);
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
-                       dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
-                       (const T*)x.data_ptr(), (const T*)y.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       partial_g.data_ptr<float>(),
-                       partial_gx.data_ptr<float>(),
-                       rows, C, relu ? 1 : 0, (float)act_max);
+    if (vec16 && sizeof(T) == 2)
+      hipLaunchKernelGGL((bn_act_bwd_reduce_vec_kernel<T>),
+                         dim3(gx_blocks, gy), dim3(kBlock), 0, stream,
+                         (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                         (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         partial_g.data_ptr<float>(),
+                         partial_gx.data_ptr<float>(),
+                         rows, C, relu ? 1 : 0, (float)act_max);
+    else
+      hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
+                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         partial_g.data_ptr<float>(),
+                         partial_gx.data_ptr<float>(),
+                         rows, C, relu ? 1 : 0, (float)act_max);
     hipLaunchKernelGGL(bn_partials_reduce_kernel,
                        dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0,
                        stream, partial_g.data_ptr<float>(),
                        partial_gx.data_ptr<float>(), sum_g.data_ptr<float>(),
                        sum_gx.data_ptr<float>(), gy, C);
+    bool vec = sizeof(T) == 2 && (C & 7) == 0;
+    int vblocks = (int)std::min<int64_t>((n / 8 + kBlock - 1) / kBlock, 8192);
     if (training) {
-      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
-                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
-                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
-                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
-                         1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
+      if (vec)
+        hipLaunchKernelGGL((bn_act_bwd_apply_vec_kernel<T, true>),
+                           dim3(vblocks), dim3(kBlock), 0, stream,
+                           (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                           (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                           sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(),
+                           (T*)gx.data_ptr(), n / 8, C / 8,
+                           1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
+      else
+        hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
+                           dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                           (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                           sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                           1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
     } else {
-      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, false>), dim3(eblocks),
-                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
-                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
-                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
-                         1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
+      if (vec)
+        hipLaunchKernelGGL((bn_act_bwd_apply_vec_kernel<T, false>),
+                           dim3(vblocks), dim3(kBlock), 0, stream,
+                           (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                           (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                           sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(),
+                           (T*)gx.data_ptr(), n / 8, C / 8,
+                           1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
+      else
+        hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, false>), dim3(eblocks),
+                           dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                           (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                           sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                           1.0f / (float)rows, relu ? 1 : 0, (float)act_max);
     }
   });
   HIP_CHECK_LAST();
@@ -401,22 +667,35 @@ std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
   auto opts = x.options().dtype(torch::kFloat32);
   auto sum_g = torch::empty({C}, opts);
   auto sum_gx = torch::empty({C}, opts);
-  int gx_blocks = (C + 63) / 64;
-  int ngrp = kBlock / std::min(C, 64);  // first block's packing
-  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+  bool vec16 = x.element_size() == 2 && (C & 7) == 0;
+  int span = vec16 ? std::min(C / 8, 64) : std::min(C, 64);
+  int gx_blocks = vec16 ? (C / 8 + 63) / 64 : (C + 63) / 64;
+  int ngrp = kBlock / span;  // first block's packing
+  int rper = vec16 ? 1 : 4;
+  int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
                                   std::max(1, 4096 / ngrp));
   auto partial_g = torch::empty({gy, C}, opts);
   auto partial_gx = torch::empty({gy, C}, opts);
   auto stream = c10::hip::getCurrentHIPStream();
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd_reduce", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
-                       dim3(kBlock), 0, stream,
-                       (const T*)g.data_ptr(), (const T*)x.data_ptr(),
-                       (const T*)y.data_ptr(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), partial_g.data_ptr<float>(),
-                       partial_gx.data_ptr<float>(), rows, C, relu ? 1 : 0,
-                       (float)act_max);
+    if (vec16 && sizeof(T) == 2)
+      hipLaunchKernelGGL((bn_act_bwd_reduce_vec_kernel<T>),
+                         dim3(gx_blocks, gy), dim3(kBlock), 0, stream,
+                         (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                         (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         partial_g.data_ptr<float>(),
+                         partial_gx.data_ptr<float>(), rows, C, relu ? 1 : 0,
+                         (float)act_max);
+    else
+      hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T>), dim3(gx_blocks, gy),
+                         dim3(kBlock), 0, stream,
+                         (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                         (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), partial_g.data_ptr<float>(),
+                         partial_gx.data_ptr<float>(), rows, C, relu ? 1 : 0,
+                         (float)act_max);
   });
   hipLaunchKernelGGL(bn_partials_reduce_kernel,
                      dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0, stream,
@@ -439,22 +718,45 @@ torch::Tensor bn_act_bwd_apply(torch::Tensor g, torch::Tensor x,
   NN_DISPATCH(x.scalar_type(), "bn_act_bwd_apply", [&] {
     using T = typename DevT<scalar_t>::type;
     auto stream = c10::hip::getCurrentHIPStream();
-    if (training)
-      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
-                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
-                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
-                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
-                         1.0f / (float)count, relu ? 1 : 0, (float)act_max);
-    else
-      hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, false>), dim3(eblocks),
-                         dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
-                         (const T*)x.data_ptr(), (const T*)y.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
-                         sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
-                         1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+    bool vec = sizeof(T) == 2 && (C & 7) == 0;
+    int vblocks = (int)std::min<int64_t>((n / 8 + kBlock - 1) / kBlock, 8192);
+    if (training) {
+      if (vec)
+        hipLaunchKernelGGL((bn_act_bwd_apply_vec_kernel<T, true>),
+                           dim3(vblocks), dim3(kBlock), 0, stream,
+                           (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                           (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                           sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(),
+                           (T*)gx.data_ptr(), n / 8, C / 8,
+                           1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+      else
+        hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, true>), dim3(eblocks),
+                           dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                           (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                           sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                           1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+    } else {
+      if (vec)
+        hipLaunchKernelGGL((bn_act_bwd_apply_vec_kernel<T, false>),
+                           dim3(vblocks), dim3(kBlock), 0, stream,
+                           (const T*)g.data_ptr(), (const T*)x.data_ptr(),
+                           (const T*)y.data_ptr(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                           sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(),
+                           (T*)gx.data_ptr(), n / 8, C / 8,
+                           1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+      else
+        hipLaunchKernelGGL((bn_act_bwd_apply_kernel<T, false>), dim3(eblocks),
+                           dim3(kBlock), 0, stream, (const T*)g.data_ptr(),
+                           (const T*)x.data_ptr(), (const T*)y.data_ptr(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), sum_g.data_ptr<float>(),
+                           sum_gx.data_ptr<float>(), (T*)gx.data_ptr(), n, C,
+                           1.0f / (float)count, relu ? 1 : 0, (float)act_max);
+    }
   });
   HIP_CHECK_LAST();
   return gx;
@@ -488,18 +790,27 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
                                running_var.scalar_type() == torch::kFloat32),
               "bn_stats_finalize: running stats must be f32");
   auto opts = x.options().dtype(torch::kFloat32);
-  int gx = (C + 63) / 64;
-  int ngrp = kBlock / std::min(C, 64);  // first block's packing
-  int gy = (int)std::min<int64_t>((rows + 4 * ngrp - 1) / (4 * ngrp),
+  bool vec16 = x.element_size() == 2 && (C & 7) == 0;
+  int span = vec16 ? std::min(C / 8, 64) : std::min(C, 64);
+  int gx = vec16 ? (C / 8 + 63) / 64 : (C + 63) / 64;
+  int ngrp = kBlock / span;  // first block's packing
+  int rper = vec16 ? 2 : 4;
+  int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
                                   std::max(1, 4096 / ngrp));
   auto partial_sum = torch::empty({gy, C}, opts);
   auto partial_sumsq = torch::empty({gy, C}, opts);
   auto stream = c10::hip::getCurrentHIPStream();
   NN_DISPATCH(x.scalar_type(), "bn_stats_finalize", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
-                       stream, (const T*)ptr, partial_sum.data_ptr<float>(),
-                       partial_sumsq.data_ptr<float>(), rows, C);
+    if (vec16 && sizeof(T) == 2)
+      hipLaunchKernelGGL((bn_stats_vec_kernel<T>), dim3(gx, gy), dim3(kBlock),
+                         0, stream, (const T*)ptr,
+                         partial_sum.data_ptr<float>(),
+                         partial_sumsq.data_ptr<float>(), rows, C);
+    else
+      hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(gx, gy), dim3(kBlock), 0,
+                         stream, (const T*)ptr, partial_sum.data_ptr<float>(),
+                         partial_sumsq.data_ptr<float>(), rows, C);
   });
   float n = (float)rows;
   float unbias = n / std::max(n - 1.0f, 1.0f);

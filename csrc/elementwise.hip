@@ -56,6 +56,42 @@ __global__ void fake_quant_kernel(const T* __restrict__ x, T* __restrict__ out,
   }
 }
 
+// 4-wide vector-load form (16-bit dtypes, n % 4 == 0): one 8-B load/store
+// per quad instead of four 2-B ones; same Philox counter scheme, so the
+// stochastic-rounding draws are bit-identical to the scalar form
+template <typename T, bool STOCH>
+__global__ void fake_quant_vec_kernel(const T* __restrict__ x,
+                                      T* __restrict__ out, int64_t n_vec,
+                                      float min_value, float inv_scale,
+                                      float scale, float qmax, float stoch,
+                                      uint64_t seed,
+                                      const int64_t* __restrict__ seed_base) {
+  struct alignas(8) V4 { T v[4]; };
+  seed = graph_seed(seed_base, seed);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_vec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    V4 xv = ((const V4*)x)[i];
+    float u[4] = {0.0f, 0.0f, 0.0f, 0.0f};
+    if (STOCH) {
+      Philox4 ph = philox4x32(seed, (uint64_t)i);
+      u[0] = (2.0f * u01(ph.x) - 1.0f) * stoch;
+      u[1] = (2.0f * u01(ph.y) - 1.0f) * stoch;
+      u[2] = (2.0f * u01(ph.z) - 1.0f) * stoch;
+      u[3] = (2.0f * u01(ph.w) - 1.0f) * stoch;
+    }
+    V4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float q = (to_f32(xv.v[j]) - min_value) * inv_scale;
+      if (STOCH) q += u[j];
+      q = fminf(fmaxf(q, 0.0f), qmax);
+      q = nearbyintf(q);
+      o.v[j] = from_f32<T>(q * scale + min_value);
+    }
+    ((V4*)out)[i] = o;
+  }
+}
+
 // --------------------------------------------------------------------------
 // STE mask: grad_in = grad_out * (min <= x <= max)
 // --------------------------------------------------------------------------
@@ -67,6 +103,32 @@ __global__ void ste_mask_kernel(const T* __restrict__ g, const T* __restrict__ x
        i += (int64_t)gridDim.x * blockDim.x) {
     float xv = to_f32(x[i]);
     out[i] = (xv >= min_value && xv <= max_value) ? g[i] : from_f32<T>(0.0f);
+  }
+}
+
+// 8-wide 16-B vector form (16-bit dtypes, n % 8 == 0): the scalar form was
+// issue-bound at 2 B per lane per load
+template <typename T>
+struct alignas(16) Pack8 { T v[8]; };
+
+template <typename T>
+__global__ void ste_mask_vec_kernel(const T* __restrict__ g,
+                                    const T* __restrict__ x,
+                                    T* __restrict__ out, int64_t n_vec,
+                                    float min_value, float max_value) {
+  using V = Pack8<T>;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_vec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    V gv = ((const V*)g)[i];
+    V xv = ((const V*)x)[i];
+    V o;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      float f = to_f32(xv.v[u]);
+      o.v[u] = (f >= min_value && f <= max_value) ? gv.v[u]
+                                                  : from_f32<T>(0.0f);
+    }
+    ((V*)out)[i] = o;
   }
 }
 
@@ -169,15 +231,29 @@ torch::Tensor fake_quant_fwd(torch::Tensor x, int64_t num_bits, double min_value
   NN_DISPATCH(xc.scalar_type(),
                                   "fake_quant_fwd", [&] {
     using T = typename DevT<scalar_t>::type;
-    if (stochastic > 0) {
+    auto stream = c10::hip::getCurrentHIPStream();
+    if (sizeof(T) == 2 && (n & 3) == 0) {
+      if (stochastic > 0)
+        hipLaunchKernelGGL((fake_quant_vec_kernel<T, true>),
+                           dim3(grid_1d(n / 4)), dim3(kBlock), 0, stream,
+                           (const T*)xc.data_ptr(), (T*)out.data_ptr(), n / 4,
+                           (float)min_value, 1.0f / scale, scale, qmax,
+                           (float)stochastic, (uint64_t)seed, g_seed_base);
+      else
+        hipLaunchKernelGGL((fake_quant_vec_kernel<T, false>),
+                           dim3(grid_1d(n / 4)), dim3(kBlock), 0, stream,
+                           (const T*)xc.data_ptr(), (T*)out.data_ptr(), n / 4,
+                           (float)min_value, 1.0f / scale, scale, qmax, 0.0f,
+                           (uint64_t)seed, g_seed_base);
+    } else if (stochastic > 0) {
       hipLaunchKernelGGL((fake_quant_kernel<T, true>), dim3(grid_1d((n + 3) / 4)),
-                         dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                         dim3(kBlock), 0, stream,
                          (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
                          (float)min_value, 1.0f / scale, scale, qmax,
                          (float)stochastic, (uint64_t)seed, g_seed_base);
     } else {
       hipLaunchKernelGGL((fake_quant_kernel<T, false>), dim3(grid_1d((n + 3) / 4)),
-                         dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                         dim3(kBlock), 0, stream,
                          (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
                          (float)min_value, 1.0f / scale, scale, qmax, 0.0f,
                          (uint64_t)seed, g_seed_base);
@@ -196,10 +272,19 @@ torch::Tensor ste_mask(torch::Tensor grad, torch::Tensor x, double min_value,
   NN_DISPATCH(g.scalar_type(),
                                   "ste_mask", [&] {
     using T = typename DevT<scalar_t>::type;
-    hipLaunchKernelGGL((ste_mask_kernel<T>), dim3(grid_1d(n)), dim3(kBlock), 0,
-                       c10::hip::getCurrentHIPStream(), (const T*)g.data_ptr(),
-                       (const T*)xc.data_ptr(), (T*)out.data_ptr(), n,
-                       (float)min_value, (float)max_value);
+    if (sizeof(T) == 2 && (n & 7) == 0) {
+      hipLaunchKernelGGL((ste_mask_vec_kernel<T>), dim3(grid_1d(n / 8)),
+                         dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                         (const T*)g.data_ptr(), (const T*)xc.data_ptr(),
+                         (T*)out.data_ptr(), n / 8, (float)min_value,
+                         (float)max_value);
+    } else {
+      hipLaunchKernelGGL((ste_mask_kernel<T>), dim3(grid_1d(n)), dim3(kBlock),
+                         0, c10::hip::getCurrentHIPStream(),
+                         (const T*)g.data_ptr(), (const T*)xc.data_ptr(),
+                         (T*)out.data_ptr(), n, (float)min_value,
+                         (float)max_value);
+    }
   });
   HIP_CHECK_LAST();
   return out;
