@@ -195,6 +195,26 @@ class CondConv2d(nn.Module):
         bias = None
         if self.bias is not None:
             bias = torch.matmul(routing_weights, self.bias).view(B * self.out_channels)
+        # Per-sample conv via unfold + one batched GEMM (K13): the
+        # reference's batch*groups grouped conv (conv2d_layers.py:204-221)
+        # is replaced by ops.per_sample_conv2d -- no F.conv2d on this path.
+        if self.dilation == (1, 1):
+            xs = x
+            padding = self.padding
+            if self.dynamic_padding:
+                ih, iw = x.size()[-2:]
+                pad_h = _calc_same_pad(ih, self.kernel_size[0], self.stride[0], 1)
+                pad_w = _calc_same_pad(iw, self.kernel_size[1], self.stride[1], 1)
+                xs = F.pad(x, (pad_w // 2, pad_w - pad_w // 2,
+                               pad_h // 2, pad_h - pad_h // 2))
+                padding = (0, 0)
+            per_bias = None
+            if bias is not None:
+                per_bias = bias.view(B, self.out_channels)
+            return ops.per_sample_conv2d(xs, weight, per_bias,
+                                         stride=self.stride, padding=padding,
+                                         groups=self.groups)
+        # dilated CondConv (unused by the model zoo): grouped-conv route
         x = x.view(1, B * C, H, W)
         if self.dynamic_padding:
             out = conv2d_same(x, weight, bias, stride=self.stride,

@@ -1075,3 +1075,43 @@ def kth_percentile(x, pctl):
     if use_native(x):
         return ext().kth_percentile(x.contiguous().view(-1), float(pctl))
     return ref.kth_percentile(x, pctl)
+
+
+# ---------------------------------------------------------------------------
+# Per-sample (conditionally-parameterized) grouped conv
+# ---------------------------------------------------------------------------
+
+
+def per_sample_conv2d(x, weight, bias=None, stride=1, padding=0, groups=1):
+    """Convolution where every sample has its OWN filter bank (CondConv,
+    reference conv2d_layers.py:152-240).
+
+    The reference runs this as one grouped conv with batch*groups groups --
+    a shape cuDNN special-cases but that maps terribly to grouped-conv
+    kernels. Here it is unfold + ONE batched GEMM: im2col columns
+    [B, g, L, C/g*R*S] contracted against the per-sample filters
+    [B, g, C/g*R*S, K/g] via a single rocBLAS strided-batched GEMM (plain
+    library GEMMs are the one sanctioned library path). Fully
+    differentiable (double-backward included) through torch autograd.
+
+    x: [B, C, H, W]; weight: [B*K, C/g, R, S] (per-sample filters, sample-
+    major as CondConv produces them); returns [B, K, OH, OW].
+    """
+    B, C, H, W = x.shape
+    BK, Cg, R, S = weight.shape
+    K = BK // B
+    st = (stride, stride) if isinstance(stride, int) else tuple(stride)
+    pd = (padding, padding) if isinstance(padding, int) else tuple(padding)
+    OH = (H + 2 * pd[0] - R) // st[0] + 1
+    OW = (W + 2 * pd[1] - S) // st[1] + 1
+    cols = F.unfold(x, (R, S), padding=pd, stride=st)  # [B, C*R*S, L]
+    L = cols.shape[-1]
+    g = groups
+    # columns laid out [g, C/g*R*S] along dim 1
+    cols = cols.view(B, g, Cg * R * S, L)
+    w = weight.view(B, g, K // g, Cg * R * S)
+    out = torch.matmul(w, cols)                        # [B, g, K/g, L]
+    out = out.reshape(B, K, OH, OW)
+    if bias is not None:
+        out = out + bias.view(B, K, 1, 1)
+    return out

@@ -400,3 +400,36 @@ def test_nvnovograd_step():
     assert losses[-1] < 0.2 * losses[0]
     state = opt.state[w]
     assert state['exp_avg_sq'].dim() == 0  # layer-wise scalar moment
+
+
+def test_condconv_per_sample_matches_grouped_oracle():
+    """CondConv2d's unfold+batched-GEMM path (K13 native route) must equal
+    the reference's batch*groups grouped-conv formulation."""
+    import torch
+    import torch.nn.functional as F
+
+    from noisynet_amd.models.conv2d_layers import CondConv2d, conv2d_same
+
+    torch.manual_seed(0)
+    for groups, pad in [(1, ''), (1, 1), (4, 1)]:
+        m = CondConv2d(8, 16, kernel_size=3, stride=1, padding=pad,
+                       groups=groups, bias=True, num_experts=4)
+        x = torch.randn(5, 8, 10, 10, requires_grad=True)
+        rw = torch.softmax(torch.randn(5, 4), dim=1)
+        out = m(x, rw)
+        B = 5
+        w = torch.matmul(rw, m.weight).view(
+            B * m.out_channels, m.in_channels // m.groups, 3, 3)
+        b = torch.matmul(rw, m.bias).view(B * m.out_channels)
+        x2 = x.view(1, B * 8, 10, 10)
+        if m.dynamic_padding:
+            ref = conv2d_same(x2, w, b, stride=m.stride, padding=m.padding,
+                              dilation=(1, 1), groups=m.groups * B)
+        else:
+            ref = F.conv2d(x2, w, b, stride=m.stride, padding=m.padding,
+                           dilation=(1, 1), groups=m.groups * B)
+        ref = ref.permute([1, 0, 2, 3]).reshape(B, m.out_channels,
+                                                ref.shape[-2], ref.shape[-1])
+        assert (out - ref).abs().max().item() < 1e-5, (groups, pad)
+        out.sum().backward()
+        assert x.grad is not None and m.weight.grad is not None
