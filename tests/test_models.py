@@ -433,3 +433,54 @@ def test_condconv_per_sample_matches_grouped_oracle():
         assert (out - ref).abs().max().item() < 1e-5, (groups, pad)
         out.sum().backward()
         assert x.grad is not None and m.weight.grad is not None
+
+
+def test_plot_surface_roundtrip(tmp_path):
+    """--write artifacts (layers/array_names/input_sizes/layer_power .npy)
+    round-trip through the multi-model comparison plot (reference
+    plot_histograms.py:513-605)."""
+    import numpy as np
+    import torch
+
+    from noisynet_amd import plot_histograms as ph
+
+    # synthesize two "result dirs" of capture artifacts
+    dirs = []
+    for run in range(2):
+        d = tmp_path / ("run%d" % run)
+        d.mkdir()
+        torch.manual_seed(run)
+        layers = []
+        sizes = []
+        for l in range(4):
+            per_layer = []
+            x = torch.randn(2, 3, 8, 8)
+            w = torch.randn(4, 3, 3, 3) * 0.2
+            y = torch.nn.functional.conv2d(x, w)
+            ph.get_layers(per_layer, x, w, y, stride=1, padding=0,
+                          layer='conv')
+            layers.append(per_layer)
+            sizes.append(int(np.prod(w.shape[1:])))
+        names = ph._CAPTURE_NAMES[:len(layers[0])]
+        np.save(d / 'layers.npy', np.array(layers, dtype=object),
+                allow_pickle=True)
+        np.save(d / 'array_names.npy', np.array(names, dtype=object),
+                allow_pickle=True)
+        np.save(d / 'input_sizes.npy', np.array(sizes), allow_pickle=True)
+        np.save(d / 'layer_power.npy', np.arange(4, dtype=np.float64),
+                allow_pickle=True)
+        (d / 'model_epoch_3_acc_77.10.pth').write_bytes(b'')
+        dirs.append(str(d))
+
+    out = ph.compare_result_dirs(dirs, var='seed', vars=[0, 1])
+    assert out is not None
+    import os
+    assert os.path.exists(out)
+    # and the single-capture grid path
+    out2 = ph.plot_layers(num_layers=4, models=[dirs[0]], epoch=1, i=2,
+                          layers=np.load(os.path.join(dirs[0], 'layers.npy'),
+                                         allow_pickle=True),
+                          names=list(np.load(
+                              os.path.join(dirs[0], 'array_names.npy'),
+                              allow_pickle=True)), acc=50.0, tag='t')
+    assert os.path.exists(out2)
