@@ -104,6 +104,9 @@ torch::Tensor conv_wgrad_from_col(torch::Tensor gy, torch::Tensor col,
 torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
                                 int64_t stride, int64_t pad, int64_t R,
                                 int64_t S);
+torch::Tensor conv_wgrad_patch(torch::Tensor gy, torch::Tensor x,
+                               int64_t stride, int64_t pad, int64_t R,
+                               int64_t S);
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w);
 torch::Tensor linear_dgrad(torch::Tensor gy, torch::Tensor w);
 torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x);
@@ -162,6 +165,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);
   m.def("conv_wgrad_im2col", &conv_wgrad_im2col);
+  m.def("conv_wgrad_patch", &conv_wgrad_patch);
   m.def("conv_wgrad_from_col", &conv_wgrad_from_col);
   m.def("im2col_materialize", &im2col_materialize);
   m.def("linear_fwd", &linear_fwd);

@@ -1109,6 +1109,119 @@ void wgrad_mk4_kernel(const T* __restrict__ gy, const T* __restrict__ x,
       }
 }
 
+// --------------------------------------------------------------------------
+// Patch wgrad: the im2col matrix is never materialized. The x operand of
+// the 128x128 wgrad GEMM is gathered straight from the (channel-padded)
+// NHWC input: column index (tap r,s | channel c) decodes to the 16-B
+// aligned address ((n*H + oh+r)*W + ow+s)*Cp + c. Removes the global
+// im2col pass (2.7 GB write+read at batch 8192 for conv2) entirely;
+// concurrent c-tiles of the same m-slice re-read an ~36 KB input window,
+// so the repeated taps come from L2, not HBM. Per-tap channel padding to
+// Cp = round8(C) keeps every 8-wide load 16-B aligned (the flat layout's
+// odd-C columns forced scalar staging).
+template <typename T>
+DEV_INLINE void wgrad_patch_stage(char* lds, const T* __restrict__ x,
+                                  int64_t m0, int64_t M, int H, int W, int OH,
+                                  int OW, int S, int Cp, int CRSp, int stride,
+                                  int pad, int c0) {
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int idx = threadIdx.x + i * kBlock;  // 0..2047
+    int mi = idx >> 4;                   // m within the 128-round
+    int seg = idx & 15;                  // 8-col group
+    int col0 = c0 + seg * 8;
+    int tap = col0 / Cp;                 // Cp % 8 == 0, col0 % 8 == 0:
+    int cin = col0 - tap * Cp;           // a group never crosses a tap
+    int r = tap / S, s = tap - (tap / S) * S;
+    int64_t m = m0 + mi;
+    T vals[8];
+    bool ok = (m < M) && (col0 < CRSp);
+    int64_t addr = 0;
+    if (ok) {
+      int64_t n = m / ((int64_t)OH * OW);
+      int p = (int)(m - n * OH * OW);
+      int oh = p / OW, ow = p - (p / OW) * OW;
+      int ih = oh * stride - pad + r, iw = ow * stride - pad + s;
+      ok = ih >= 0 && ih < H && iw >= 0 && iw < W;
+      addr = ((n * H + ih) * (int64_t)W + iw) * Cp + cin;
+    }
+    if (ok) {
+      *(bf16x8*)vals = *(const bf16x8*)(x + addr);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = from_f32<T>(0.0f);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int row = seg * 8 + j;
+      *(T*)(lds + row * WG_LSTR + wg_swz(row, mi >> 3) * 16 + (mi & 7) * 2) =
+          vals[j];
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(kBlock)
+void conv_wgrad_patch_kernel(const T* __restrict__ gy,
+                             const T* __restrict__ x,
+                             float* __restrict__ dw /* [z][Kp, CRSp] f32 */,
+                             int64_t M, int Kp, int H, int W, int OH, int OW,
+                             int S, int Cp, int CRSp, int stride, int pad,
+                             int mchunks_per_block) {
+  int c0 = blockIdx.x * 128;
+  int k0 = blockIdx.y * 128;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* g_lds = smem;                        // [128 k rows][128 m]
+  char* x_lds = smem + 128 * WG_LSTR;        // [128 crs rows][128 m]
+
+  int wid = threadIdx.x / WAVE;
+  int wm = wid >> 1, wn = wid & 1;
+  int lane = threadIdx.x & (WAVE - 1);
+  f32x4 acc[4][4] = {};
+
+  int64_t m_start = (int64_t)blockIdx.z * mchunks_per_block * WG_BK;
+  int64_t m_end = m_start + (int64_t)mchunks_per_block * WG_BK;
+  if (m_end > M) m_end = M;
+
+  for (int64_t m0 = m_start; m0 < m_end; m0 += WG_BK) {
+    wgrad_mk4_stage(g_lds, gy, m0, M, Kp, k0);
+    wgrad_patch_stage(x_lds, x, m0, M, H, W, OH, OW, S, Cp, CRSp, stride,
+                      pad, c0);
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        int arow = wm * 64 + fm * 16 + (lane & 15);
+        auto a = *(typename Mma<T>::frag*)(
+            g_lds + arow * WG_LSTR +
+            wg_swz(arow, ks * 4 + (lane >> 4)) * 16);
+#pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          int brow = wn * 64 + fn * 16 + (lane & 15);
+          auto b = *(typename Mma<T>::frag*)(
+              x_lds + brow * WG_LSTR +
+              wg_swz(brow, ks * 4 + (lane >> 4)) * 16);
+          Mma<T>::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  float* slab = dw + (int64_t)blockIdx.z * Kp * CRSp;
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int k = k0 + wm * 64 + fm * 16 + 4 * (lane >> 4) + reg;
+        int c = c0 + wn * 64 + fn * 16 + (lane & 15);
+        if (k < Kp && c < CRSp) slab[(int64_t)k * CRSp + c] = acc[fm][fn][reg];
+      }
+}
+
 template <typename scalar_t> struct DevT { using type = scalar_t; };
 template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct DevT<at::Half> { using type = _Float16; };
@@ -2442,6 +2555,60 @@ torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
   check_cl(x, "conv_wgrad_im2col x");
   auto col = im2col_materialize(x, gy.size(1), stride, pad, R, S);
   return conv_wgrad_from_col(gy, col, x.size(1), R, S);
+}
+
+// Patch wgrad: no materialized im2col (see conv_wgrad_patch_kernel).
+// 16-bit dtypes; any stride/pad.
+torch::Tensor conv_wgrad_patch(torch::Tensor gy, torch::Tensor x,
+                               int64_t stride, int64_t pad, int64_t R,
+                               int64_t S) {
+  check_cl(gy, "conv_wgrad_patch gy");
+  check_cl(x, "conv_wgrad_patch x");
+  TORCH_CHECK(x.element_size() == 2, "conv_wgrad_patch: 16-bit only");
+  int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int64_t K = gy.size(1), OH = gy.size(2), OW = gy.size(3);
+  int64_t M = N * OH * OW;
+  int64_t Cp = (C + 7) & ~7;
+  int64_t CRSp = R * S * Cp;
+
+  // raw NHWC views (zero-copy for channels_last)
+  auto x2 = x.permute({0, 2, 3, 1}).reshape({N * H * W, C});
+  torch::Tensor xp = (Cp == C) ? x2.contiguous() : pad_cols8(x2.contiguous(), Cp);
+  auto gy2 = gy.permute({0, 2, 3, 1}).reshape({M, K});
+  int64_t Kp = (K + 7) & ~7;
+  torch::Tensor gyp = (Kp == K) ? gy2.contiguous()
+                                : pad_cols8(gy2.contiguous(), Kp);
+
+  int ctiles = (int)((CRSp + 127) / 128);
+  int ktiles = (int)((Kp + 127) / 128);
+  int64_t mtotal = (M + WG_BK - 1) / WG_BK;
+  int64_t target_z = std::max<int64_t>(1, 2048 / std::max(1, ctiles * ktiles));
+  int chunks = (int)((mtotal + target_z - 1) / target_z);
+  int mslices = (int)((mtotal + chunks - 1) / chunks);
+  dim3 grid(ctiles, ktiles, mslices);
+  size_t lds = (size_t)256 * WG_LSTR;
+  auto parts = torch::empty({(int64_t)mslices, Kp, CRSp},
+                            x.options().dtype(torch::kFloat32));
+  NN_DISPATCH(gy.scalar_type(), "conv_wgrad_patch", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto* kfn = &conv_wgrad_patch_kernel<T>;
+    if (lds > 64 * 1024)
+      hipFuncSetAttribute((const void*)kfn,
+                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                          (int)lds);
+    hipLaunchKernelGGL(kfn, grid, dim3(kBlock), lds,
+                       c10::hip::getCurrentHIPStream(),
+                       (const T*)gyp.data_ptr(), (const T*)xp.data_ptr(),
+                       parts.data_ptr<float>(), M, (int)Kp, (int)H, (int)W,
+                       (int)OH, (int)OW, (int)S, (int)Cp, (int)CRSp,
+                       (int)stride, (int)pad, chunks);
+  });
+  HIP_CHECK_LAST();
+  auto dw = parts.sum(0).narrow(0, 0, K)
+                .view({K, R, S, Cp}).narrow(3, 0, C)
+                .permute({0, 3, 1, 2})
+                .contiguous(at::MemoryFormat::ChannelsLast);
+  return dw.to(x.scalar_type());
 }
 
 // wgrad when the flat im2col matrix already exists (shared from the

@@ -134,9 +134,13 @@ def _conv_wgrad_raw(g, x, stride, padding, w_shape, col=None):
         if col is not None and R * S > 1:
             # flat im2col matrix shared from the forward pass
             return ext().conv_wgrad_from_col(_nhwc(g), col, x.shape[1], R, S)
+        if R * S > 1 and x.element_size() == 2:
+            # patch wgrad: x operand gathered in-kernel from the NHWC
+            # input -- no materialized im2col pass at all
+            return ext().conv_wgrad_patch(_nhwc(g), _nhwc(x), stride,
+                                          padding, R, S)
         if R * S > 1:
-            # im2col-GEMM wgrad when the buffer is affordable (<2 GB):
-            # one coalesced materialization pass beats per-K-chunk gather
+            # fp32: im2col-GEMM wgrad when the buffer is affordable (<2 GB)
             C_in = x.shape[1]
             cols_p = (R * S * C_in if C_in % 8 == 0
                       else (R * S * C_in + 31) // 32 * 32)

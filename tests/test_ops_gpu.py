@@ -819,3 +819,38 @@ def test_bn_stats_odd_channels():
         ref_var = x.float().var(dim=(0, 2, 3), unbiased=False)
         assert torch.allclose(mean.cpu(), ref_mean.cpu(), atol=5e-3), C
         assert torch.allclose(var.cpu(), ref_var.cpu(), atol=5e-3, rtol=1e-2), C
+
+
+def test_conv_wgrad_patch_matches_reference():
+    """Patch wgrad (in-kernel im2col gather) vs torch fp32 conv2d_weight."""
+    torch.manual_seed(21)
+    shapes = [
+        # (N, C, H, W, K, R, stride, pad): conv2-like, conv1-like, padded,
+        # strided (ResNet downsample shape)
+        (16, 65, 14, 14, 120, 5, 1, 0),
+        (8, 3, 32, 32, 65, 5, 1, 0),
+        (4, 24, 16, 16, 32, 3, 1, 1),
+        (4, 32, 16, 16, 64, 3, 2, 1),
+    ]
+    for (N, C, H, W, K, R, stride, pad) in shapes:
+        x = to_bf16_representable(torch.randn(N, C, H, W) * 0.5)
+        OH = (H + 2 * pad - R) // stride + 1
+        gy = to_bf16_representable(torch.randn(N, K, OH, OH) * 0.5)
+        got = ops.ext().conv_wgrad_patch(
+            cl(gy.cuda().bfloat16()), cl(x.cuda().bfloat16()),
+            stride, pad, R, R)
+        ref = torch.nn.grad.conv2d_weight(
+            x.float(), (K, C, R, R), gy.float(), stride, pad)
+        err = (got.float().cpu() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 0.02, ((N, C, H, W, K, R, stride, pad),
+                                    err / scale)
+
+
+def test_conv_wgrad_patch_deterministic():
+    torch.manual_seed(3)
+    x = cl(torch.randn(8, 65, 14, 14).cuda().bfloat16())
+    gy = cl(torch.randn(8, 120, 10, 10).cuda().bfloat16())
+    a = ops.ext().conv_wgrad_patch(gy, x, 1, 0, 5, 5)
+    b = ops.ext().conv_wgrad_patch(gy, x, 1, 0, 5, 5)
+    assert torch.equal(a, b)
