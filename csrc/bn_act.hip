@@ -153,21 +153,39 @@ __global__ void bn_stats_vec_kernel(const T* __restrict__ x,
   }
 }
 
-// fixed-order reduction of two [gy, C] partial arrays to [C]
+// Deterministic reduction of two [gy, C] partial arrays to [C].
+// Parallel over rows as well as channels: 8 fixed row-strided streams per
+// 32-channel chunk, combined in fixed stream order (the one-thread-per-
+// channel serial loop was 4 waves crawling 1024 rows -- 7 ms/step on
+// MobileNetV2). Every partition is a pure function of thread indices, so
+// same-input runs are bit-identical.
 __global__ void bn_partials_reduce_kernel(const float* __restrict__ pa,
                                           const float* __restrict__ pb,
                                           float* __restrict__ outa,
                                           float* __restrict__ outb, int gy,
                                           int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float la[256], lb[256];
+  int c = blockIdx.x * 32 + (threadIdx.x & 31);
+  int rg = threadIdx.x >> 5;  // 8 row streams
   float a = 0.0f, b = 0.0f;
-  for (int r = 0; r < gy; ++r) {
-    a += pa[(int64_t)r * C + c];
-    b += pb[(int64_t)r * C + c];
+  if (c < C) {
+    for (int r = rg; r < gy; r += 8) {
+      a += pa[(int64_t)r * C + c];
+      b += pb[(int64_t)r * C + c];
+    }
   }
-  outa[c] = a;
-  outb[c] = b;
+  la[threadIdx.x] = a;
+  lb[threadIdx.x] = b;
+  __syncthreads();
+  if (threadIdx.x < 32 && c < C) {
+    float sa = 0.0f, sb = 0.0f;
+    for (int g = 0; g < 8; ++g) {
+      sa += la[g * 32 + threadIdx.x];
+      sb += lb[g * 32 + threadIdx.x];
+    }
+    outa[c] = sa;
+    outb[c] = sb;
+  }
 }
 
 template <typename T>
@@ -231,13 +249,26 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partial_sum,
                                    int C, float inv_n, float unbias,
                                    float momentum, float eps,
                                    int has_running) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  // fixed-order partials reduce folded in (determinism, see bn_stats_kernel)
+  // deterministic partials reduce folded in, parallel over 8 fixed row
+  // streams per 32-channel chunk (see bn_partials_reduce_kernel)
+  __shared__ float la[256], lb[256];
+  int c = blockIdx.x * 32 + (threadIdx.x & 31);
+  int rg = threadIdx.x >> 5;
+  float a = 0.0f, b = 0.0f;
+  if (c < C) {
+    for (int r = rg; r < gy; r += 8) {
+      a += partial_sum[(int64_t)r * C + c];
+      b += partial_sumsq[(int64_t)r * C + c];
+    }
+  }
+  la[threadIdx.x] = a;
+  lb[threadIdx.x] = b;
+  __syncthreads();
+  if (threadIdx.x >= 32 || c >= C) return;
   float s = 0.0f, q = 0.0f;
-  for (int r = 0; r < gy; ++r) {
-    s += partial_sum[(int64_t)r * C + c];
-    q += partial_sumsq[(int64_t)r * C + c];
+  for (int g = 0; g < 8; ++g) {
+    s += la[g * 32 + threadIdx.x];
+    q += lb[g * 32 + threadIdx.x];
   }
   float m = s * inv_n;
   float v = fmaxf(q * inv_n - m * m, 0.0f);
@@ -295,7 +326,7 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
                          partial_sumsq.data_ptr<float>(), rows, C);
   });
   hipLaunchKernelGGL(bn_partials_reduce_kernel,
-                     dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0, stream,
+                     dim3((C + 31) / 32), dim3(256), 0, stream,
                      partial_sum.data_ptr<float>(),
                      partial_sumsq.data_ptr<float>(), sum.data_ptr<float>(),
                      sumsq.data_ptr<float>(), gy, C);
@@ -600,7 +631,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
                          partial_gx.data_ptr<float>(),
                          rows, C, relu ? 1 : 0, (float)act_max);
     hipLaunchKernelGGL(bn_partials_reduce_kernel,
-                       dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0,
+                       dim3((C + 31) / 32), dim3(256), 0,
                        stream, partial_g.data_ptr<float>(),
                        partial_gx.data_ptr<float>(), sum_g.data_ptr<float>(),
                        sum_gx.data_ptr<float>(), gy, C);
@@ -698,7 +729,7 @@ std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
                          (float)act_max);
   });
   hipLaunchKernelGGL(bn_partials_reduce_kernel,
-                     dim3((C + kBlock - 1) / kBlock), dim3(kBlock), 0, stream,
+                     dim3((C + 31) / 32), dim3(256), 0, stream,
                      partial_g.data_ptr<float>(), partial_gx.data_ptr<float>(),
                      sum_g.data_ptr<float>(), sum_gx.data_ptr<float>(), gy, C);
   HIP_CHECK_LAST();
@@ -816,7 +847,7 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
   float unbias = n / std::max(n - 1.0f, 1.0f);
   auto mean = torch::empty({C}, opts);
   auto invstd = torch::empty({C}, opts);
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 31) / 32), dim3(256), 0,
                      stream, partial_sum.data_ptr<float>(),
                      partial_sumsq.data_ptr<float>(),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),

@@ -389,12 +389,16 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   TORCH_CHECK(taps <= 64, "dwconv_wgrad: filter too large");
   constexpr int VEC = 8;
   bool vec = (C % VEC == 0) && gy.element_size() == 2 && taps <= 9;
+  // 5x5 filters: acc[25][8] would spill; a 4-wide vector still beats the
+  // scalar form 4x on address math
+  bool vec4 = !vec && (C % 4 == 0) && gy.element_size() == 2 && taps <= 25;
   torch::Tensor parts;
   NN_DISPATCH(gy.scalar_type(), "dwconv_wgrad", [&] {
     using T = typename DevT<scalar_t>::type;
     auto stream = c10::hip::getCurrentHIPStream();
-    if (vec) {
-      int Cv = C / VEC;
+    if (vec || vec4) {
+      int V = vec ? VEC : 4;
+      int Cv = C / V;
       int cspan_v = std::min(Cv, kBlock);
       int pgrp = kBlock / cspan_v;
       int cblocks = (Cv + cspan_v - 1) / cspan_v;
@@ -406,11 +410,20 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       int64_t rows = (int64_t)mslices * pgrp;
       parts = torch::zeros({rows, (int64_t)C * taps},
                            x.options().dtype(torch::kFloat32));
-      hipLaunchKernelGGL((dwconv_wgrad_vec_kernel<T, VEC, 9>),
-                         dim3(cblocks, mslices), dim3(kBlock), 0, stream,
-                         (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
-                         parts.data_ptr<float>(), Cv, C, H, W, OH, OW, (int)R,
-                         (int)S, (int)stride, (int)pad, npix, cspan_v, pgrp);
+      if (vec)
+        hipLaunchKernelGGL((dwconv_wgrad_vec_kernel<T, VEC, 9>),
+                           dim3(cblocks, mslices), dim3(kBlock), 0, stream,
+                           (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                           parts.data_ptr<float>(), Cv, C, H, W, OH, OW,
+                           (int)R, (int)S, (int)stride, (int)pad, npix,
+                           cspan_v, pgrp);
+      else
+        hipLaunchKernelGGL((dwconv_wgrad_vec_kernel<T, 4, 25>),
+                           dim3(cblocks, mslices), dim3(kBlock), 0, stream,
+                           (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                           parts.data_ptr<float>(), Cv, C, H, W, OH, OW,
+                           (int)R, (int)S, (int)stride, (int)pad, npix,
+                           cspan_v, pgrp);
     } else {
       int cspan = std::min(C, kBlock);
       int pgrp = kBlock / cspan;
