@@ -388,7 +388,13 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   int64_t npix = (int64_t)N * OH * OW;
   TORCH_CHECK(taps <= 64, "dwconv_wgrad: filter too large");
   constexpr int VEC = 8;
-  bool vec = (C % VEC == 0) && gy.element_size() == 2 && taps <= 9;
+  // NOISYNET_DW_WGRAD_VEC4=1: force the narrower accumulator form for
+  // 3x3 too (occupancy experiment; acc[9][8] is 72 VGPRs)
+  static const bool force4 = [] {
+    const char* e = getenv("NOISYNET_DW_WGRAD_VEC4");
+    return e && e[0] == '1';
+  }();
+  bool vec = !force4 && (C % VEC == 0) && gy.element_size() == 2 && taps <= 9;
   // 5x5 filters: acc[25][8] would spill; a 4-wide vector still beats the
   // scalar form 4x on address math
   bool vec4 = !vec && (C % 4 == 0) && gy.element_size() == 2 && taps <= 25;
