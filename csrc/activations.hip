@@ -72,6 +72,76 @@ __global__ void act_bwd_kernel(const T* __restrict__ g, const T* __restrict__ x,
     gx[i] = from_f32<T>(to_f32(g[i]) * act_bwd_f(act, to_f32(x[i])));
 }
 
+// ---------------------------------------------------------------------------
+// Squeeze-Excite gating: y = x * gate(s[b,c]) with a per-(batch,channel)
+// scalar, NHWC. The eager broadcast-mul + its backward reduce were ~6 ms
+// of an EfficientNet-B0 step; fused here into one elementwise pass each
+// way plus a deterministic per-(b,c) reduce.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void se_scale_fwd_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ s,
+                                    T* __restrict__ y, int64_t n, int C,
+                                    int64_t hwC, int act) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t b = i / hwC;
+    float gate = act_fwd_f(act, to_f32(s[b * C + c]));
+    y[i] = from_f32<T>(to_f32(x[i]) * gate);
+  }
+}
+
+template <typename T>
+__global__ void se_scale_bwd_gx_kernel(const T* __restrict__ g,
+                                       const T* __restrict__ s,
+                                       T* __restrict__ gx, int64_t n, int C,
+                                       int64_t hwC, int act) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    int64_t b = i / hwC;
+    float gate = act_fwd_f(act, to_f32(s[b * C + c]));
+    gx[i] = from_f32<T>(to_f32(g[i]) * gate);
+  }
+}
+
+// gs[b,c] = (sum_hw g*x) * act'(s[b,c]); one block per (b, 64-channel
+// chunk), 4 fixed row streams combined in order (deterministic)
+template <typename T>
+__global__ void se_scale_bwd_gs_kernel(const T* __restrict__ g,
+                                       const T* __restrict__ x,
+                                       const T* __restrict__ s,
+                                       T* __restrict__ gs, int64_t hw, int C,
+                                       int act) {
+  __shared__ float ls[kBlock];
+  int64_t b = blockIdx.x;
+  int cbase = blockIdx.y * 64;
+  int cw = C - cbase;
+  if (cw > 64) cw = 64;
+  int c_l = (cw == 64) ? (threadIdx.x & 63) : (threadIdx.x % cw);
+  int rg = (cw == 64) ? (threadIdx.x >> 6) : (threadIdx.x / cw);
+  int ngrp = kBlock / cw;
+  float acc = 0.0f;
+  if (rg < ngrp) {
+    int c = cbase + c_l;
+    const T* gb = g + b * hw * C;
+    const T* xb = x + b * hw * C;
+    for (int64_t r = rg; r < hw; r += ngrp)
+      acc += to_f32(gb[r * C + c]) * to_f32(xb[r * C + c]);
+  }
+  ls[threadIdx.x] = acc;
+  __syncthreads();
+  if (threadIdx.x < cw) {
+    float sum = 0.0f;
+    for (int gp = 0; gp < ngrp; ++gp) sum += ls[gp * cw + threadIdx.x];
+    int c = cbase + threadIdx.x;
+    gs[b * C + c] = from_f32<T>(
+        sum * act_bwd_f(act, to_f32(s[b * C + c])));
+  }
+}
+
 template <typename scalar_t> struct DevT { using type = scalar_t; };
 template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct DevT<at::Half> { using type = _Float16; };
@@ -106,4 +176,52 @@ torch::Tensor act_bwd(torch::Tensor g, torch::Tensor x, int64_t act) {
   });
   HIP_CHECK_LAST();
   return gx;
+}
+
+torch::Tensor se_scale_fwd(torch::Tensor x, torch::Tensor s, int64_t act) {
+  TORCH_CHECK(x.dim() == 4 &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int C = (int)x.size(1);
+  int64_t hw = (int64_t)x.size(2) * x.size(3);
+  auto sc = s.reshape({x.size(0), C}).contiguous();
+  auto y = torch::empty_like(x);
+  int64_t n = x.numel();
+  NN_DISPATCH(x.scalar_type(), "se_scale_fwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    hipLaunchKernelGGL((se_scale_fwd_kernel<T>), dim3(grid_1d(n)),
+                       dim3(kBlock), 0, c10::hip::getCurrentHIPStream(),
+                       (const T*)x.data_ptr(), (const T*)sc.data_ptr(),
+                       (T*)y.data_ptr(), n, C, hw * C, (int)act);
+  });
+  HIP_CHECK_LAST();
+  return y;
+}
+
+std::vector<torch::Tensor> se_scale_bwd(torch::Tensor g, torch::Tensor x,
+                                        torch::Tensor s, int64_t act) {
+  TORCH_CHECK(x.dim() == 4 &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  auto gc = g.contiguous(at::MemoryFormat::ChannelsLast);
+  int64_t B = x.size(0);
+  int C = (int)x.size(1);
+  int64_t hw = (int64_t)x.size(2) * x.size(3);
+  auto sc = s.reshape({B, C}).contiguous();
+  auto gx = torch::empty_like(gc);
+  auto gs = torch::empty_like(sc);
+  int64_t n = x.numel();
+  NN_DISPATCH(x.scalar_type(), "se_scale_bwd", [&] {
+    using T = typename DevT<scalar_t>::type;
+    auto stream = c10::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL((se_scale_bwd_gx_kernel<T>), dim3(grid_1d(n)),
+                       dim3(kBlock), 0, stream, (const T*)gc.data_ptr(),
+                       (const T*)sc.data_ptr(), (T*)gx.data_ptr(), n, C,
+                       hw * C, (int)act);
+    hipLaunchKernelGGL((se_scale_bwd_gs_kernel<T>),
+                       dim3((unsigned)B, (C + 63) / 64), dim3(kBlock), 0,
+                       stream, (const T*)gc.data_ptr(),
+                       (const T*)x.data_ptr(), (const T*)sc.data_ptr(),
+                       (T*)gs.data_ptr(), hw, C, (int)act);
+  });
+  HIP_CHECK_LAST();
+  return {gx, gs};
 }

@@ -79,6 +79,9 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
 
 // activations.hip
 torch::Tensor act_fwd(torch::Tensor x, int64_t act);
+torch::Tensor se_scale_fwd(torch::Tensor x, torch::Tensor s, int64_t act);
+std::vector<torch::Tensor> se_scale_bwd(torch::Tensor g, torch::Tensor x,
+                                        torch::Tensor s, int64_t act);
 torch::Tensor act_bwd(torch::Tensor g, torch::Tensor x, int64_t act);
 
 // percentile.hip
@@ -155,6 +158,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_wgrad", &dwconv_wgrad);
   m.def("act_fwd", &act_fwd);
   m.def("act_bwd", &act_bwd);
+  m.def("se_scale_fwd", &se_scale_fwd);
+  m.def("se_scale_bwd", &se_scale_bwd);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_fwd", &avgpool_fwd);

@@ -69,6 +69,24 @@ def get_padding_value(padding, kernel_size, **kwargs):
     return padding, dynamic
 
 
+class NativeBatchNorm2d(nn.BatchNorm2d):
+    """nn.BatchNorm2d routed through the fused BN HIP kernel
+    (csrc/bn_act.hip: one stats pass + one finalize kernel instead of
+    MIOpen's 5-kernel chain). State-dict identical to nn.BatchNorm2d."""
+
+    def forward(self, x):
+        if self.weight is None:  # affine=False: rare, keep torch
+            return super().forward(x)
+        if self.training and self.track_running_stats \
+                and self.num_batches_tracked is not None:
+            self.num_batches_tracked.add_(1)
+        mom = self.momentum if self.momentum is not None else 0.1
+        return ops.bn_act(x, self.weight, self.bias, self.running_mean,
+                          self.running_var,
+                          self.training or not self.track_running_stats,
+                          mom, self.eps, relu=False, act_max=0.0)
+
+
 class NativeConv2d(nn.Conv2d):
     """nn.Conv2d whose forward dispatches to the MFMA / depthwise HIP
     kernels for the standard cases (state_dict-identical to nn.Conv2d)."""

@@ -20,7 +20,8 @@ import torch.nn as nn
 from .. import ops
 from .activations import HardSigmoid, HardSwish, Sigmoid, Swish
 from .adaptive_avgmax_pool import SelectAdaptivePool2d
-from .conv2d_layers import CondConv2d, NativeConv2d, select_conv2d
+from .conv2d_layers import (CondConv2d, NativeBatchNorm2d, NativeConv2d,
+                            select_conv2d)
 
 
 def make_divisible(v, divisor=8, min_value=None):
@@ -150,6 +151,11 @@ class SqueezeExcite(nn.Module):
         x_se = self.conv_reduce(x_se)
         x_se = self.act1(x_se)
         x_se = self.conv_expand(x_se)
+        if x.is_cuda:
+            # fused gate*broadcast-scale (csrc/activations.hip se_scale)
+            gate = ('hard_sigmoid'
+                    if isinstance(self.gate_fn, HardSigmoid) else 'sigmoid')
+            return ops.se_scale(x, x_se, gate)
         return x * self.gate_fn(x_se)
 
 
@@ -160,7 +166,7 @@ class ConvBnAct(nn.Module):
         norm_kwargs = norm_kwargs or {}
         self.conv = select_conv2d(in_chs, out_chs, kernel_size, stride=stride,
                                   dilation=dilation, padding=pad_type)
-        self.bn1 = nn.BatchNorm2d(out_chs, **norm_kwargs)
+        self.bn1 = NativeBatchNorm2d(out_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
 
     def forward(self, x):
@@ -182,14 +188,14 @@ class DepthwiseSeparableConv(nn.Module):
         self.conv_dw = select_conv2d(in_chs, in_chs, dw_kernel_size,
                                      stride=stride, dilation=dilation,
                                      padding=pad_type, depthwise=True)
-        self.bn1 = nn.BatchNorm2d(in_chs, **norm_kwargs)
+        self.bn1 = NativeBatchNorm2d(in_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
         if self.has_se:
             self.se = SqueezeExcite(in_chs, se_ratio=se_ratio,
                                     act_layer=act_layer)
         self.conv_pw = select_conv2d(in_chs, out_chs, pw_kernel_size,
                                      padding=pad_type)
-        self.bn2 = nn.BatchNorm2d(out_chs, **norm_kwargs)
+        self.bn2 = NativeBatchNorm2d(out_chs, **norm_kwargs)
         self.act2 = act_layer(inplace=True) if self.has_pw_act else nn.Identity()
 
     def forward(self, x):
@@ -228,13 +234,13 @@ class InvertedResidual(nn.Module):
 
         self.conv_pw = select_conv2d(in_chs, mid_chs, exp_kernel_size,
                                      padding=pad_type, **conv_kwargs)
-        self.bn1 = nn.BatchNorm2d(mid_chs, **norm_kwargs)
+        self.bn1 = NativeBatchNorm2d(mid_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
         self.conv_dw = select_conv2d(mid_chs, mid_chs, dw_kernel_size,
                                      stride=stride, dilation=dilation,
                                      padding=pad_type, depthwise=True,
                                      **conv_kwargs)
-        self.bn2 = nn.BatchNorm2d(mid_chs, **norm_kwargs)
+        self.bn2 = NativeBatchNorm2d(mid_chs, **norm_kwargs)
         self.act2 = act_layer(inplace=True)
         if self.has_se:
             self.se = SqueezeExcite(mid_chs, se_ratio=se_ratio,
@@ -242,7 +248,7 @@ class InvertedResidual(nn.Module):
                                     act_layer=act_layer)
         self.conv_pwl = select_conv2d(mid_chs, out_chs, pw_kernel_size,
                                       padding=pad_type, **conv_kwargs)
-        self.bn3 = nn.BatchNorm2d(out_chs, **norm_kwargs)
+        self.bn3 = NativeBatchNorm2d(out_chs, **norm_kwargs)
 
     def forward(self, x):
         residual = x
@@ -292,7 +298,7 @@ class EdgeResidual(nn.Module):
 
         self.conv_exp = select_conv2d(in_chs, mid_chs, exp_kernel_size,
                                       padding=pad_type)
-        self.bn1 = nn.BatchNorm2d(mid_chs, **norm_kwargs)
+        self.bn1 = NativeBatchNorm2d(mid_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
         if self.has_se:
             self.se = SqueezeExcite(mid_chs, se_ratio=se_ratio,
@@ -300,7 +306,7 @@ class EdgeResidual(nn.Module):
                                     act_layer=act_layer)
         self.conv_pwl = select_conv2d(mid_chs, out_chs, pw_kernel_size,
                                       stride=stride, padding=pad_type)
-        self.bn2 = nn.BatchNorm2d(out_chs, **norm_kwargs)
+        self.bn2 = NativeBatchNorm2d(out_chs, **norm_kwargs)
 
     def forward(self, x):
         residual = x
@@ -400,7 +406,7 @@ class EfficientNet(nn.Module):
                                    channel_divisor, channel_min)
         self.conv_stem = select_conv2d(in_chans, stem_size, 3, stride=2,
                                        padding=pad_type)
-        self.bn1 = nn.BatchNorm2d(stem_size, **norm_kwargs)
+        self.bn1 = NativeBatchNorm2d(stem_size, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
 
         builder = EfficientNetBuilder(channel_multiplier, channel_divisor,
@@ -411,7 +417,7 @@ class EfficientNet(nn.Module):
 
         self.conv_head = select_conv2d(head_chs, num_features, 1,
                                        padding=pad_type)
-        self.bn2 = nn.BatchNorm2d(num_features, **norm_kwargs)
+        self.bn2 = NativeBatchNorm2d(num_features, **norm_kwargs)
         self.act2 = act_layer(inplace=True)
         self.global_pool = SelectAdaptivePool2d(pool_type=global_pool)
         self.classifier = nn.Linear(num_features * self.global_pool.feat_mult(),

@@ -1119,3 +1119,39 @@ def per_sample_conv2d(x, weight, bias=None, stride=1, padding=0, groups=1):
     if bias is not None:
         out = out + bias.view(B, K, 1, 1)
     return out
+
+
+# ---------------------------------------------------------------------------
+# Squeeze-Excite gating (fused broadcast-scale + per-(b,c) reduce backward)
+# ---------------------------------------------------------------------------
+
+
+_SE_GATES = {'sigmoid': 4, 'hard_sigmoid': 3}
+
+
+class SeScale(torch.autograd.Function):
+    """y = x * gate(s) with per-(batch, channel) logits s [B,C,1,1].
+
+    Replaces the eager broadcast multiply + backward reduce of the SE
+    block (reference efficientnet SqueezeExcite, models/efficientnet.py:
+    449-466) with one fused pass each way."""
+
+    @staticmethod
+    def forward(ctx, x, s, gate_code):
+        ctx.gate_code = gate_code
+        ctx.save_for_backward(x, s)
+        return ext().se_scale_fwd(x, s, gate_code)
+
+    @staticmethod
+    def backward(ctx, g):
+        x, s = ctx.saved_tensors
+        gx, gs = ext().se_scale_bwd(g, x, s, ctx.gate_code)
+        return gx, gs.view_as(s), None
+
+
+def se_scale(x, s, gate='sigmoid'):
+    if use_native(x, s) and x.dim() == 4:
+        return SeScale.apply(_nhwc(x), s, _SE_GATES[gate])
+    gate_fn = torch.sigmoid if gate == 'sigmoid' \
+        else lambda t: F.hardsigmoid(t)
+    return x * gate_fn(s)

@@ -854,3 +854,58 @@ def test_conv_wgrad_patch_deterministic():
     a = ops.ext().conv_wgrad_patch(gy, x, 1, 0, 5, 5)
     b = ops.ext().conv_wgrad_patch(gy, x, 1, 0, 5, 5)
     assert torch.equal(a, b)
+
+
+def test_se_scale_matches_eager():
+    """Fused SE gating (fwd + both backward outputs) vs eager broadcast."""
+    torch.manual_seed(5)
+    for gate in ('sigmoid', 'hard_sigmoid'):
+        x = to_bf16_representable(torch.randn(4, 32, 7, 7))
+        s = to_bf16_representable(torch.randn(4, 32, 1, 1))
+        xg = cl(x.cuda().bfloat16()).requires_grad_(True)
+        sg = s.cuda().bfloat16().requires_grad_(True)
+        y = ops.se_scale(xg, sg, gate)
+        g = to_bf16_representable(torch.randn_like(y.cpu().float())).cuda().bfloat16()
+        y.backward(cl(g))
+
+        xe = x.float().requires_grad_(True)
+        se = s.float().requires_grad_(True)
+        gate_fn = torch.sigmoid if gate == 'sigmoid' else F.hardsigmoid
+        ye = xe * gate_fn(se)
+        ye.backward(g.float().cpu())
+
+        assert torch.allclose(y.float().cpu(), ye.detach(), atol=2e-2)
+        assert torch.allclose(xg.grad.float().cpu(), xe.grad, atol=2e-2)
+        assert torch.allclose(sg.grad.float().cpu(), se.grad,
+                              atol=0.05, rtol=0.05), gate
+
+
+def test_native_batchnorm2d_matches_torch():
+    """NativeBatchNorm2d (fused kernel) vs nn.BatchNorm2d: output, grads,
+    running stats, train and eval."""
+    from noisynet_amd.models.conv2d_layers import NativeBatchNorm2d
+    torch.manual_seed(6)
+    C = 24
+    ours = NativeBatchNorm2d(C).cuda()
+    ref = torch.nn.BatchNorm2d(C).cuda()
+    ref.load_state_dict(ours.state_dict())
+    x = torch.randn(8, C, 9, 9).cuda()
+    xo = cl(x.clone()).requires_grad_(True)
+    xr = x.clone().requires_grad_(True)
+    for train in (True, False):
+        ours.train(train)
+        ref.train(train)
+        yo = ours(xo)
+        yr = ref(xr)
+        assert torch.allclose(yo.float(), cl(yr).float(), atol=1e-4), train
+        yo.sum().backward()
+        yr.sum().backward()
+        assert torch.allclose(xo.grad.float(), cl(xr.grad).float(), atol=1e-4)
+        assert torch.allclose(ours.weight.grad, ref.weight.grad, atol=1e-3)
+        xo.grad = None
+        xr.grad = None
+        ours.zero_grad()
+        ref.zero_grad()
+    assert torch.allclose(ours.running_mean, ref.running_mean, atol=1e-5)
+    assert torch.allclose(ours.running_var, ref.running_var, atol=1e-5)
+    assert int(ours.num_batches_tracked) == int(ref.num_batches_tracked)
