@@ -130,8 +130,10 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
 
 def _conv_wgrad_raw(g, x, stride, padding, w_shape, col=None):
     if use_native(g, x):
+        import os
         R, S = w_shape[2], w_shape[3]
-        if col is not None and R * S > 1:
+        if col is not None and R * S > 1 \
+                and not os.environ.get("NOISYNET_WGRAD_NO_COL"):
             # flat im2col matrix shared from the forward pass
             return ext().conv_wgrad_from_col(_nhwc(g), col, x.shape[1], R, S)
         if R * S > 1 and x.element_size() == 2:
