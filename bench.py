@@ -69,8 +69,8 @@ def parse_args():
     args = p.parse_args()
     if args.batch == 0:
         # throughput-optimal per-GPU batch on 288 GB HBM3E (weak scaling
-        # keeps this fixed as N grows)
-        args.batch = 8192 if args.model == "noisynet" else 256
+        # keeps this fixed as N grows; 16384 measured +1.6% over 8192)
+        args.batch = 16384 if args.model == "noisynet" else 256
     return args
 
 

@@ -1,5 +1,11 @@
 import sqlite3, glob, re, collections, sys
-db_path = sorted(glob.glob(sys.argv[1] + '/runc/*_results.db'))[0]
+import os
+arg = sys.argv[1]
+cands = [arg] if os.path.isfile(arg) else (
+    sorted(glob.glob(arg + '/runc/*_results.db'))
+    or sorted(glob.glob(arg + '/*_results.db'))
+    or sorted(glob.glob(arg + '/**/*_results.db', recursive=True)))
+db_path = cands[0]
 db = sqlite3.connect(db_path); cur = db.cursor()
 t = [r[0] for r in cur.execute("SELECT name FROM sqlite_master WHERE type='table'") if r[0].startswith('rocpd_kernel_dispatch')][0]
 sfx = t[len('rocpd_kernel_dispatch_'):]

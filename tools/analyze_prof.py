@@ -2,7 +2,13 @@
 per-kernel totals for one steady-state step (delimited by sgd bursts)."""
 import sqlite3, glob, re, collections, sys
 
-db_path = sorted(glob.glob(sys.argv[1] + '/runc/*_results.db'))[0]
+import os
+arg = sys.argv[1]
+cands = [arg] if os.path.isfile(arg) else (
+    sorted(glob.glob(arg + '/runc/*_results.db'))
+    or sorted(glob.glob(arg + '/*_results.db'))
+    or sorted(glob.glob(arg + '/**/*_results.db', recursive=True)))
+db_path = cands[0]
 db = sqlite3.connect(db_path)
 cur = db.cursor()
 t = [r[0] for r in cur.execute(
