@@ -412,7 +412,7 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       // enough blocks to fill the chip; partial memory stays bounded
       int mslices = (int)std::min<int64_t>(
           (nchunks + pgrp - 1) / pgrp,
-          std::max<int64_t>(1, 512 / std::max(1, cblocks)));
+          std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
       int64_t rows = (int64_t)mslices * pgrp;
       parts = torch::zeros({rows, (int64_t)C * taps},
                            x.options().dtype(torch::kFloat32));
@@ -437,7 +437,7 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       int64_t nchunks = (npix + 7) / 8;
       int mslices = (int)std::min<int64_t>(
           (nchunks + pgrp - 1) / pgrp,
-          std::max<int64_t>(1, 512 / std::max(1, cblocks)));
+          std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
       int64_t rows = (int64_t)mslices * pgrp;
       parts = torch::zeros({rows, (int64_t)C * taps},
                            x.options().dtype(torch::kFloat32));
