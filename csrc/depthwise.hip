@@ -197,11 +197,12 @@ __global__ void dwconv_wgrad_vec_kernel(const T* __restrict__ gy,
                                         int R, int S, int stride, int pad,
                                         int64_t npix, int cspan_v, int pgrp) {
   using V = VecT<T, VEC>;
+  // dynamic LDS: [cspan_v][taps][VEC] f32 block accumulation region
+  extern __shared__ float lregion[];
   int cv_l = threadIdx.x % cspan_v;
   int grp = threadIdx.x / cspan_v;
-  if (grp >= pgrp) return;
   int cv = blockIdx.x * cspan_v + cv_l;
-  if (cv >= Cv) return;
+  bool active = (grp < pgrp) && (cv < Cv);
   int c0 = cv * VEC;
   int taps = R * S;
   float acc[MAXTAPS][VEC];
@@ -212,39 +213,57 @@ __global__ void dwconv_wgrad_vec_kernel(const T* __restrict__ gy,
 
   constexpr int CHUNK = 8;
   int64_t nchunks = (npix + CHUNK - 1) / CHUNK;
-  for (int64_t chunk = (int64_t)blockIdx.y * pgrp + grp; chunk < nchunks;
-       chunk += (int64_t)gridDim.y * pgrp) {
+  if (active) {
+    for (int64_t chunk = (int64_t)blockIdx.y * pgrp + grp; chunk < nchunks;
+         chunk += (int64_t)gridDim.y * pgrp) {
 #pragma unroll
-    for (int uu = 0; uu < CHUNK; ++uu) {
-      int64_t pix = chunk * CHUNK + uu;
-      if (pix >= npix) break;
-      int64_t t = pix;
-      int ow = (int)(t % OW);
-      t /= OW;
-      int oh = (int)(t % OH);
-      int64_t nb = t / OH;
-      V gv = *(const V*)&gy[pix * C + c0];
-      for (int r = 0; r < R; ++r) {
-        int ih = oh * stride - pad + r;
-        if (ih < 0 || ih >= H) continue;
-        for (int s = 0; s < S; ++s) {
-          int iw = ow * stride - pad + s;
-          if (iw < 0 || iw >= W) continue;
-          V xv = *(const V*)&x[((nb * H + ih) * W + iw) * C + c0];
+      for (int uu = 0; uu < CHUNK; ++uu) {
+        int64_t pix = chunk * CHUNK + uu;
+        if (pix >= npix) break;
+        int64_t t = pix;
+        int ow = (int)(t % OW);
+        t /= OW;
+        int oh = (int)(t % OH);
+        int64_t nb = t / OH;
+        V gv = *(const V*)&gy[pix * C + c0];
+        for (int r = 0; r < R; ++r) {
+          int ih = oh * stride - pad + r;
+          if (ih < 0 || ih >= H) continue;
+          for (int s = 0; s < S; ++s) {
+            int iw = ow * stride - pad + s;
+            if (iw < 0 || iw >= W) continue;
+            V xv = *(const V*)&x[((nb * H + ih) * W + iw) * C + c0];
 #pragma unroll
-          for (int u = 0; u < VEC; ++u)
-            acc[r * S + s][u] += to_f32(gv.v[u]) * to_f32(xv.v[u]);
+            for (int u = 0; u < VEC; ++u)
+              acc[r * S + s][u] += to_f32(gv.v[u]) * to_f32(xv.v[u]);
+          }
         }
       }
     }
   }
-  // one partial row per (blockIdx.y, grp): plain stores, no atomics
-  int64_t row = (int64_t)blockIdx.y * pgrp + grp;
-  float* out = partials + row * ((int64_t)C * taps);
-  for (int t = 0; t < taps; ++t)
+  // ordered in-block combine across the pixel groups (deterministic:
+  // group index order), so each BLOCK emits one partial row -- pgrp x
+  // fewer partial rows lets the grid grow to occupancy-filling sizes
+  // without exploding the partials reduce
+  for (int g2 = 0; g2 < pgrp; ++g2) {
+    if (grp == g2 && cv < Cv) {
+      for (int t = 0; t < taps; ++t)
 #pragma unroll
-    for (int u = 0; u < VEC; ++u)
-      out[(int64_t)(c0 + u) * taps + t] = acc[t][u];
+        for (int u = 0; u < VEC; ++u) {
+          int slot = (cv_l * taps + t) * VEC + u;
+          lregion[slot] = (g2 == 0 ? 0.0f : lregion[slot]) + acc[t][u];
+        }
+    }
+    __syncthreads();
+  }
+  if (grp == 0 && cv < Cv) {
+    float* out = partials + (int64_t)blockIdx.y * ((int64_t)C * taps);
+    for (int t = 0; t < taps; ++t)
+#pragma unroll
+      for (int u = 0; u < VEC; ++u)
+        out[(int64_t)(c0 + u) * taps + t] =
+            lregion[(cv_l * taps + t) * VEC + u];
+  }
 }
 
 template <typename T, int MAXTAPS>
@@ -409,27 +428,40 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       int pgrp = kBlock / cspan_v;
       int cblocks = (Cv + cspan_v - 1) / cspan_v;
       int64_t nchunks = (npix + 7) / 8;
-      // enough blocks to fill the chip; partial memory stays bounded
+      // one partial row PER BLOCK (in-block ordered combine), so the grid
+      // can grow to occupancy-filling sizes: at 512 blocks the kernel sat
+      // 86% memory-wait with ~2 waves/SIMD
       int mslices = (int)std::min<int64_t>(
           (nchunks + pgrp - 1) / pgrp,
           std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
-      int64_t rows = (int64_t)mslices * pgrp;
-      parts = torch::zeros({rows, (int64_t)C * taps},
+      parts = torch::zeros({(int64_t)mslices, (int64_t)C * taps},
                            x.options().dtype(torch::kFloat32));
-      if (vec)
-        hipLaunchKernelGGL((dwconv_wgrad_vec_kernel<T, VEC, 9>),
-                           dim3(cblocks, mslices), dim3(kBlock), 0, stream,
+      size_t lds = (size_t)cspan_v * taps * V * sizeof(float);
+      if (vec) {
+        auto* kfn = &dwconv_wgrad_vec_kernel<T, VEC, 9>;
+        if (lds > 64 * 1024)
+          hipFuncSetAttribute((const void*)kfn,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+        hipLaunchKernelGGL(kfn,
+                           dim3(cblocks, mslices), dim3(kBlock), lds, stream,
                            (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                            parts.data_ptr<float>(), Cv, C, H, W, OH, OW,
                            (int)R, (int)S, (int)stride, (int)pad, npix,
                            cspan_v, pgrp);
-      else
-        hipLaunchKernelGGL((dwconv_wgrad_vec_kernel<T, 4, 25>),
-                           dim3(cblocks, mslices), dim3(kBlock), 0, stream,
+      } else {
+        auto* kfn = &dwconv_wgrad_vec_kernel<T, 4, 25>;
+        if (lds > 64 * 1024)
+          hipFuncSetAttribute((const void*)kfn,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)lds);
+        hipLaunchKernelGGL(kfn,
+                           dim3(cblocks, mslices), dim3(kBlock), lds, stream,
                            (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                            parts.data_ptr<float>(), Cv, C, H, W, OH, OW,
                            (int)R, (int)S, (int)stride, (int)pad, npix,
                            cspan_v, pgrp);
+      }
     } else {
       int cspan = std::min(C, kBlock);
       int pgrp = kBlock / cspan;
@@ -437,7 +469,7 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       int64_t nchunks = (npix + 7) / 8;
       int mslices = (int)std::min<int64_t>(
           (nchunks + pgrp - 1) / pgrp,
-          std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
+          std::max<int64_t>(1, 512 / std::max(1, cblocks)));
       int64_t rows = (int64_t)mslices * pgrp;
       parts = torch::zeros({rows, (int64_t)C * taps},
                            x.options().dtype(torch::kFloat32));

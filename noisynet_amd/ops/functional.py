@@ -886,7 +886,10 @@ class ActFn(torch.autograd.Function):
     def backward(ctx, g):
         (x,) = ctx.saved_tensors
         if use_native(x):
-            return ext().act_bwd(g.contiguous(), x, _ACT_IDS[ctx.act]), None
+            # the C++ wrapper aligns g to x's memory format; a plain
+            # .contiguous() here forced an NCHW round-trip of every
+            # channels_last activation grad (5.3 ms/step on EffNet-B0)
+            return ext().act_bwd(g, x, _ACT_IDS[ctx.act]), None
         return _act_ref_bwd(g, x, ctx.act), None
 
 
@@ -932,7 +935,8 @@ class ReluClip(torch.autograd.Function):
     def backward(ctx, g):
         (y,) = ctx.saved_tensors
         if use_native(y):
-            return (ext().relu_clip_bwd(g.contiguous(), y, bool(ctx.relu),
+            # wrapper aligns g to y's memory format (no NCHW round-trip)
+            return (ext().relu_clip_bwd(g, y, bool(ctx.relu),
                                         float(ctx.act_max)), None, None)
         mask = torch.ones_like(y)
         if ctx.relu:
