@@ -428,12 +428,13 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
       int pgrp = kBlock / cspan_v;
       int cblocks = (Cv + cspan_v - 1) / cspan_v;
       int64_t nchunks = (npix + 7) / 8;
-      // one partial row PER BLOCK (in-block ordered combine), so the grid
-      // can grow to occupancy-filling sizes: at 512 blocks the kernel sat
-      // 86% memory-wait with ~2 waves/SIMD
+      // one partial row PER BLOCK (in-block ordered combine) keeps the
+      // partials reduce 21x smaller; 512 blocks measured best (2048
+      // thrashed L2: each block's 8-pixel tap windows are only reused
+      // while resident)
       int mslices = (int)std::min<int64_t>(
           (nchunks + pgrp - 1) / pgrp,
-          std::max<int64_t>(1, 2048 / std::max(1, cblocks)));
+          std::max<int64_t>(1, 512 / std::max(1, cblocks)));
       parts = torch::zeros({(int64_t)mslices, (int64_t)C * taps},
                            x.options().dtype(torch::kFloat32));
       size_t lds = (size_t)cspan_v * taps * V * sizeof(float);
