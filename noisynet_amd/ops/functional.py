@@ -123,6 +123,17 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
             dx = ext().conv_fwd_fused(_nhwc(g), w2, w2, empty, 1,
                                       R - 1 - padding, 0, zero_f, 0, False)[0]
             return dx
+        if R * S == 1 and stride == 1 and padding == 0:
+            # pointwise dgrad IS a GEMM: dx[M,C] = g[M,K] @ W[K,C]
+            gn = _nhwc(g)
+            K = gn.shape[1]
+            m = gn.shape[0] * gn.shape[2] * gn.shape[3]
+            g2 = gn.permute(0, 2, 3, 1).reshape(m, K)
+            dx = ext().linear_dgrad(g2.contiguous(),
+                                    w.reshape(w.shape[0], w.shape[1]))
+            return dx.view(gn.shape[0], gn.shape[2], gn.shape[3],
+                           w.shape[1]).permute(0, 3, 1, 2).contiguous(
+                               memory_format=torch.channels_last)
         return ext().conv_dgrad(_nhwc(g), _nhwc(w), stride, padding,
                                 x_shape[2], x_shape[3])
     return torch.nn.grad.conv2d_input(x_shape, w, g, stride, padding)
@@ -141,6 +152,19 @@ def _conv_wgrad_raw(g, x, stride, padding, w_shape, col=None):
             # input -- no materialized im2col pass at all
             return ext().conv_wgrad_patch(_nhwc(g), _nhwc(x), stride,
                                           padding, R, S)
+        if R * S == 1 and stride == 1 and padding == 0:
+            # pointwise conv wgrad IS a GEMM wgrad: flat NHWC views into
+            # the 128-deep pipelined mk kernels (the generic conv wgrad
+            # kernel was 2 ms/step of MobileNetV2's expand/project convs)
+            gn, xn = _nhwc(g), _nhwc(x)
+            K = gn.shape[1]
+            C_in = xn.shape[1]
+            m = gn.shape[0] * gn.shape[2] * gn.shape[3]
+            g2 = gn.permute(0, 2, 3, 1).reshape(m, K)
+            x2 = xn.permute(0, 2, 3, 1).reshape(m, C_in)
+            dw = ext().linear_wgrad(g2.contiguous(), x2.contiguous())
+            return dw.view(K, C_in, 1, 1).contiguous(
+                memory_format=torch.channels_last)
         if R * S > 1:
             # fp32: im2col-GEMM wgrad when the buffer is affordable (<2 GB)
             C_in = x.shape[1]
