@@ -266,6 +266,112 @@ __global__ void dwconv_wgrad_vec_kernel(const T* __restrict__ gy,
   }
 }
 
+// Sliding-window form for the dominant 3x3/stride-1 case: each thread
+// walks a whole OUTPUT ROW of one image, keeping the 3x3 input window of
+// VEC channels in registers -- 4 vector loads per output pixel (3 new
+// window columns + gy) instead of 10, and VEC=4 keeps the accumulator
+// bank at ~36 VGPRs so ~7 waves/SIMD hide the walk's load latency.
+// Same one-partial-row-per-block in-block ordered combine as the generic
+// vec kernel.
+template <typename T, int VEC>
+__global__ void dwconv_wgrad_win3_kernel(const T* __restrict__ gy,
+                                         const T* __restrict__ x,
+                                         float* __restrict__ partials,
+                                         int Cv, int C, int H, int W, int OH,
+                                         int OW, int pad, int64_t nrows,
+                                         int cspan_v, int pgrp) {
+  using V = VecT<T, VEC>;
+  extern __shared__ float lregion3[];
+  int cv_l = threadIdx.x % cspan_v;
+  int grp = threadIdx.x / cspan_v;
+  int cv = blockIdx.x * cspan_v + cv_l;
+  bool active = (grp < pgrp) && (cv < Cv);
+  int c0 = cv * VEC;
+  float acc[9][VEC];
+#pragma unroll
+  for (int t = 0; t < 9; ++t)
+#pragma unroll
+    for (int u = 0; u < VEC; ++u) acc[t][u] = 0.0f;
+
+  if (active) {
+    for (int64_t row = (int64_t)blockIdx.y * pgrp + grp; row < nrows;
+         row += (int64_t)gridDim.y * pgrp) {
+      int oh = (int)(row % OH);
+      int64_t nb = row / OH;
+      const T* xb = x + nb * ((int64_t)H * W) * C + c0;
+      const T* gb = gy + (row * OW) * C + c0;
+      int ih0 = oh - pad;  // input rows ih0 .. ih0+2
+      // window columns w0|w1|w2 = input cols (ow-1, ow, ow+1) x 3 rows
+      float w0[3][VEC], w1[3][VEC], w2[3][VEC];
+#pragma unroll
+      for (int r = 0; r < 3; ++r)
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) { w0[r][u] = 0.0f; w1[r][u] = 0.0f; }
+      // preload column for iw = 0 into w1 (iw = -1 stays zero in w0)
+#pragma unroll
+      for (int r = 0; r < 3; ++r) {
+        int ih = ih0 + r;
+        if (ih >= 0 && ih < H) {
+          V v = *(const V*)&xb[((int64_t)ih * W + 0) * C];
+#pragma unroll
+          for (int u = 0; u < VEC; ++u) w1[r][u] = to_f32(v.v[u]);
+        }
+      }
+      for (int ow = 0; ow < OW; ++ow) {
+        int iw2 = ow + 1;  // rightmost window column (pad=1)
+#pragma unroll
+        for (int r = 0; r < 3; ++r) {
+          int ih = ih0 + r;
+          if (ih >= 0 && ih < H && iw2 < W) {
+            V v = *(const V*)&xb[((int64_t)ih * W + iw2) * C];
+#pragma unroll
+            for (int u = 0; u < VEC; ++u) w2[r][u] = to_f32(v.v[u]);
+          } else {
+#pragma unroll
+            for (int u = 0; u < VEC; ++u) w2[r][u] = 0.0f;
+          }
+        }
+        V gv = *(const V*)&gb[(int64_t)ow * C];
+#pragma unroll
+        for (int r = 0; r < 3; ++r)
+#pragma unroll
+          for (int u = 0; u < VEC; ++u) {
+            float g = to_f32(gv.v[u]);
+            acc[r * 3 + 0][u] += g * w0[r][u];
+            acc[r * 3 + 1][u] += g * w1[r][u];
+            acc[r * 3 + 2][u] += g * w2[r][u];
+          }
+#pragma unroll
+        for (int r = 0; r < 3; ++r)
+#pragma unroll
+          for (int u = 0; u < VEC; ++u) {
+            w0[r][u] = w1[r][u];
+            w1[r][u] = w2[r][u];
+          }
+      }
+    }
+  }
+  // ordered in-block combine (see dwconv_wgrad_vec_kernel)
+  for (int g2 = 0; g2 < pgrp; ++g2) {
+    if (grp == g2 && cv < Cv) {
+      for (int t = 0; t < 9; ++t)
+#pragma unroll
+        for (int u = 0; u < VEC; ++u) {
+          int slot = (cv_l * 9 + t) * VEC + u;
+          lregion3[slot] = (g2 == 0 ? 0.0f : lregion3[slot]) + acc[t][u];
+        }
+    }
+    __syncthreads();
+  }
+  if (grp == 0 && cv < Cv) {
+    float* out = partials + (int64_t)blockIdx.y * ((int64_t)C * 9);
+    for (int t = 0; t < 9; ++t)
+#pragma unroll
+      for (int u = 0; u < VEC; ++u)
+        out[(int64_t)(c0 + u) * 9 + t] = lregion3[(cv_l * 9 + t) * VEC + u];
+  }
+}
+
 template <typename T, int MAXTAPS>
 __global__ void dwconv_wgrad_kernel(const T* __restrict__ gy,
                                     const T* __restrict__ x,
@@ -418,10 +524,34 @@ torch::Tensor dwconv_wgrad(torch::Tensor gy, torch::Tensor x, int64_t stride,
   // scalar form 4x on address math
   bool vec4 = !vec && (C % 4 == 0) && gy.element_size() == 2 && taps <= 25;
   torch::Tensor parts;
+  static const bool no_win3 = [] {
+    const char* e = getenv("NOISYNET_DW_NO_WIN3");
+    return e && e[0] == '1';
+  }();
+  bool win3 = !no_win3 && taps == 9 && stride == 1 && pad == 1 &&
+              (C % 4 == 0) && gy.element_size() == 2;
   NN_DISPATCH(gy.scalar_type(), "dwconv_wgrad", [&] {
     using T = typename DevT<scalar_t>::type;
     auto stream = c10::hip::getCurrentHIPStream();
-    if (vec || vec4) {
+    if (win3) {
+      constexpr int V4 = 4;
+      int Cv = C / V4;
+      int cspan_v = std::min(Cv, kBlock);
+      int pgrp = kBlock / cspan_v;
+      int cblocks = (Cv + cspan_v - 1) / cspan_v;
+      int64_t nrows = (int64_t)N * OH;
+      int mslices = (int)std::min<int64_t>(
+          (nrows + pgrp - 1) / pgrp,
+          std::max<int64_t>(1, 1024 / std::max(1, cblocks)));
+      parts = torch::zeros({(int64_t)mslices, (int64_t)C * 9},
+                           x.options().dtype(torch::kFloat32));
+      size_t lds = (size_t)cspan_v * 9 * V4 * sizeof(float);
+      hipLaunchKernelGGL((dwconv_wgrad_win3_kernel<T, V4>),
+                         dim3(cblocks, mslices), dim3(kBlock), lds, stream,
+                         (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                         parts.data_ptr<float>(), Cv, C, H, W, OH, OW,
+                         (int)pad, nrows, cspan_v, pgrp);
+    } else if (vec || vec4) {
       int V = vec ? VEC : 4;
       int Cv = C / V;
       int cspan_v = std::min(Cv, kBlock);
