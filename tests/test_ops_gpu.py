@@ -451,24 +451,27 @@ def test_model_learns_gpu_noisefree():
 
 def test_depthwise_conv_gpu_matches_torch():
     torch.manual_seed(13)
-    for stride, pad, k in ((1, 1, 3), (2, 2, 5)):
-        x = to_bf16_representable(torch.randn(4, 32, 14, 14)).to(dev()).bfloat16()
-        w = to_bf16_representable(torch.randn(32, 1, k, k) * 0.2).to(dev()).bfloat16()
+    # covers the win3 sliding-window wgrad (3x3/s1/p1), the 5x5 VEC4 path,
+    # the stride-2 vec path and an odd-channel scalar fallback
+    for stride, pad, k, C in ((1, 1, 3, 32), (2, 2, 5, 32), (2, 1, 3, 96),
+                              (1, 1, 3, 24), (1, 1, 3, 25)):
+        x = to_bf16_representable(torch.randn(4, C, 14, 14)).to(dev()).bfloat16()
+        w = to_bf16_representable(torch.randn(C, 1, k, k) * 0.2).to(dev()).bfloat16()
         y = ops.ext().dwconv_fwd(cl(x), w.contiguous(),
                                  torch.empty(0, device=dev(), dtype=x.dtype),
                                  stride, pad)
-        ye = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad, 1, 32)
+        ye = F.conv2d(x.float().cpu(), w.float().cpu(), None, stride, pad, 1, C)
         rel = (y.float().cpu() - ye).abs().max() / (ye.abs().max() + 1e-6)
         assert rel < 0.02, rel
         g = to_bf16_representable(torch.randn_like(ye)).to(dev()).bfloat16()
         dx = ops.ext().dwconv_dgrad(cl(g), w.contiguous(), stride, pad, 14, 14)
-        dxe = torch.nn.grad.conv2d_input((4, 32, 14, 14), w.float().cpu(),
-                                         g.float().cpu(), stride, pad, 1, 32)
+        dxe = torch.nn.grad.conv2d_input((4, C, 14, 14), w.float().cpu(),
+                                         g.float().cpu(), stride, pad, 1, C)
         rel = (dx.float().cpu() - dxe).abs().max() / (dxe.abs().max() + 1e-6)
         assert rel < 0.02, rel
         dw = ops.ext().dwconv_wgrad(cl(g), cl(x), stride, pad, k, k)
-        dwe = torch.nn.grad.conv2d_weight(x.float().cpu(), (32, 1, k, k),
-                                          g.float().cpu(), stride, pad, 1, 32)
+        dwe = torch.nn.grad.conv2d_weight(x.float().cpu(), (C, 1, k, k),
+                                          g.float().cpu(), stride, pad, 1, C)
         rel = (dw.float().cpu() - dwe).abs().max() / (dwe.abs().max() + 1e-6)
         assert rel < 0.03, rel
 
