@@ -484,3 +484,31 @@ def test_plot_surface_roundtrip(tmp_path):
                               os.path.join(dirs[0], 'array_names.npy'),
                               allow_pickle=True)), acc=50.0, tag='t')
     assert os.path.exists(out2)
+
+
+def test_tf_preprocessing_transform():
+    """TF-exact preprocessing without TF: padded center crop + bicubic
+    (eval) and distorted-bbox random crop (train), CHW uint8 out."""
+    import numpy as np
+    from PIL import Image
+
+    from noisynet_amd.timm.data.tf_preprocessing import (
+        TfPreprocessTransform, _center_crop_box)
+
+    img = Image.fromarray(
+        (np.random.RandomState(0).rand(300, 400, 3) * 255).astype('uint8'))
+    ev = TfPreprocessTransform(is_training=False, size=224)
+    out = ev(img)
+    assert out.shape == (3, 224, 224) and out.dtype == np.uint8
+    # the eval crop is the reference's size/(size+32)*min(h,w) center box
+    box = _center_crop_box(400, 300, 224)
+    assert box[2] - box[0] == box[3] - box[1] == int(224 / 256 * 300)
+    tr = TfPreprocessTransform(is_training=True, size=224)
+    out2 = tr(img)
+    assert out2.shape == (3, 224, 224)
+    # bytes input path (the reference feeds encoded jpeg bytes)
+    import io
+    buf = io.BytesIO()
+    img.save(buf, format='JPEG')
+    out3 = ev(buf.getvalue())
+    assert out3.shape == (3, 224, 224)
