@@ -110,7 +110,11 @@ def create_loader(dataset, input_size, batch_size, is_training=False,
     from .transforms import transforms_imagenet_eval, transforms_imagenet_train
     size = input_size[-1] if isinstance(input_size, (tuple, list)) else input_size
     if dataset.transform is None:
-        if is_training:
+        if tf_preprocessing:
+            from .tf_preprocessing import TfPreprocessTransform
+            dataset.transform = TfPreprocessTransform(
+                is_training=is_training, size=size)
+        elif is_training:
             dataset.transform = transforms_imagenet_train(img_size=size)
         else:
             dataset.transform = transforms_imagenet_eval(img_size=size)
