@@ -57,6 +57,12 @@ def train_one(current, epochs, X, y, Xt, yt, device, seed=11, batch=64):
     curve = []
     step = 0
     for epoch in range(epochs):
+        # step-decay schedule (the reference trains with LR decay,
+        # noisynet.py:1176-1231; constant LR left late epochs unstable)
+        if epoch in (int(epochs * 0.5), int(epochs * 0.75),
+                     int(epochs * 0.9)):
+            for group in opt.param_groups:
+                group['lr'] *= 0.2
         model.train()
         perm = torch.randperm(n, device=device)
         for i in range(n // batch):
