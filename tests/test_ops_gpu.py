@@ -951,8 +951,8 @@ def test_condconv_per_sample_gpu():
     w = torch.matmul(rw.float(), m.weight.float()).view(
         B * m.out_channels, 16, 3, 3)
     b = torch.matmul(rw.float(), m.bias.float()).view(B * m.out_channels)
-    ref = F.conv2d(x.detach().float().view(1, B * 16, 10, 10), w, b,
-                   stride=1, padding=1, groups=B)
+    ref = F.conv2d(x.detach().float().contiguous().view(1, B * 16, 10, 10),
+                   w, b, stride=1, padding=1, groups=B)
     ref = ref.permute([1, 0, 2, 3]).reshape(B, 24, 10, 10)
     rel = (out.float() - ref).abs().max() / (ref.abs().max() + 1e-6)
     assert rel < 0.05, rel.item()
