@@ -96,10 +96,15 @@ def _patch_eligible(x, w, padding):
             and x.shape[2] * wp * c_pad * eb + 128 <= 96 * 1024)
 
 
+import os as _os
+
+_PW_GEMM = _os.environ.get("NOISYNET_PW_GEMM", "1") == "1"
+
+
 def _conv_fwd_raw(x, w, bias, stride, padding):
     if use_native(x, w):
         R, S = w.shape[2], w.shape[3]
-        if R * S == 1 and stride == 1 and padding == 0 \
+        if _PW_GEMM and R * S == 1 and stride == 1 and padding == 0 \
                 and x.element_size() == 2:
             # pointwise conv IS a plain GEMM: rocBLAS bf16 (the streaming
             # conv kernel ran these short-contraction shapes at ~1% of
@@ -139,12 +144,16 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
                                       R - 1 - padding, 0, zero_f, 0, False)[0]
             return dx
         if R * S == 1 and stride == 1 and padding == 0:
-            # pointwise dgrad IS a GEMM: dx[M,C] = g[M,K] @ W[K,C] -> rocBLAS
+            # pointwise dgrad IS a GEMM: dx[M,C] = g[M,K] @ W[K,C]
             gn = _nhwc(g)
             K = gn.shape[1]
             m = gn.shape[0] * gn.shape[2] * gn.shape[3]
             g2 = gn.permute(0, 2, 3, 1).reshape(m, K)
-            dx = torch.matmul(g2, w.reshape(w.shape[0], w.shape[1]))
+            if _PW_GEMM:
+                dx = torch.matmul(g2, w.reshape(w.shape[0], w.shape[1]))
+            else:
+                dx = ext().linear_dgrad(g2.contiguous(),
+                                        w.reshape(w.shape[0], w.shape[1]))
             return dx.view(gn.shape[0], gn.shape[2], gn.shape[3],
                            w.shape[1]).permute(0, 3, 1, 2).contiguous(
                                memory_format=torch.channels_last)
