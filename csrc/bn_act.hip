@@ -306,7 +306,7 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x) {
   int ngrp = kBlock / span;  // first block's packing
   int rper = vec16 ? 2 : 4;
   int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
-                                  std::max(1, 2048 / gx));
+                                  std::max(1, 1024 / gx));
   auto partial_sum = torch::empty({gy, C}, opts);
   auto partial_sumsq = torch::empty({gy, C}, opts);
   auto sum = torch::empty({C}, opts);
@@ -633,7 +633,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor g, torch::Tensor x,
   int ngrp = kBlock / span;  // first block's packing
   int rper = vec16 ? 2 : 4;
   int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
-                                  std::max(1, 2048 / gx_blocks));
+                                  std::max(1, 1024 / gx_blocks));
   auto partial_g = torch::empty({gy, C}, opts);
   auto partial_gx = torch::empty({gy, C}, opts);
   int64_t n = x.numel();
@@ -731,7 +731,7 @@ std::vector<torch::Tensor> bn_act_bwd_reduce(torch::Tensor g, torch::Tensor x,
   int ngrp = kBlock / span;  // first block's packing
   int rper = vec16 ? 2 : 4;
   int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
-                                  std::max(1, 2048 / gx_blocks));
+                                  std::max(1, 1024 / gx_blocks));
   auto partial_g = torch::empty({gy, C}, opts);
   auto partial_gx = torch::empty({gy, C}, opts);
   auto stream = c10::hip::getCurrentHIPStream();
@@ -854,7 +854,7 @@ std::vector<torch::Tensor> bn_stats_finalize(torch::Tensor x,
   int ngrp = kBlock / span;  // first block's packing
   int rper = vec16 ? 2 : 4;
   int gy = (int)std::min<int64_t>((rows + rper * ngrp - 1) / (rper * ngrp),
-                                  std::max(1, 2048 / gx));
+                                  std::max(1, 1024 / gx));
   auto partial_sum = torch::empty({gy, C}, opts);
   auto partial_sumsq = torch::empty({gy, C}, opts);
   auto stream = c10::hip::getCurrentHIPStream();

@@ -98,7 +98,7 @@ def _patch_eligible(x, w, padding):
 
 import os as _os
 
-_PW_GEMM = _os.environ.get("NOISYNET_PW_GEMM", "1") == "1"
+_PW_GEMM = _os.environ.get("NOISYNET_PW_GEMM", "0") == "1"  # A/B: custom kernels beat rocBLAS at these shapes (MNv2 8.48k vs 8.10k)
 
 
 def _conv_fwd_raw(x, w, bias, stride, padding):
