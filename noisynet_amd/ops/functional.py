@@ -18,6 +18,8 @@ Autograd structure mirrors the reference semantics:
     noisynet.py:1348-1476) work through the custom kernels.
 """
 
+import os as _os
+
 import torch
 import torch.nn.functional as F
 
@@ -96,8 +98,6 @@ def _patch_eligible(x, w, padding):
             and x.shape[2] * wp * c_pad * eb + 128 <= 96 * 1024)
 
 
-import os as _os
-
 _PW_GEMM = _os.environ.get("NOISYNET_PW_GEMM", "0") == "1"  # A/B: custom kernels beat rocBLAS at these shapes (MNv2 8.48k vs 8.10k)
 
 
@@ -164,10 +164,9 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
 
 def _conv_wgrad_raw(g, x, stride, padding, w_shape, col=None):
     if use_native(g, x):
-        import os
         R, S = w_shape[2], w_shape[3]
         if col is not None and R * S > 1 \
-                and not os.environ.get("NOISYNET_WGRAD_NO_COL"):
+                and not _os.environ.get("NOISYNET_WGRAD_NO_COL"):
             # flat im2col matrix shared from the forward pass
             return ext().conv_wgrad_from_col(_nhwc(g), col, x.shape[1], R, S)
         if R * S > 1 and x.element_size() == 2:
