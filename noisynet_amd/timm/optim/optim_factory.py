@@ -80,6 +80,24 @@ def create_optimizer(args, model, filter_bias_and_bn=True):
     elif opt_lower == 'nvnovograd':
         optimizer = NvNovoGrad(parameters, lr=args.lr,
                                weight_decay=weight_decay, eps=args.opt_eps)
+    # apex Fused* names (reference optim_factory.py:75-97): this framework's
+    # SGD/Adam/AdamW ARE fused-kernel updates (csrc/optimizer.hip), so the
+    # names resolve to them instead of requiring apex
+    elif opt_lower == 'fusedsgd':
+        optimizer = native_optim.SGD(parameters, lr=args.lr,
+                                     momentum=args.momentum,
+                                     weight_decay=weight_decay, nesterov=True)
+    elif opt_lower == 'fusedadam':
+        optimizer = native_optim.Adam(parameters, lr=args.lr,
+                                      weight_decay=weight_decay,
+                                      eps=args.opt_eps)
+    elif opt_lower == 'fusedadamw':
+        optimizer = native_optim.AdamW(parameters, lr=args.lr,
+                                       weight_decay=weight_decay,
+                                       eps=args.opt_eps)
+    elif opt_lower == 'fusednovograd':
+        optimizer = NvNovoGrad(parameters, lr=args.lr,
+                               weight_decay=weight_decay, eps=args.opt_eps)
     else:
         raise ValueError("Invalid optimizer %s" % args.opt)
 

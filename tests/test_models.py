@@ -512,3 +512,26 @@ def test_tf_preprocessing_transform():
     img.save(buf, format='JPEG')
     out3 = ev(buf.getvalue())
     assert out3.shape == (3, 224, 224)
+
+
+def test_optim_factory_fused_names():
+    """apex Fused* names resolve to the native fused-kernel optimizers
+    (reference optim_factory.py:75-97 requires apex; ours ARE fused)."""
+    import types
+
+    import torch
+
+    from noisynet_amd.timm.optim.optim_factory import create_optimizer
+
+    m = torch.nn.Linear(4, 4)
+    expected = {'fusedsgd': 'SGD', 'fusedadam': 'Adam',
+                'fusedadamw': 'AdamW', 'fusednovograd': 'NvNovoGrad'}
+    for name, cls in expected.items():
+        args = types.SimpleNamespace(opt=name, weight_decay=1e-4, lr=0.01,
+                                     momentum=0.9, opt_eps=1e-8)
+        opt = create_optimizer(args, m)
+        assert type(opt).__name__ == cls, name
+        loss = m(torch.randn(2, 4)).sum()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
