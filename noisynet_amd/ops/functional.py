@@ -98,25 +98,22 @@ def _patch_eligible(x, w, padding):
             and x.shape[2] * wp * c_pad * eb + 128 <= 96 * 1024)
 
 
-_PW_GEMM = _os.environ.get("NOISYNET_PW_GEMM", "0") == "1"  # A/B: custom kernels beat rocBLAS at these shapes (MNv2 8.48k vs 8.10k)
-
-
 def _conv_fwd_raw(x, w, bias, stride, padding):
     if use_native(x, w):
         R, S = w.shape[2], w.shape[3]
-        if _PW_GEMM and R * S == 1 and stride == 1 and padding == 0 \
-                and x.element_size() == 2:
-            # pointwise conv IS a plain GEMM: rocBLAS bf16 (the streaming
-            # conv kernel ran these short-contraction shapes at ~1% of
-            # MFMA peak -- 3.1 ms of a MobileNetV2 step)
+        if R * S == 1 and stride == 1 and padding == 0 \
+                and x.element_size() == 2 and x.shape[1] % 8 == 0:
+            # pointwise conv IS a plain GEMM -> the lean 128x128 gemm_xwt
+            # kernel (the streaming conv kernel ran these short-contraction
+            # shapes at ~1% of MFMA peak; rocBLAS measured slower too)
             xn = _nhwc(x)
             B, C = xn.shape[0], xn.shape[1]
             OH, OW = xn.shape[2], xn.shape[3]
             x2 = xn.permute(0, 2, 3, 1).reshape(B * OH * OW, C)
-            y2 = torch.matmul(x2, w.reshape(w.shape[0], C).t())
+            empty_b = (bias.to(torch.float32) if bias is not None else
+                       torch.empty(0, device=x.device, dtype=torch.float32))
+            y2 = ext().gemm_xwt(x2, w.reshape(w.shape[0], C), empty_b)
             y = y2.view(B, OH, OW, w.shape[0]).permute(0, 3, 1, 2)
-            if bias is not None:
-                y = y + bias.view(1, -1, 1, 1)
             return y.contiguous(memory_format=torch.channels_last)
         if _patch_eligible(x, w, padding):
             empty = torch.empty(0, device=x.device, dtype=x.dtype)
@@ -149,11 +146,8 @@ def _conv_dgrad_raw(g, w, stride, padding, x_shape):
             K = gn.shape[1]
             m = gn.shape[0] * gn.shape[2] * gn.shape[3]
             g2 = gn.permute(0, 2, 3, 1).reshape(m, K)
-            if _PW_GEMM:
-                dx = torch.matmul(g2, w.reshape(w.shape[0], w.shape[1]))
-            else:
-                dx = ext().linear_dgrad(g2.contiguous(),
-                                        w.reshape(w.shape[0], w.shape[1]))
+            dx = ext().linear_dgrad(g2.contiguous(),
+                                    w.reshape(w.shape[0], w.shape[1]))
             return dx.view(gn.shape[0], gn.shape[2], gn.shape[3],
                            w.shape[1]).permute(0, 3, 1, 2).contiguous(
                                memory_format=torch.channels_last)
