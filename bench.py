@@ -178,6 +178,12 @@ def run_top1(bench, device, dtype, rank, distributed):
     start_calibration(model)
     curve = []
     for epoch in range(bench.top1):
+        # step LR decay, as in tools/accuracy_curve.py (constant LR left
+        # late epochs unstable)
+        if epoch in (int(bench.top1 * 0.5), int(bench.top1 * 0.75),
+                     int(bench.top1 * 0.9)):
+            for group in optimizer.param_groups:
+                group['lr'] *= 0.2
         model.train()
         perm = torch.randperm(n, device=device)
         for i in range(n // bs):
