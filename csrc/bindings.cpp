@@ -110,7 +110,6 @@ torch::Tensor conv_wgrad_im2col(torch::Tensor gy, torch::Tensor x,
 torch::Tensor conv_wgrad_patch(torch::Tensor gy, torch::Tensor x,
                                int64_t stride, int64_t pad, int64_t R,
                                int64_t S);
-torch::Tensor gemm_xwt(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w);
 torch::Tensor linear_dgrad(torch::Tensor gy, torch::Tensor w);
 torch::Tensor linear_wgrad(torch::Tensor gy, torch::Tensor x);
@@ -174,7 +173,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_wgrad_patch", &conv_wgrad_patch);
   m.def("conv_wgrad_from_col", &conv_wgrad_from_col);
   m.def("im2col_materialize", &im2col_materialize);
-  m.def("gemm_xwt", &gemm_xwt);
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_dgrad", &linear_dgrad);
   m.def("linear_wgrad", &linear_wgrad);

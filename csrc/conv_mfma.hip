@@ -1110,97 +1110,6 @@ void wgrad_mk4_kernel(const T* __restrict__ gy, const T* __restrict__ x,
 }
 
 // --------------------------------------------------------------------------
-// Lean GEMM: out[M, K] = X[M, C] @ W[K, C]^T, 128x128 tiles, 128-deep
-// contraction rounds over C. Serves pointwise (1x1) conv forward and the
-// plain linear forward/dgrad -- the generic streaming conv kernel ran
-// those short-contraction shapes at ~1% of MFMA peak (3+ ms of a
-// MobileNetV2 step). Both operands are [out-row, contraction] row-major,
-// so staging is a straight swizzled copy (no transpose scatter).
-// --------------------------------------------------------------------------
-template <typename T>
-DEV_INLINE void gemm_xwt_stage(char* lds, const T* __restrict__ src,
-                               int64_t row0, int64_t rows, int C, int c0) {
-#pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    int idx = threadIdx.x + i * kBlock;  // 0..2047
-    int row = idx >> 4;                  // 0..127
-    int seg = idx & 15;                  // 16-B chunk of the 128-col row
-    int c = c0 + seg * 8;
-    int64_t r = row0 + row;
-    T vals[8];
-    if (r < rows && c + 8 <= C) {
-      *(bf16x8*)vals = *(const bf16x8*)(src + r * C + c);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vals[j] = (r < rows && c + j < C) ? src[r * C + c + j]
-                                          : from_f32<T>(0.0f);
-    }
-    *(bf16x8*)(lds + row * WG_LSTR + wg_swz(row, seg) * 16) = *(bf16x8*)vals;
-  }
-}
-
-template <typename T>
-__global__ __launch_bounds__(kBlock)
-void gemm_xwt_kernel(const T* __restrict__ X, const T* __restrict__ W,
-                     const float* __restrict__ bias, T* __restrict__ out,
-                     int64_t M, int K, int C) {
-  int k0 = blockIdx.x * 128;
-  int64_t m0 = (int64_t)blockIdx.y * 128;
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* a_lds = smem;                        // [128 m rows][128 c]
-  char* b_lds = smem + 128 * WG_LSTR;        // [128 k rows][128 c]
-
-  int wid = threadIdx.x / WAVE;
-  int wm = wid >> 1, wn = wid & 1;
-  int lane = threadIdx.x & (WAVE - 1);
-  f32x4 acc[4][4] = {};
-
-  int nrounds = (C + WG_BK - 1) / WG_BK;
-  for (int cr = 0; cr < nrounds; ++cr) {
-    int c0 = cr * WG_BK;
-    gemm_xwt_stage(a_lds, X, m0, M, C, c0);
-    gemm_xwt_stage(b_lds, W, k0, (int64_t)K, C, c0);
-    __syncthreads();
-    int ks_max = (C - c0 + 31) >> 5;
-    if (ks_max > 4) ks_max = 4;
-    for (int ks = 0; ks < ks_max; ++ks) {
-#pragma unroll
-      for (int fm = 0; fm < 4; ++fm) {
-        int arow = wm * 64 + fm * 16 + (lane & 15);
-        auto a = *(typename Mma<T>::frag*)(
-            a_lds + arow * WG_LSTR +
-            wg_swz(arow, ks * 4 + (lane >> 4)) * 16);
-#pragma unroll
-        for (int fn = 0; fn < 4; ++fn) {
-          int brow = wn * 64 + fn * 16 + (lane & 15);
-          auto b = *(typename Mma<T>::frag*)(
-              b_lds + brow * WG_LSTR +
-              wg_swz(brow, ks * 4 + (lane >> 4)) * 16);
-          Mma<T>::mma(a, b, acc[fm][fn]);
-        }
-      }
-    }
-    __syncthreads();
-  }
-
-#pragma unroll
-  for (int fm = 0; fm < 4; ++fm)
-#pragma unroll
-    for (int fn = 0; fn < 4; ++fn)
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        int64_t m = m0 + wm * 64 + fm * 16 + 4 * (lane >> 4) + reg;
-        int k = k0 + wn * 64 + fn * 16 + (lane & 15);
-        if (m < M && k < K) {
-          float v = acc[fm][fn][reg];
-          if (bias != nullptr) v += bias[k];
-          out[m * K + k] = from_f32<T>(v);
-        }
-      }
-}
-
-// --------------------------------------------------------------------------
 // Patch wgrad: the im2col matrix is never materialized. The x operand of
 // the 128x128 wgrad GEMM is gathered straight from the (channel-padded)
 // NHWC input: column index (tap r,s | channel c) decodes to the 16-B
@@ -1522,44 +1431,8 @@ ConvGeom linear_geom(const torch::Tensor& x, int64_t out_features) {
 }
 }  // namespace
 
-// out[M, K] = X @ W^T with the lean 128x128 GEMM kernel (16-bit, C % 8)
-torch::Tensor gemm_xwt(torch::Tensor x, torch::Tensor w, torch::Tensor bias) {
-  TORCH_CHECK(x.dim() == 2 && x.is_contiguous());
-  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
-  TORCH_CHECK(x.size(1) == w.size(1), "gemm_xwt: inner dims");
-  TORCH_CHECK(x.element_size() == 2 && (x.size(1) % 8) == 0,
-              "gemm_xwt: 16-bit, C % 8 == 0");
-  int64_t M = x.size(0);
-  int K = (int)w.size(0);
-  int C = (int)x.size(1);
-  bool has_bias = bias.numel() > 0;
-  torch::Tensor bias_f;
-  if (has_bias) bias_f = bias.to(torch::kFloat32).contiguous();
-  auto out = torch::empty({M, (int64_t)K}, x.options());
-  dim3 grid((K + 127) / 128, (unsigned)((M + 127) / 128));
-  size_t lds = (size_t)256 * WG_LSTR;
-  NN_DISPATCH(x.scalar_type(), "gemm_xwt", [&] {
-    using T = typename DevT<scalar_t>::type;
-    auto* kfn = &gemm_xwt_kernel<T>;
-    if (lds > 64 * 1024)
-      hipFuncSetAttribute((const void*)kfn,
-                          hipFuncAttributeMaxDynamicSharedMemorySize,
-                          (int)lds);
-    hipLaunchKernelGGL(kfn, grid, dim3(kBlock), lds,
-                       c10::hip::getCurrentHIPStream(),
-                       (const T*)x.data_ptr(), (const T*)w.data_ptr(),
-                       has_bias ? bias_f.data_ptr<float>() : nullptr,
-                       (T*)out.data_ptr(), M, K, C);
-  });
-  HIP_CHECK_LAST();
-  return out;
-}
-
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
-  if (x.element_size() == 2 && (x.size(1) % 8) == 0 && x.is_contiguous()) {
-    return gemm_xwt(x, w, torch::empty({0}, x.options()));
-  }
   auto g = linear_geom(x, w.size(0));
   auto out = torch::empty({x.size(0), w.size(0)}, x.options());
   dim3 grid((g.K + BN - 1) / BN, (int)((g.M + BM - 1) / BM));

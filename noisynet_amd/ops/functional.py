@@ -101,20 +101,6 @@ def _patch_eligible(x, w, padding):
 def _conv_fwd_raw(x, w, bias, stride, padding):
     if use_native(x, w):
         R, S = w.shape[2], w.shape[3]
-        if R * S == 1 and stride == 1 and padding == 0 \
-                and x.element_size() == 2 and x.shape[1] % 8 == 0:
-            # pointwise conv IS a plain GEMM -> the lean 128x128 gemm_xwt
-            # kernel (the streaming conv kernel ran these short-contraction
-            # shapes at ~1% of MFMA peak; rocBLAS measured slower too)
-            xn = _nhwc(x)
-            B, C = xn.shape[0], xn.shape[1]
-            OH, OW = xn.shape[2], xn.shape[3]
-            x2 = xn.permute(0, 2, 3, 1).reshape(B * OH * OW, C)
-            empty_b = (bias.to(torch.float32) if bias is not None else
-                       torch.empty(0, device=x.device, dtype=torch.float32))
-            y2 = ext().gemm_xwt(x2, w.reshape(w.shape[0], C), empty_b)
-            y = y2.view(B, OH, OW, w.shape[0]).permute(0, 3, 1, 2)
-            return y.contiguous(memory_format=torch.channels_last)
         if _patch_eligible(x, w, padding):
             empty = torch.empty(0, device=x.device, dtype=x.dtype)
             zero_f = torch.zeros(1, device=x.device, dtype=torch.float32)
