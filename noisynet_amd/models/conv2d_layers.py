@@ -36,6 +36,8 @@ def conv2d_same(x, weight, bias=None, stride=(1, 1), padding=(0, 0),
                       pad_h // 2, pad_h - pad_h // 2])
     if groups == 1 and dilation == (1, 1) and pad_h % 2 == 0 and pad_w % 2 == 0:
         return ops.conv2d(x, weight, bias, stride[0], 0)
+    # documented exception: dilated / grouped-non-depthwise SAME convs are
+    # outside the model zoo's hot paths (no registered model hits this)
     return F.conv2d(x, weight, bias, stride, (0, 0), dilation, groups)
 
 
@@ -99,6 +101,8 @@ class NativeConv2d(nn.Conv2d):
                 and self.dilation == (1, 1)):
             return ops.depthwise_conv2d(x, self.weight, self.bias,
                                         self.stride, self.padding)
+        # documented exception: dilation / grouped-non-depthwise -- not on
+        # any registered model's hot path
         return F.conv2d(x, self.weight, self.bias, self.stride, self.padding,
                         self.dilation, self.groups)
 
