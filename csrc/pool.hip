@@ -11,9 +11,6 @@ namespace {
 constexpr int kBlock = 256;
 
 template <typename T>
-struct alignas(16) PoolPack8 { T v[8]; };
-
-template <typename T>
 __global__ void maxpool2x2_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                       uint8_t* __restrict__ code, int64_t n_out,
                                       int C, int H, int W, int OH, int OW) {
@@ -94,7 +91,7 @@ DEV_INLINE void flat_g2l(T* dst, const T* __restrict__ src, int64_t e0,
        e += (int64_t)blockDim.x * 8) {
     if (e >= e0 && e + 8 <= e1) {
       T vals[8];
-      *(PoolPack8<T>*)vals = *(const PoolPack8<T>*)(src + e);
+      *(bf16x8*)vals = *(const bf16x8*)(src + e);
 #pragma unroll
       for (int j = 0; j < 8; ++j) dst[e - e0 + j] = vals[j];
     } else {
@@ -117,7 +114,7 @@ DEV_INLINE void flat_l2g(T* __restrict__ dst, const T* src, int64_t e0,
       T vals[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) vals[j] = src[e - e0 + j];
-      *(PoolPack8<T>*)(dst + e) = *(PoolPack8<T>*)vals;
+      *(bf16x8*)(dst + e) = *(bf16x8*)vals;
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
