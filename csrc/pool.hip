@@ -74,187 +74,6 @@ __global__ void maxpool2x2_bwd_kernel(const T* __restrict__ g,
   }
 }
 
-// ---------------------------------------------------------------------------
-// LDS-staged 2x2 pool (16-bit, even H and W). Odd C makes every direct
-// NHWC access 2-B-misaligned, so the simple kernels issue scalar loads
-// (44% instruction-wait, ~3x off bandwidth). Each block copies a
-// G-output-row slab: the slab's FLAT global ranges are moved with aligned
-// 16-B vectors; the pooling itself runs LDS->LDS where alignment is free.
-// ---------------------------------------------------------------------------
-
-template <typename T>
-DEV_INLINE void flat_g2l(T* dst, const T* __restrict__ src, int64_t e0,
-                         int64_t e1) {
-  // global [e0, e1) -> LDS dst[0 ..), 16-B window-aligned
-  int64_t a0 = e0 & ~(int64_t)7;
-  for (int64_t e = a0 + (int64_t)threadIdx.x * 8; e < e1;
-       e += (int64_t)blockDim.x * 8) {
-    if (e >= e0 && e + 8 <= e1) {
-      T vals[8];
-      *(bf16x8*)vals = *(const bf16x8*)(src + e);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[e - e0 + j] = vals[j];
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int64_t ee = e + j;
-        if (ee >= e0 && ee < e1) dst[ee - e0] = src[ee];
-      }
-    }
-  }
-}
-
-template <typename T>
-DEV_INLINE void flat_l2g(T* __restrict__ dst, const T* src, int64_t e0,
-                         int64_t e1) {
-  int64_t a0 = e0 & ~(int64_t)7;
-  for (int64_t e = a0 + (int64_t)threadIdx.x * 8; e < e1;
-       e += (int64_t)blockDim.x * 8) {
-    if (e >= e0 && e + 8 <= e1) {
-      T vals[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vals[j] = src[e - e0 + j];
-      *(bf16x8*)(dst + e) = *(bf16x8*)vals;
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int64_t ee = e + j;
-        if (ee >= e0 && ee < e1) dst[ee] = src[ee - e0];
-      }
-    }
-  }
-}
-
-DEV_INLINE void flat_l2g_u8(uint8_t* __restrict__ dst, const uint8_t* src,
-                            int64_t e0, int64_t e1) {
-  int64_t a0 = e0 & ~(int64_t)15;
-  for (int64_t e = a0 + (int64_t)threadIdx.x * 16; e < e1;
-       e += (int64_t)blockDim.x * 16) {
-    if (e >= e0 && e + 16 <= e1) {
-      uint8_t vals[16];
-#pragma unroll
-      for (int j = 0; j < 16; ++j) vals[j] = src[e - e0 + j];
-      *(uint4*)(dst + e) = *(uint4*)vals;
-    } else {
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        int64_t ee = e + j;
-        if (ee >= e0 && ee < e1) dst[ee] = src[ee - e0];
-      }
-    }
-  }
-}
-
-DEV_INLINE void flat_g2l_u8(uint8_t* dst, const uint8_t* __restrict__ src,
-                            int64_t e0, int64_t e1) {
-  int64_t a0 = e0 & ~(int64_t)15;
-  for (int64_t e = a0 + (int64_t)threadIdx.x * 16; e < e1;
-       e += (int64_t)blockDim.x * 16) {
-    if (e >= e0 && e + 16 <= e1) {
-      uint4 v = *(const uint4*)(src + e);
-      *(uint4*)(dst + (e - e0)) = v;
-    } else {
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        int64_t ee = e + j;
-        if (ee >= e0 && ee < e1) dst[ee - e0] = src[ee];
-      }
-    }
-  }
-}
-
-template <typename T>
-__global__ void maxpool2x2_fwd_lds_kernel(const T* __restrict__ x,
-                                          T* __restrict__ y,
-                                          uint8_t* __restrict__ code, int C,
-                                          int W, int OH, int OW, int G,
-                                          int slabs_per_img, int64_t nslabs) {
-  extern __shared__ __attribute__((aligned(16))) char raw[];
-  for (int64_t sb = blockIdx.x; sb < nslabs; sb += gridDim.x) {
-    int64_t nb = sb / slabs_per_img;
-    int og0 = (int)(sb - nb * slabs_per_img) * G;
-    int rows = OH - og0 < G ? OH - og0 : G;
-    int64_t in_row = (int64_t)W * C;        // one input row
-    int64_t out_row = (int64_t)OW * C;
-    T* in_slab = (T*)raw;                   // 2*rows input rows
-    T* out_slab = in_slab + 2 * (int64_t)G * in_row;
-    uint8_t* code_slab = (uint8_t*)(out_slab + (int64_t)G * out_row);
-
-    int64_t H = (int64_t)OH * 2;
-    int64_t ie0 = (nb * H + 2 * og0) * in_row;
-    flat_g2l(in_slab, x, ie0, ie0 + 2 * (int64_t)rows * in_row);
-    __syncthreads();
-    int64_t total = (int64_t)rows * out_row;
-    for (int64_t t = threadIdx.x; t < total; t += blockDim.x) {
-      int lr = (int)(t / out_row);
-      int rem = (int)(t - (int64_t)lr * out_row);
-      int ow = rem / C;
-      int c = rem - ow * C;
-      const T* base = in_slab + (2 * (int64_t)lr * W + 2 * ow) * C + c;
-      float v00 = to_f32(base[0]);
-      float v01 = to_f32(base[C]);
-      float v10 = to_f32(base[in_row]);
-      float v11 = to_f32(base[in_row + C]);
-      float m = v00;
-      int k = 0;
-      if (v01 > m) { m = v01; k = 1; }
-      if (v10 > m) { m = v10; k = 2; }
-      if (v11 > m) { m = v11; k = 3; }
-      out_slab[t] = from_f32<T>(m);
-      code_slab[t] = (uint8_t)k;
-    }
-    __syncthreads();
-    int64_t oe0 = (nb * OH + og0) * out_row;
-    flat_l2g(y, out_slab, oe0, oe0 + total);
-    flat_l2g_u8(code, code_slab, oe0, oe0 + total);
-    __syncthreads();
-  }
-}
-
-template <typename T>
-__global__ void maxpool2x2_bwd_lds_kernel(const T* __restrict__ g,
-                                          const uint8_t* __restrict__ code,
-                                          T* __restrict__ gx, int C, int W,
-                                          int OH, int OW, int G,
-                                          int slabs_per_img, int64_t nslabs) {
-  extern __shared__ __attribute__((aligned(16))) char raw[];
-  for (int64_t sb = blockIdx.x; sb < nslabs; sb += gridDim.x) {
-    int64_t nb = sb / slabs_per_img;
-    int og0 = (int)(sb - nb * slabs_per_img) * G;
-    int rows = OH - og0 < G ? OH - og0 : G;
-    int64_t in_row = (int64_t)W * C;
-    int64_t out_row = (int64_t)OW * C;
-    T* in_slab = (T*)raw;                   // 2*rows of gx
-    T* g_slab = in_slab + 2 * (int64_t)G * in_row;
-    uint8_t* code_slab = (uint8_t*)(g_slab + (int64_t)G * out_row);
-
-    int64_t oe0 = (nb * OH + og0) * out_row;
-    int64_t ototal = (int64_t)rows * out_row;
-    flat_g2l(g_slab, g, oe0, oe0 + ototal);
-    flat_g2l_u8(code_slab, code, oe0, oe0 + ototal);
-    __syncthreads();
-    const T zero = from_f32<T>(0.0f);
-    for (int64_t t = threadIdx.x; t < ototal; t += blockDim.x) {
-      int lr = (int)(t / out_row);
-      int rem = (int)(t - (int64_t)lr * out_row);
-      int ow = rem / C;
-      int c = rem - ow * C;
-      int k = code_slab[t];
-      T gv = g_slab[t];
-      T* base = in_slab + (2 * (int64_t)lr * W + 2 * ow) * C + c;
-      base[0] = (k == 0) ? gv : zero;
-      base[C] = (k == 1) ? gv : zero;
-      base[in_row] = (k == 2) ? gv : zero;
-      base[in_row + C] = (k == 3) ? gv : zero;
-    }
-    __syncthreads();
-    int64_t H = (int64_t)OH * 2;
-    int64_t ie0 = (nb * H + 2 * og0) * in_row;
-    flat_l2g(gx, in_slab, ie0, ie0 + 2 * (int64_t)rows * in_row);
-    __syncthreads();
-  }
-}
-
 template <typename scalar_t> struct DevT { using type = scalar_t; };
 template <> struct DevT<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct DevT<at::Half> { using type = _Float16; };
@@ -273,28 +92,10 @@ std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor x) {
   NN_DISPATCH(x.scalar_type(),
                                   "maxpool2x2_fwd", [&] {
     using T = typename DevT<scalar_t>::type;
-    int64_t per_row = 2 * (int64_t)W * C * sizeof(T)
-                      + (int64_t)OW * C * sizeof(T) + (int64_t)OW * C;
-    if (sizeof(T) == 2 && (H % 2) == 0 && (W % 2) == 0
-        && per_row <= 56 * 1024) {
-      int G = (int)std::min<int64_t>(OH, (56 * 1024) / per_row);
-      int spi = (OH + G - 1) / G;
-      int64_t nslabs = (int64_t)N * spi;
-      int lb = (int)std::min<int64_t>(nslabs, 8192);
-      hipLaunchKernelGGL((maxpool2x2_fwd_lds_kernel<T>), dim3(lb),
-                         dim3(kBlock), (size_t)G * per_row,
-                         c10::hip::getCurrentHIPStream(),
-                         (const T*)x.data_ptr(), (T*)y.data_ptr(),
-                         code.data_ptr<uint8_t>(), C, W, OH, OW, G, spi,
-                         nslabs);
-    } else {
-      hipLaunchKernelGGL((maxpool2x2_fwd_kernel<T>), dim3(blocks),
-                         dim3(kBlock), 0,
-                         c10::hip::getCurrentHIPStream(),
-                         (const T*)x.data_ptr(),
-                         (T*)y.data_ptr(), code.data_ptr<uint8_t>(), n_out, C,
-                         H, W, OH, OW);
-    }
+    hipLaunchKernelGGL((maxpool2x2_fwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)x.data_ptr(),
+                       (T*)y.data_ptr(), code.data_ptr<uint8_t>(), n_out, C, H,
+                       W, OH, OW);
   });
   HIP_CHECK_LAST();
   return {y, code};
@@ -313,28 +114,10 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor g, torch::Tensor code, int64_t H,
   NN_DISPATCH(g.scalar_type(),
                                   "maxpool2x2_bwd", [&] {
     using T = typename DevT<scalar_t>::type;
-    int64_t per_row = 2 * (int64_t)W * C * sizeof(T)
-                      + (int64_t)OW * C * sizeof(T) + (int64_t)OW * C;
-    if (sizeof(T) == 2 && (H % 2) == 0 && (W % 2) == 0
-        && per_row <= 56 * 1024) {
-      int G = (int)std::min<int64_t>(OH, (56 * 1024) / per_row);
-      int spi = (OH + G - 1) / G;
-      int64_t nslabs = (int64_t)N * spi;
-      int lb = (int)std::min<int64_t>(nslabs, 8192);
-      hipLaunchKernelGGL((maxpool2x2_bwd_lds_kernel<T>), dim3(lb),
-                         dim3(kBlock), (size_t)G * per_row,
-                         c10::hip::getCurrentHIPStream(),
-                         (const T*)g.data_ptr(), code.data_ptr<uint8_t>(),
-                         (T*)gx.data_ptr(), C, (int)W, OH, OW, G, spi,
-                         nslabs);
-    } else {
-      hipLaunchKernelGGL((maxpool2x2_bwd_kernel<T>), dim3(blocks),
-                         dim3(kBlock), 0,
-                         c10::hip::getCurrentHIPStream(),
-                         (const T*)g.data_ptr(),
-                         code.data_ptr<uint8_t>(), (T*)gx.data_ptr(), n_out,
-                         C, (int)H, (int)W, OH, OW);
-    }
+    hipLaunchKernelGGL((maxpool2x2_bwd_kernel<T>), dim3(blocks), dim3(kBlock), 0,
+                       c10::hip::getCurrentHIPStream(), (const T*)g.data_ptr(),
+                       code.data_ptr<uint8_t>(), (T*)gx.data_ptr(), n_out, C,
+                       (int)H, (int)W, OH, OW);
   });
   HIP_CHECK_LAST();
   return gx;
