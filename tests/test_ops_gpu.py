@@ -959,3 +959,19 @@ def test_condconv_per_sample_gpu():
     out.sum().backward()
     assert x.grad is not None and m.weight.grad is not None
 
+
+
+def test_maxpool2x2_even_dims_odd_channels():
+    """2x2 pool fwd+bwd vs torch on even spatial dims with odd channel
+    counts (the NHWC alignment corner)."""
+    torch.manual_seed(31)
+    for N, C, H, W in ((4, 65, 28, 28), (8, 120, 10, 10), (2, 7, 8, 12)):
+        x = to_bf16_representable(torch.randn(N, C, H, W)).cuda().bfloat16()
+        y, codev = ops.ext().maxpool2x2_fwd(cl(x))
+        ye = F.max_pool2d(x.float().cpu(), 2)
+        assert torch.allclose(y.float().cpu(), ye, atol=1e-2), (N, C, H, W)
+        g = to_bf16_representable(torch.randn_like(ye)).cuda().bfloat16()
+        gx = ops.ext().maxpool2x2_bwd(cl(g), codev, H, W)
+        xr = x.float().cpu().requires_grad_(True)
+        F.max_pool2d(xr, 2).backward(g.float().cpu())
+        assert torch.allclose(gx.float().cpu(), xr.grad, atol=1e-2), (N, C)
