@@ -27,9 +27,12 @@ from noisynet_amd.quant import finish_calibration, start_calibration  # noqa: E4
 
 
 def train_one(current, epochs, X, y, Xt, yt, device, seed=11, batch=64):
+    # exact README flagship flags (README.md:6-9); the reference's default
+    # optimizer is AdamW (noisynet.py:249) -- forcing SGD here was measured
+    # ~50 points worse at 1 nA
     argv = ['--q_a', '4', '--act_max', '5', '--w_max1', '0.3', '--LR',
             '0.005', '--L2_1', '0.0005', '--L2_2', '0.0002', '--batch_size',
-            str(batch), '--optim', 'SGD', '--calculate_running']
+            str(batch), '--calculate_running']
     if current > 0:
         argv = ['--current', str(current)] + argv
     args = build_noisynet_parser().parse_args(argv)
@@ -40,7 +43,7 @@ def train_one(current, epochs, X, y, Xt, yt, device, seed=11, batch=64):
     model = model.to(device)
     if device.type == 'cuda':
         model = model.to(memory_format=torch.channels_last)
-    opt = native_optim.SGD([
+    groups = [
         {'params': model.conv1.parameters(), 'weight_decay': args.L2_1,
          'lr': args.LR, 'clamp': (-args.w_max1, args.w_max1)},
         {'params': model.conv2.parameters(), 'weight_decay': args.L2_2,
@@ -51,7 +54,12 @@ def train_one(current, epochs, X, y, Xt, yt, device, seed=11, batch=64):
          'lr': args.LR},
         {'params': [p for m in (model.bn1, model.bn2, model.bn3, model.bn4)
                     for p in m.parameters()], 'weight_decay': 0.0,
-         'lr': args.LR}], lr=args.LR, momentum=args.momentum, nesterov=True)
+         'lr': args.LR}]
+    if args.optim == 'SGD':
+        opt = native_optim.SGD(groups, lr=args.LR, momentum=args.momentum,
+                               nesterov=True)
+    else:  # AdamW: the reference's (and this driver's) default
+        opt = native_optim.AdamW(groups, lr=args.LR)
     n = X.shape[0]
     start_calibration(model)
     curve = []
