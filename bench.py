@@ -15,8 +15,8 @@ Each timed step is a FULL training step on a FRESH batch sliced from a
 GPU-resident dataset, with the reference's GPU crop/flip augmentation
 (noisynet.py:1261-1269) inside the timed region: forward (fused
 quant+conv+sigma+noise kernels), fused softmax-xent, backward (dgrad/
-wgrad), bucketed RCCL all-reduce overlapped with backward, fused SGD
-update with weight clamp.
+wgrad), bucketed RCCL all-reduce overlapped with backward, fused AdamW
+update with the weight clamp folded in (the recipe's default optimizer).
 """
 
 import argparse
@@ -78,8 +78,7 @@ def flagship_args(bench):
     """BASELINE.json config 2: NoisyNet CIFAR-10 4-bit, I_max=1nA, act_max=5."""
     argv = ["--q_a", "4", "--act_max", "5", "--w_max1", "0.3",
             "--LR", "0.005", "--L2_1", "0.0005", "--L2_2", "0.0002",
-            "--batch_size", str(bench.batch), "--optim", "SGD",
-            "--calculate_running"]
+            "--batch_size", str(bench.batch), "--calculate_running"]
     if bench.no_augment:
         argv += ["--no-augment"]
     if not bench.no_noise:
@@ -171,8 +170,8 @@ def run_top1(bench, device, dtype, rank, distributed):
                     for p in m.parameters()], "weight_decay": 0.0,
          "lr": args.LR},
     ]
-    optimizer = native_optim.SGD(param_groups, lr=args.LR,
-                                 momentum=args.momentum, nesterov=True)
+    # the recipe's real optimizer: AdamW (reference noisynet.py:249 default)
+    optimizer = native_optim.AdamW(param_groups, lr=args.LR)
     bs = args.batch_size
     n = X.shape[0]
     start_calibration(model)
@@ -275,8 +274,9 @@ def main():
                         for p in m.parameters()], "weight_decay": 0.0,
              "lr": args.LR},
         ]
-        optimizer = native_optim.SGD(param_groups, lr=args.LR,
-                                     momentum=args.momentum, nesterov=True)
+        # fused AdamW: the flagship recipe's default optimizer
+        # (reference noisynet.py:249)
+        optimizer = native_optim.AdamW(param_groups, lr=args.LR)
     else:
         optimizer = native_optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                                      weight_decay=1e-4, nesterov=False)
@@ -412,6 +412,7 @@ def main():
                 "current_nA": (0 if (bench.no_noise or bench.model != "noisynet")
                                else 1),
                 "act_max": 5 if bench.model == "noisynet" else 0,
+                "optim": ("adamw" if bench.model == "noisynet" else "sgd"),
                 "parallelism": "dp%d" % n_gpus,
             },
         }

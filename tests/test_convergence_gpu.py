@@ -95,8 +95,7 @@ def test_flagship_convergence_top1_gpu():
     reaches real held-out accuracy on the learnable synthetic CIFAR."""
     argv = ['--current', '1', '--q_a', '4', '--act_max', '5', '--w_max1',
             '0.3', '--LR', '0.005', '--L2_1', '0.0005', '--L2_2', '0.0002',
-            '--batch_size', '64', '--optim', 'SGD', '--calculate_running',
-            '--no-augment']
+            '--batch_size', '64', '--calculate_running', '--no-augment']
     args = build_noisynet_parser().parse_args(argv)
     broadcast_per_layer(args)
     torch.manual_seed(11)
@@ -113,8 +112,8 @@ def test_flagship_convergence_top1_gpu():
         memory_format=torch.channels_last)
     yt = torch.from_numpy(tel).cuda()
 
-    opt = native_optim.SGD(model.parameters(), lr=args.LR, momentum=0.9,
-                           nesterov=True)
+    # the recipe's real optimizer: AdamW (reference noisynet.py:249 default)
+    opt = native_optim.AdamW(model.parameters(), lr=args.LR)
     bs = 64
     start_calibration(model)
     model.train()
@@ -138,8 +137,7 @@ def test_flagship_convergence_top1_gpu():
             out = model(Xt[i:i + 1000], 0, 100)
             correct += int((out.argmax(1) == yt[i:i + 1000]).sum())
     top1 = 100.0 * correct / Xt.shape[0]
-    # random = 10%; the noisy quantized model must genuinely classify after
-    # this short run (measured ~29% at 10 epochs and still climbing; the
-    # full convergence curve lives in profiles/accuracy_curve.md via
-    # tools/accuracy_curve.py / bench.py --top1)
-    assert top1 > 20.0, top1
+    # random = 10%; with the recipe's real optimizer (AdamW) the noisy
+    # quantized model classifies strongly within a few epochs (full curve:
+    # profiles/accuracy_curve.md)
+    assert top1 > 60.0, top1
