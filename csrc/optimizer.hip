@@ -73,7 +73,12 @@ void sgd_step(torch::Tensor p, torch::Tensor g, torch::Tensor buf, double lr,
   bool mom = momentum != 0.0;
   bool clamp = cmax > cmin;
   TORCH_CHECK(!mom || buf.numel() == n, "sgd_step: momentum buffer size");
-  auto gc = g.contiguous();
+  // align the grad to the PARAM's raw layout: a plain .contiguous() turned
+  // channels_last grads into NCHW order, silently pairing every 4-D conv
+  // param element with the WRONG grad element (models trained through the
+  // channels_last path learned drastically worse; the buffers from
+  // zeros_like(p) already share p's layout)
+  auto gc = g.contiguous(p.suggest_memory_format());
   NN_DISPATCH(p.scalar_type(),
                                   "sgd_step", [&] {
     using T = typename DevT<scalar_t>::type;
@@ -105,7 +110,7 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
   bool clamp = cmax > cmin;
   float bc1 = 1.0f - ::powf((float)beta1, (float)step);
   float bc2 = 1.0f - ::powf((float)beta2, (float)step);
-  auto gc = g.contiguous();
+  auto gc = g.contiguous(p.suggest_memory_format());  // see sgd_step
   NN_DISPATCH(p.scalar_type(),
                                   "adamw_step", [&] {
     using T = typename DevT<scalar_t>::type;

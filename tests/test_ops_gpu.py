@@ -975,3 +975,41 @@ def test_maxpool2x2_even_dims_odd_channels():
         xr = x.float().cpu().requires_grad_(True)
         F.max_pool2d(xr, 2).backward(g.float().cpu())
         assert torch.allclose(gx.float().cpu(), xr.grad, atol=1e-2), (N, C)
+
+
+def test_fused_optimizers_channels_last_alignment():
+    """Fused SGD/AdamW on CHANNELS_LAST params must update each element
+    with ITS OWN gradient: regression for the wrapper's plain
+    .contiguous() silently re-ordering channels_last grads to NCHW (every
+    cl-trained model mis-learned; the NCHW driver was unaffected)."""
+    torch.manual_seed(41)
+    from noisynet_amd import optim as native_optim
+    for opt_cls, torch_cls, kw in (
+            (native_optim.SGD, torch.optim.SGD,
+             dict(lr=0.05, momentum=0.9, nesterov=True)),
+            (native_optim.AdamW, torch.optim.AdamW, dict(lr=0.05))):
+        w0 = torch.randn(16, 8, 3, 3)
+        p_cl = torch.nn.Parameter(
+            w0.clone().cuda().contiguous(memory_format=torch.channels_last))
+        p_ref = torch.nn.Parameter(w0.clone().cuda())
+        opt = opt_cls([p_cl], **kw)
+        ref = torch_cls([p_ref], **kw)
+        for i in range(3):
+            g = torch.randn(16, 8, 3, 3).cuda()
+            # grads arrive channels_last (autograd mirrors the producer)
+            p_cl.grad = g.contiguous(memory_format=torch.channels_last)
+            p_ref.grad = g.clone()
+            opt.step()
+            ref.step()
+        d = (p_cl.detach().float().contiguous() -
+             p_ref.detach().float()).abs().max().item()
+        assert d < 1e-4, (opt_cls.__name__, d)
+        # and NCHW grads on cl params must also align
+        g = torch.randn(16, 8, 3, 3).cuda()
+        p_cl.grad = g.clone()
+        p_ref.grad = g.clone()
+        opt.step()
+        ref.step()
+        d = (p_cl.detach().float().contiguous() -
+             p_ref.detach().float()).abs().max().item()
+        assert d < 1e-4, (opt_cls.__name__, 'nchw-grad', d)
