@@ -135,8 +135,6 @@ def run_top1(bench, device, dtype, rank, distributed):
     """Train the flagship to convergence on the learnable synthetic CIFAR
     and return held-out top-1 (the accuracy half of the BASELINE metric,
     reference README.md:6-13 semantics on synthetic data)."""
-    import numpy as np
-
     args = flagship_args(bench)
     args.batch_size = 64  # the reference recipe's batch (noisynet.py default)
     args.augment = True
