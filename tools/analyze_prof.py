@@ -20,7 +20,7 @@ rows = list(cur.execute(
         FROM rocpd_kernel_dispatch_{sfx} k
         JOIN rocpd_info_kernel_symbol_{sfx} ks ON k.kernel_id = ks.id
         ORDER BY k.start"""))
-opt = [r for r in rows if 'sgd' in str(r[2]).lower()]
+opt = [r for r in rows if 'sgd' in str(r[2]).lower() or 'adamw' in str(r[2]).lower()]
 steps, cs = [], [opt[0]]
 for r in opt[1:]:
     if r[0] - cs[-1][0] > 1e6:
