@@ -535,3 +535,55 @@ def test_optim_factory_fused_names():
         opt.zero_grad()
         loss.backward()
         opt.step()
+
+
+def test_nvnovograd_amsgrad_branch():
+    """NvNovoGrad AMSGrad variant: the max-of-second-moments denominator
+    (reference nvnovograd.py:100-106)."""
+    import torch
+
+    from noisynet_amd.timm.optim import NvNovoGrad
+
+    torch.manual_seed(1)
+    w = torch.nn.Parameter(torch.randn(3, 3))
+    opt = NvNovoGrad([w], lr=0.05, amsgrad=True)
+    losses = []
+    for _ in range(25):
+        loss = (w ** 2).sum()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < 0.3 * losses[0]
+    st = opt.state[w]
+    assert 'max_exp_avg_sq' in st
+    assert float(st['max_exp_avg_sq']) >= float(st['exp_avg_sq']) - 1e-8
+
+
+def test_scheduler_step_update_per_iteration():
+    """Cosine scheduler's per-iteration step_update lowers LR smoothly
+    between epoch boundaries (reference cosine_lr.py + per-iteration
+    t_in_epochs=False use, train_efficientnet.py:544-545)."""
+    import types
+
+    import torch
+
+    from noisynet_amd.timm.scheduler import create_scheduler
+
+    m = torch.nn.Linear(2, 2)
+    opt = torch.optim.SGD(m.parameters(), lr=0.4)
+    args = types.SimpleNamespace(sched='cosine', epochs=4, min_lr=1e-5,
+                                 warmup_lr=1e-4, warmup_epochs=1,
+                                 decay_rate=0.1, decay_epochs=1,
+                                 cooldown_epochs=0, lr_cycle_mul=1.0,
+                                 lr_cycle_limit=1, seed=0, lr=0.4,
+                                 patience_epochs=2)
+    sched, num_epochs = create_scheduler(args, opt)
+    assert num_epochs >= 4
+    sched.step(0)
+    lrs = []
+    for update in range(0, 40, 10):
+        sched.step_update(update)
+        lrs.append(opt.param_groups[0]['lr'])
+    # warmup: LR should be rising early and stay bounded by base lr
+    assert lrs[0] <= lrs[-1] <= 0.4 + 1e-9
